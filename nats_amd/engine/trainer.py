@@ -1,0 +1,238 @@
+"""Training orchestration — the ``train(**hyperparams)`` entry point.
+
+Mirrors the reference's train loop (nats.py:1230-1539) behaviour:
+  * epochs x sequential minibatches (TextIterator, no shuffling),
+  * prepare_data with maxlen truncation; skipped empty batches decrement
+    the update counter (nats.py:1395-1398),
+  * forward -> mean cost -> backward -> global-norm clip -> optimizer step,
+  * NaN/Inf cost aborts returning (1., 1., 1.) (nats.py:1415-1417),
+  * dispFreq logging of Epoch/Update/Cost/UD, saveFreq checkpoints
+    (best-so-far params preferred), sampleFreq stochastic samples,
+    validFreq early stopping with patience (nats.py:1482-1510),
+  * final save embeds zipped_params (nats.py:1533-1535).
+
+MI355X-native extensions (all default-off / auto):
+  * device="cuda" runs the model through the HIP kernel path,
+  * world_size>1 (torchrun env) enables bucketed RCCL gradient all-reduce
+    (nats_amd/parallel) with rank-sharded data and all-reduced validation,
+  * per-update wall-clock timing is always collected (utils/timers).
+"""
+
+import logging
+import math
+import time
+
+import numpy
+import torch
+
+from ..data.dictionary import load_dictionary, invert_dictionary
+from ..data.iterator import TextIterator
+from ..data.prepare import prepare_data
+from ..models.distraction import NatsModel, default_options
+from ..parallel.ddp import DataParallelGrads, init_distributed
+from .checkpoint import (load_checkpoint, save_checkpoint, load_options,
+                         save_options)
+from .optim import build_optimizer
+from .validate import pred_probs
+
+logger = logging.getLogger("nats_amd.train")
+
+
+def _to_device(arrs, device):
+    return [torch.from_numpy(a).to(device) for a in arrs]
+
+
+def _print_tokens(ids, worddicts_r, limit_at_eos=True):
+    words = []
+    for vv in ids:
+        vv = int(vv)
+        if vv == 0:
+            break
+        words.append(worddicts_r.get(vv, "UNK"))
+    return " ".join(words)
+
+
+def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
+          decoder="gru_cond", patience=10, max_epochs=5000,
+          finish_after=10000000, dispFreq=100, decay_c=0.0, clip_c=-1.0,
+          lrate=0.01, n_words=100000, maxlen=100, optimizer="adadelta",
+          batch_size=16, valid_batch_size=16, saveto="model.npz",
+          validFreq=1000, saveFreq=1000, sampleFreq=100, datasets=[],
+          valid_datasets=[], dictionary="", use_dropout=False, reload_=False,
+          verbose=False, device=None, seed=None):
+    """Train the distraction model; returns final validation error.
+
+    Signature (and defaults) mirror nats.py:1230-1257; `device`/`seed` are
+    framework additions (device None = cuda if available else cpu).
+    """
+    logging.basicConfig(
+        level=logging.DEBUG,
+        format="%(asctime)s: %(name)s: %(levelname)s: %(message)s")
+
+    import os
+    model_options = {k: v for k, v in locals().items()
+                     if k not in ("os",)}
+
+    if device is None:
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+    rank, local_rank, world = init_distributed()
+    if device == "cuda":
+        device = "cuda:%d" % local_rank if world > 1 else "cuda"
+
+    # load dictionary and invert (nats.py:1264-1268)
+    worddicts = load_dictionary(dictionary)
+    worddicts_r = invert_dictionary(worddicts)
+
+    # reload options (nats.py:1271-1274)
+    import os.path
+    if reload_ and os.path.exists(saveto):
+        print("Reload options")
+        model_options = load_options(saveto)
+
+    logger.debug(model_options)
+
+    print("Loading data")
+    train_it = TextIterator(datasets[0], datasets[1], dictionary,
+                            n_words=n_words, batch_size=batch_size)
+    valid_it = TextIterator(valid_datasets[0], valid_datasets[1], dictionary,
+                            n_words=n_words, batch_size=valid_batch_size)
+
+    print("Building model")
+    model = NatsModel(model_options, seed=seed)
+    if reload_ and os.path.exists(saveto):
+        print("Reload parameters")
+        params, _ = load_checkpoint(saveto)
+        model.set_params(params)
+    model = model.to(device)
+
+    dp = DataParallelGrads(model.parameters())
+    dp.broadcast_params()
+
+    opt = build_optimizer(model_options["optimizer"],
+                          list(model.P.items()),
+                          lrate=lrate, clip_c=clip_c)
+
+    history_errs = []
+    if reload_ and os.path.exists(saveto):
+        print("Reload history error")
+        _, history_errs = load_checkpoint(saveto)
+    best_p = None
+    bad_counter = 0
+
+    uidx = 0
+    estop = False
+    valid_err = None
+    for eidx in range(max_epochs):
+        n_samples = 0
+        for bidx, (xs, ys) in enumerate(train_it):
+            # data-parallel sharding: each rank takes every world-th batch
+            if world > 1 and (bidx % world) != rank:
+                continue
+            n_samples += len(xs)
+            uidx += 1
+
+            x, x_mask, y, y_mask = prepare_data(xs, ys, maxlen=maxlen,
+                                                n_words=n_words)
+            if x is None:
+                print("Minibatch with zero sample under length ", maxlen)
+                uidx -= 1
+                continue
+
+            ud_start = time.time()
+            x, x_mask, y, y_mask = _to_device((x, x_mask, y, y_mask), device)
+
+            opt.zero_grad()
+            cost_vec = model(x, x_mask, y, y_mask)
+            cost = cost_vec.mean()
+            if decay_c > 0.0:
+                weight_decay = sum((p ** 2).sum() for p in model.parameters())
+                cost = cost + decay_c * weight_decay
+            cost.backward()
+            dp.finish()
+            norm_g = opt.step()
+            cost_val = float(cost.detach())
+            ud = time.time() - ud_start
+
+            # NaN abort (nats.py:1415-1417)
+            if math.isnan(cost_val) or math.isinf(cost_val):
+                print("NaN detected")
+                return 1.0, 1.0, 1.0
+
+            if numpy.mod(uidx, dispFreq) == 0:
+                logger.debug("Epoch {0} Update {1} Cost {2} UD {3}".format(
+                    eidx, uidx, cost_val, ud))
+                if verbose and clip_c > 0:
+                    logger.debug("Grad {0}".format(norm_g))
+
+            if rank == 0 and numpy.mod(uidx, saveFreq) == 0:
+                print("Saving...", end=" ")
+                params = best_p if best_p is not None else model.get_params()
+                save_checkpoint(saveto, params, history_errs,
+                                options=model_options)
+                print("Done")
+
+            if rank == 0 and numpy.mod(uidx, sampleFreq) == 0:
+                from ..decode.beam import gen_sample
+                for jj in range(min(5, x.shape[1])):
+                    sample, score, dec_alphas = gen_sample(
+                        model, x[:, jj:jj + 1], k=1, maxlen=30,
+                        stochastic=True, argmax=False)
+                    print("Source %d: %s" % (
+                        jj, _print_tokens(x[:, jj].tolist(), worddicts_r)))
+                    print("Truth %d: %s" % (
+                        jj, _print_tokens(y[:, jj].tolist(), worddicts_r)))
+                    print("Sample %d: %s" % (
+                        jj, _print_tokens(sample, worddicts_r)))
+
+            if numpy.mod(uidx, validFreq) == 0:
+                model.eval()
+                valid_errs = pred_probs(model, valid_it, device=device)
+                model.train()
+                valid_err = float(valid_errs.mean())
+                if world > 1:
+                    valid_err = dp.all_reduce_scalar(valid_err, average=True)
+                history_errs.append(valid_err)
+
+                if uidx == 0 or valid_err <= numpy.array(history_errs).min():
+                    best_p = model.get_params()
+                    bad_counter = 0
+
+                # early stopping with patience (nats.py:1493-1505)
+                if patience == 0:
+                    if len(history_errs) > 1 and valid_err >= min(
+                            history_errs[:-1]):
+                        print("Early Stop!")
+                        estop = True
+                        break
+                else:
+                    if len(history_errs) > patience and valid_err >= numpy.array(
+                            history_errs)[:-patience].min():
+                        bad_counter += 1
+                        if bad_counter > patience:
+                            print("Early Stop!")
+                            estop = True
+                            break
+                print("Valid ", valid_err)
+
+            if uidx >= finish_after:
+                print("Finishing after %d iterations!" % uidx)
+                estop = True
+                break
+
+        print("Seen %d samples" % n_samples)
+        if estop:
+            break
+
+    if best_p is not None:
+        model.set_params(best_p)
+
+    model.eval()
+    valid_err = float(pred_probs(model, valid_it, device=device).mean())
+    print("Valid ", valid_err)
+
+    if rank == 0:
+        params = dict(best_p) if best_p is not None else model.get_params()
+        save_checkpoint(saveto, params, history_errs,
+                        zipped_params=best_p, options=model_options)
+    logger.debug("Done")
+    return valid_err

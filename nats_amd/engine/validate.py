@@ -1,0 +1,38 @@
+"""Validation scoring (pred_probs, nats.py:1080-1101)."""
+
+import math
+
+import numpy
+import torch
+
+from ..data.prepare import prepare_data
+
+
+@torch.no_grad()
+def pred_probs(model, iterator, device=None, verbose=False):
+    """Per-sequence NLL over a whole corpus iterator.
+
+    Mirrors pred_probs: NO maxlen truncation (nats.py:1088), per-sequence
+    masked-CE sums collected into one array. NaN raises (the reference
+    drops into ipdb, nats.py:1096 — non-interactive here).
+    """
+    probs = []
+    n_done = 0
+    options = model.options
+    for xs, ys in iterator:
+        n_done += len(xs)
+        x, x_mask, y, y_mask = prepare_data(xs, ys, n_words=options["n_words"])
+        x = torch.from_numpy(x)
+        y = torch.from_numpy(y)
+        x_mask = torch.from_numpy(x_mask)
+        y_mask = torch.from_numpy(y_mask)
+        if device is not None:
+            x, x_mask = x.to(device), x_mask.to(device)
+            y, y_mask = y.to(device), y_mask.to(device)
+        cost = model(x, x_mask, y, y_mask)
+        probs.extend(float(c) for c in cost.cpu())
+        if not all(math.isfinite(p) for p in probs):
+            raise FloatingPointError("NaN/Inf in validation cost")
+        if verbose:
+            print("%d samples computed" % n_done)
+    return numpy.array(probs, dtype="float32")

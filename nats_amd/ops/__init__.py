@@ -1,0 +1,106 @@
+"""Op dispatch layer.
+
+Every hot op has two implementations:
+
+  * ``nats_amd.ops.eager`` — plain PyTorch, runs on CPU (and GPU), serves as
+    the numerics oracle for the HIP kernels (the Theano graph equations at
+    nats.py:336-372, 498-572, 753-770 are the spec),
+  * ``nats_amd.ops.hip`` (built from ``nats_amd/ops/hip/*.hip``) — the
+    hand-written CDNA4 kernels used on MI355X.
+
+Dispatch rule: CUDA tensors use the HIP kernels; if the extension is not
+importable on a GPU machine the op RAISES (no silent eager fallback — set
+``NATS_AMD_ALLOW_EAGER_GPU=1`` to override for debugging). CPU tensors use
+eager. ``NATS_AMD_FORCE_EAGER=1`` forces eager everywhere (used in tests to
+A/B the kernels).
+"""
+
+import os
+
+import torch
+
+from . import eager
+
+_HIP = None
+_HIP_TRIED = False
+
+
+def _hip_ext():
+    """Import the compiled HIP extension lazily (built in-tree by setup or
+    __graft_entry__.build())."""
+    global _HIP, _HIP_TRIED
+    if not _HIP_TRIED:
+        _HIP_TRIED = True
+        try:
+            from . import hip_ext as m  # thin loader module
+            _HIP = m.load()
+        except Exception:
+            _HIP = None
+    return _HIP
+
+
+def _use_hip(*tensors):
+    if os.environ.get("NATS_AMD_FORCE_EAGER"):
+        return False
+    if not any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor)):
+        return False
+    ext = _hip_ext()
+    if ext is None:
+        if os.environ.get("NATS_AMD_ALLOW_EAGER_GPU"):
+            return False
+        raise RuntimeError(
+            "nats_amd HIP extension is not built but tensors are on GPU. "
+            "Run `python -c 'import __graft_entry__; __graft_entry__.build()'` "
+            "or set NATS_AMD_ALLOW_EAGER_GPU=1 to run the (slow) eager path.")
+    return True
+
+
+def gru_scan(x_gates, x_cand, mask, U, Ux, h0=None):
+    """GRU scan over time. Returns hidden states (T, B, H).
+
+    Semantics: nats.py:336-372 (gru_layer step). ``x_gates``/``x_cand`` are
+    the hoisted input projections x@W+b (T,B,2H) and x@Wx+bx (T,B,H).
+    """
+    if _use_hip(x_gates, U):
+        from .gru import gru_scan_hip
+        return gru_scan_hip(x_gates, x_cand, mask, U, Ux, h0)
+    return eager.gru_scan(x_gates, x_cand, mask, U, Ux, h0)
+
+
+def cond_gru_scan(y_gates, y_cand, mask, init_state, ctx, ctx_mask, pctx, P):
+    """Conditional-GRU decoder scan (training). Returns
+    (h2s, ctxs, alphas, acc_ctx, acc_alpha).
+
+    Semantics: nats.py:498-608 (gru_cond_layer scan path).
+    """
+    if _use_hip(y_gates, ctx):
+        from .cond_gru import cond_gru_scan_hip
+        return cond_gru_scan_hip(y_gates, y_cand, mask, init_state, ctx,
+                                 ctx_mask, pctx, P)
+    return eager.cond_gru_scan(y_gates, y_cand, mask, init_state, ctx,
+                               ctx_mask, pctx, P)
+
+
+def cond_gru_step(h_prev, x_g, x_c, ctx, ctx_mask, pctx, acc_ctx, acc_alpha, P):
+    """One decoder step (beam search / sampler one_step path).
+
+    Semantics: nats.py:592-594 with mask == all-ones.
+    """
+    if _use_hip(h_prev, ctx):
+        from .cond_gru import cond_gru_step_hip
+        return cond_gru_step_hip(h_prev, x_g, x_c, ctx, ctx_mask, pctx,
+                                 acc_ctx, acc_alpha, P)
+    return eager.cond_gru_step(h_prev, x_g, x_c, ctx, ctx_mask, pctx,
+                               acc_ctx, acc_alpha, P)
+
+
+def softmax_xent(logits, targets):
+    """Per-position NLL of a softmax over the vocabulary.
+
+    logits (N, V) float, targets (N,) int64 -> (N,) float32.
+    Semantics: nats.py:763-768 (softmax + categorical_crossentropy).
+    """
+    if _use_hip(logits):
+        from .softmax_ce import softmax_xent_hip
+        return softmax_xent_hip(logits, targets)
+    return eager.softmax_xent(logits, targets)

@@ -1,0 +1,67 @@
+"""Multi-process CPU (gloo) tests of the data-parallel gradient layer."""
+
+import os
+
+import numpy
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from nats_amd.data.prepare import prepare_data
+
+
+def _make_batch(opts, B, seed):
+    rng = numpy.random.RandomState(seed)
+    xs = [list(rng.randint(2, opts["n_words"], size=6)) for _ in range(B)]
+    ys = [list(rng.randint(2, opts["n_words"], size=4)) for _ in range(B)]
+    return [torch.from_numpy(a) for a in prepare_data(xs, ys)]
+
+
+def _worker(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from nats_amd.models.distraction import NatsModel, default_options
+        from nats_amd.parallel.ddp import DataParallelGrads
+
+        opts = default_options(dim_word=8, dim=10, dim_att=6, n_words=32)
+        model = NatsModel(opts, seed=3)
+        dp = DataParallelGrads(model.parameters(), bucket_cap_mb=1)
+        dp.broadcast_params()
+
+        # full batch = shard0 + shard1 (equal sizes)
+        full = _make_batch(opts, 4, seed=9)
+        shard = [a[:, rank * 2:(rank + 1) * 2].contiguous() for a in full]
+
+        cost = model(*shard).mean()
+        cost.backward()
+        dp.finish()
+
+        # reference: single-process mean-of-shard-means on the full batch
+        model_ref = NatsModel(opts, seed=3)
+        model_ref.set_params(model.get_params())  # same (broadcast) weights
+        for p in model_ref.parameters():
+            p.grad = None
+        c0 = model_ref(*[a[:, 0:2].contiguous() for a in full]).mean()
+        c1 = model_ref(*[a[:, 2:4].contiguous() for a in full]).mean()
+        ((c0 + c1) / 2).backward()
+
+        for (k, p), (k2, p2) in zip(model.P.items(), model_ref.P.items()):
+            assert k == k2
+            torch.testing.assert_close(p.grad, p2.grad, rtol=1e-4, atol=1e-6)
+
+        # scalar all-reduce
+        v = dp.all_reduce_scalar(float(rank + 1), average=True)
+        assert abs(v - 1.5) < 1e-9
+        if rank == 0:
+            open(os.path.join(tmpdir, "ok"), "w").write("ok")
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ddp_grad_allreduce_gloo(tmp_path):
+    port = 29531
+    mp.spawn(_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+    assert os.path.exists(tmp_path / "ok")

@@ -1,0 +1,139 @@
+import warnings
+
+import numpy
+import pytest
+import torch
+
+from nats_amd.data.prepare import prepare_data
+from nats_amd.engine.checkpoint import (load_checkpoint, save_checkpoint,
+                                        load_options)
+from nats_amd.models.distraction import NatsModel, default_options
+from nats_amd.models.init import init_params, param_shapes
+
+
+def _batch(tiny_options, B=3, Ts=7, Tt=5, seed=0):
+    rng = numpy.random.RandomState(seed)
+    xs = [list(rng.randint(2, tiny_options["n_words"],
+                           size=rng.randint(3, Ts))) for _ in range(B)]
+    ys = [list(rng.randint(2, tiny_options["n_words"],
+                           size=rng.randint(2, Tt))) for _ in range(B)]
+    arrs = prepare_data(xs, ys)
+    return [torch.from_numpy(a) for a in arrs]
+
+
+def test_param_schema_shapes(tiny_options):
+    params = init_params(tiny_options, seed=0)
+    shapes = dict(param_shapes(tiny_options))
+    V, E = tiny_options["n_words"], tiny_options["dim_word"]
+    H, A = tiny_options["dim"], tiny_options["dim_att"]
+    assert params["Wemb"].shape == (V, E)
+    assert params["encoder_W"].shape == (E, 2 * H)
+    assert params["encoder_r_Ux"].shape == (H, H)
+    assert params["ff_state_W"].shape == (2 * H, H)
+    assert params["decoder_W_1"].shape == (2 * H, 2 * H)
+    assert params["decoder_D_wei"].shape == (1, A)
+    assert params["ff_logit_W"].shape == (E, V)
+    for k, v in params.items():
+        assert v.shape == shapes[k], k
+        assert v.dtype == numpy.float32
+
+
+def test_ortho_init():
+    from nats_amd.models.init import ortho_weight
+    W = ortho_weight(16, numpy.random.RandomState(0))
+    numpy.testing.assert_allclose(W @ W.T, numpy.eye(16), atol=1e-5)
+
+
+def test_forward_cost(tiny_options):
+    model = NatsModel(tiny_options, seed=1)
+    x, x_mask, y, y_mask = _batch(tiny_options)
+    cost = model(x, x_mask, y, y_mask)
+    assert cost.shape == (x.shape[1],)
+    assert torch.isfinite(cost).all()
+    assert (cost > 0).all()
+
+
+def test_padding_invariance(tiny_options):
+    """Extra padding timesteps beyond mask+eos must not change the cost."""
+    model = NatsModel(tiny_options, seed=1)
+    x, x_mask, y, y_mask = _batch(tiny_options, B=2)
+    cost = model(x, x_mask, y, y_mask)
+    # append 3 more all-pad rows to source
+    pad = torch.zeros(3, x.shape[1], dtype=x.dtype)
+    padm = torch.zeros(3, x.shape[1])
+    cost2 = model(torch.cat([x, pad]), torch.cat([x_mask, padm]), y, y_mask)
+    torch.testing.assert_close(cost, cost2, rtol=1e-4, atol=1e-5)
+
+
+def test_backward_all_params_get_grads(tiny_options):
+    model = NatsModel(tiny_options, seed=1)
+    x, x_mask, y, y_mask = _batch(tiny_options)
+    cost = model(x, x_mask, y, y_mask).mean()
+    cost.backward()
+    missing = [k for k, p in model.P.items() if p.grad is None]
+    assert not missing, missing
+    for k, p in model.P.items():
+        assert torch.isfinite(p.grad).all(), k
+
+
+def test_checkpoint_roundtrip(tmp_path, tiny_options):
+    model = NatsModel(tiny_options, seed=2)
+    saveto = str(tmp_path / "m.npz")
+    save_checkpoint(saveto, model.get_params(), [0.5, 0.4],
+                    options=tiny_options)
+    params, hist = load_checkpoint(saveto)
+    assert hist == [0.5, 0.4]
+    assert set(params) == set(model.get_params())
+    model2 = NatsModel(tiny_options, seed=3)
+    model2.set_params(params)
+    for k in model.P:
+        torch.testing.assert_close(model.P[k], model2.P[k])
+    opts = load_options(saveto)
+    assert opts["dim"] == tiny_options["dim"]
+
+
+def test_set_params_missing_key_warns(tiny_options):
+    model = NatsModel(tiny_options, seed=2)
+    params = model.get_params()
+    del params["decoder_D_wei"]
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        model.set_params(params)
+    assert any("decoder_D_wei" in str(x.message) for x in w)
+
+
+def test_sampler_f_init_f_next(tiny_options):
+    model = NatsModel(tiny_options, seed=4)
+    x = torch.randint(2, tiny_options["n_words"], (9, 1))
+    state, ctx = model.f_init(x)
+    H = tiny_options["dim"]
+    assert state.shape == (1, H)
+    assert ctx.shape == (9, 1, 2 * H)
+    pctx = model.project_ctx(ctx)
+    acc_ctx = torch.zeros(1, 2 * H)
+    acc_alpha = torch.zeros(1, 9)
+    y = torch.tensor([-1])
+    probs, sample, h2, alpha, ctx_t, acc_ctx, acc_alpha = model.f_next(
+        y, ctx, None, pctx, state, acc_ctx, acc_alpha)
+    assert probs.shape == (1, tiny_options["n_words"])
+    torch.testing.assert_close(probs.sum(), torch.tensor(1.0))
+    assert alpha.shape == (1, 9)
+    torch.testing.assert_close(alpha.sum(), torch.tensor(1.0))
+
+
+def test_bos_embedding_is_zero(tiny_options):
+    """y=-1 must feed a zero embedding (nats.py:827-829)."""
+    model = NatsModel(tiny_options, seed=4)
+    emb = model.embed(torch.tensor([3]))
+    assert not torch.all(emb == 0)
+    x = torch.randint(2, tiny_options["n_words"], (5, 1))
+    state, ctx = model.f_init(x)
+    pctx = model.project_ctx(ctx)
+    z = torch.zeros(1, 2 * tiny_options["dim"])
+    za = torch.zeros(1, 5)
+    out_bos = model.f_next(torch.tensor([-1]), ctx, None, pctx, state, z, za,
+                           sample_draw=False)
+    # BOS result must differ from feeding token 3
+    out_tok = model.f_next(torch.tensor([3]), ctx, None, pctx, state, z, za,
+                           sample_draw=False)
+    assert not torch.allclose(out_bos[0], out_tok[0])
