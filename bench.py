@@ -1,0 +1,168 @@
+#!/usr/bin/env python3
+"""Flagship training benchmark — train tokens/sec (whole job).
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W` runs the
+flagship training step on N GPUs of one node (launched via
+torch.distributed.run for N>1, one rank per GPU over RCCL). W untimed
+warmup steps, then exactly K timed steps bracketed by barrier +
+torch.cuda.synchronize() on both sides; rank 0 prints ONE JSON line with
+the whole-job aggregate tokens/sec (MAX elapsed over ranks).
+
+Default config is the BASELINE.json headline: CNN/DailyMail shape
+(src=800, tgt=100, vocab=30k), 1000-dim GRU, synthetic random tokens,
+random-init weights, adadelta + clip (the reference's driver config),
+bf16 compute on GPU (fp32 master weights) / fp32 on CPU.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from nats_amd.data.synthetic import synthetic_batch
+from nats_amd.engine.optim import build_optimizer
+from nats_amd.models.distraction import NatsModel, default_options
+from nats_amd.parallel.ddp import DataParallelGrads, init_distributed
+
+CONFIGS = {
+    # BASELINE.json configs[2] — the headline: CNN/DM shape, dim=1000, DP=N
+    "cnn_dm": dict(src=800, tgt=100, n_words=30000, dim=1000, dim_word=100,
+                   dim_att=100, batch=20),
+    # BASELINE.json configs[1] — LCSTS shape, 500-dim
+    "lcsts": dict(src=120, tgt=30, n_words=4000, dim=500, dim_word=100,
+                  dim_att=100, batch=20),
+    # BASELINE.json configs[4] — long-doc stress (single-layer variant)
+    "longdoc": dict(src=4000, tgt=100, n_words=30000, dim=2048, dim_word=100,
+                    dim_att=100, batch=4),
+    # tiny smoke config (CPU-capable)
+    "toy": dict(src=40, tgt=12, n_words=500, dim=64, dim_word=32, dim_att=16,
+                batch=8),
+}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--config", default="cnn_dm")
+    ap.add_argument("--batch", type=int, default=None,
+                    help="per-GPU batch override")
+    ap.add_argument("--eager", action="store_true",
+                    help="force the eager (no HIP kernels) path")
+    args = ap.parse_args()
+
+    if args.eager:
+        os.environ["NATS_AMD_FORCE_EAGER"] = "1"
+        os.environ["NATS_AMD_ALLOW_EAGER_GPU"] = "1"
+
+    use_cuda = torch.cuda.is_available()
+    cfg_name = args.config
+    if not use_cuda and cfg_name != "toy":
+        print("WARNING: no GPU — falling back to toy config", file=sys.stderr)
+        cfg_name = "toy"
+    cfg = CONFIGS[cfg_name]
+    batch = args.batch or cfg["batch"]
+
+    rank, local_rank, world = init_distributed()
+    if use_cuda:
+        device = torch.device("cuda", local_rank)
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    opts = default_options(
+        dim=cfg["dim"], dim_word=cfg["dim_word"], dim_att=cfg["dim_att"],
+        n_words=cfg["n_words"], batch_size=batch, optimizer="adadelta",
+        clip_c=100.0, maxlen=cfg["src"] + 1)
+    model = NatsModel(opts, seed=1234).to(device)
+    dp = DataParallelGrads(model.parameters())
+    dp.broadcast_params()
+    opt = build_optimizer("adadelta", list(model.P.items()),
+                          lrate=1e-4, clip_c=100.0)
+
+    rng = numpy.random.RandomState(1234 + rank)
+    x, x_mask, y, y_mask = [
+        torch.from_numpy(a).to(device) for a in synthetic_batch(
+            rng, batch, cfg["src"], cfg["tgt"], cfg["n_words"])]
+
+    amp = torch.autocast("cuda", dtype=torch.bfloat16) if use_cuda else None
+
+    def step():
+        opt.zero_grad()
+        if amp is not None:
+            with amp:
+                cost = model(x, x_mask, y, y_mask).mean()
+        else:
+            cost = model(x, x_mask, y, y_mask).mean()
+        cost.backward()
+        dp.finish()
+        opt.step()
+        return cost
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+    t0 = time.perf_counter()
+    last_cost = None
+    for _ in range(args.steps):
+        last_cost = step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks defines whole-job time
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_cuda else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    tokens_per_step = world * batch * ((cfg["src"] + 1) + (cfg["tgt"] + 1))
+    toks_per_sec = tokens_per_step * args.steps / elapsed
+    ms_per_step = 1e3 * elapsed / args.steps
+
+    if rank == 0:
+        result = {
+            "metric": "train_tokens_per_sec",
+            "value": toks_per_sec,
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "bf16" if use_cuda else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "distraction-gru-seq2seq",
+                "shape": cfg_name,
+                "dim": cfg["dim"], "dim_word": cfg["dim_word"],
+                "dim_att": cfg["dim_att"], "vocab": cfg["n_words"],
+                "src_len": cfg["src"], "tgt_len": cfg["tgt"],
+                "batch_per_gpu": batch,
+                "global_batch": batch * world,
+                "seq_len": cfg["src"],
+                "optimizer": "adadelta+clip100",
+                "parallelism": "dp%d" % world,
+                "final_cost": float(last_cost.detach()),
+            },
+        }
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,96 @@
+"""Autograd wrapper for the fused HIP GRU scan (ops/hip/gru_scan.hip).
+
+Custom BPTT: the sequential recurrences (forward h-chain, reverse dh-chain)
+run in the HIP kernels; the time-batched weight/input gradients are single
+hipBLASLt GEMMs over the stored per-step preactivation grads — the same
+split the reference's Theano graph implied (scan grads + batched dots,
+nats.py:1340).
+"""
+
+import math
+
+import torch
+import torch.nn.functional as F
+
+from . import _hip_ext
+
+JB = 16
+
+
+def _pad_to(x, rows, cols):
+    return F.pad(x, (0, cols - x.shape[1], 0, rows - x.shape[0]))
+
+
+def pack_fwd_weights(U, Ux):
+    """[ngrp*3*16, Hpad] bf16: per group g rows = [r-cols | u-cols | x-cols]
+    of output columns [g*16,(g+1)*16), transposed (row = output col)."""
+    H = Ux.shape[1]
+    ngrp = (H + JB - 1) // JB
+    rows = ngrp * JB
+    Hpad = ((H + 31) // 32) * 32
+    Ut = _pad_to(U[:, :H].t(), rows, Hpad)
+    Uu = _pad_to(U[:, H:].t(), rows, Hpad)
+    Uxt = _pad_to(Ux.t(), rows, Hpad)
+    P = torch.stack([Ut, Uu, Uxt], dim=0)            # (3, rows, Hpad)
+    P = P.view(3, ngrp, JB, Hpad).permute(1, 0, 2, 3).contiguous()
+    return P.view(ngrp * 3 * JB, Hpad).to(torch.bfloat16).contiguous()
+
+
+def pack_bwd_weights(U, Ux):
+    """[ngrp*16, K3pad] bf16: row i = [U[i, :2H] | Ux[i, :]] zero-padded."""
+    H = Ux.shape[1]
+    ngrp = (H + JB - 1) // JB
+    rows = ngrp * JB
+    K3 = 3 * H
+    K3pad = ((K3 + 31) // 32) * 32
+    cat = torch.cat([U, Ux], dim=1)                   # (H, 3H)
+    return _pad_to(cat, rows, K3pad).to(torch.bfloat16).contiguous()
+
+
+class GRUScanFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, xg, xc, mask, U, Ux, h0):
+        ext = _hip_ext()
+        xg = xg.to(torch.bfloat16).contiguous()
+        xc = xc.to(torch.bfloat16).contiguous()
+        Upk = pack_fwd_weights(U, Ux)
+        h_all, saved = ext.gru_scan_fwd(xg, xc, mask, Upk, h0)
+        ctx.save_for_backward(xc, h_all, saved, U, Ux,
+                              mask if mask is not None else torch.empty(0),
+                              h0 if h0 is not None else torch.empty(0))
+        return h_all
+
+    @staticmethod
+    def backward(ctx, dh_out):
+        ext = _hip_ext()
+        xc, h_all, saved, U, Ux, mask, h0 = ctx.saved_tensors
+        mask = mask if mask.numel() else None
+        h0v = h0 if h0.numel() else None
+        T, B, H = h_all.shape
+        Ubwd = pack_bwd_weights(U, Ux)
+        dpre_all, dh0 = ext.gru_scan_bwd(
+            dh_out.contiguous().float(), h_all, saved, xc, mask, Ubwd, h0v)
+        dxg = dpre_all[..., :2 * H]
+        dxc = dpre_all[..., 2 * H:3 * H]
+        # time-batched weight grads: dU = hprev^T @ [dpr|dpu],
+        # dUx = hprev^T @ dpxl (one GEMM each over all T*B positions)
+        if h0v is not None:
+            h_prev = torch.cat([h0v.unsqueeze(0).float(), h_all[:-1]], dim=0)
+        else:
+            h_prev = torch.cat([torch.zeros_like(h_all[:1]), h_all[:-1]],
+                               dim=0)
+        flat_h = h_prev.reshape(T * B, H).to(torch.bfloat16)
+        dU = (flat_h.t() @ dpre_all[..., :2 * H].reshape(T * B, 2 * H)).float()
+        dUx = (flat_h.t() @ dpre_all[..., 3 * H:].reshape(T * B, H)).float()
+        need_h0 = ctx.needs_input_grad[5]
+        return (dxg.to(U.dtype) if U.dtype != torch.bfloat16 else dxg,
+                dxc.to(U.dtype) if U.dtype != torch.bfloat16 else dxc,
+                None,
+                dU.to(U.dtype), dUx.to(Ux.dtype),
+                dh0 if need_h0 else None)
+
+
+def gru_scan_hip(x_gates, x_cand, mask, U, Ux, h0=None):
+    if mask is not None:
+        mask = mask.float().contiguous()
+    return GRUScanFn.apply(x_gates, x_cand, mask, U, Ux, h0)

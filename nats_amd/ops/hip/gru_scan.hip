@@ -1,0 +1,324 @@
+// Fused GRU scan (forward + backward) for CDNA4 / gfx950.
+//
+// Replaces the reference's per-timestep Theano scan body (gru_layer
+// _step_slice, nats.py:336-372; kernel rows K3 in SURVEY §2.4): the three
+// recurrent GEMMs h@U_r, h@U_u, h@Ux plus the gate nonlinearities and the
+// mask blend are ONE kernel launch per timestep, tiled for MFMA
+// (16x16x32 bf16, fp32 accumulate), with the per-step preactivations
+// exchanged through LDS between the MFMA phase and the pointwise phase.
+//
+// Weight packing (done by the python wrapper, ops/gru.py):
+//   Upk  : [ngrp*3*16, Hpad] bf16, row = output slot. Workgroup `wg` owns
+//          output columns j in [wg*16, wg*16+16); its rows are laid out
+//          [r-cols | u-cols | cand-cols] so B-fragments are contiguous
+//          16-byte loads (see common.h layout note).
+//   Hpad : H rounded up to 32; the padded K-region of every operand is
+//          zero-filled so MFMA accumulates exact zeros there.
+//
+// State: h_all (T,B,H) fp32 is the master hidden sequence; h_bf
+// ([32][Hpad] bf16, zero-padded rows/cols) mirrors the CURRENT h for the
+// next step's A-fragments.
+//
+// The backward scan (reverse-time) mirrors the split: a pointwise kernel
+// turns dh_t into gate-preactivation grads (stored to dpre_all for the
+// big time-batched dW/dU GEMMs done by hipBLASLt from python), then an
+// MFMA kernel computes the recurrent term dh_{t-1} = [dpr|dpu|dpxl] @
+// [U|Ux]^T + passthrough. BPTT semantics = tensor.grad through the scan
+// (nats.py:1340).
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int JB = 16;  // output columns per workgroup
+
+// ---------------- forward step ----------------
+// block: 384 threads = 6 waves; wave w -> (m = w/3, g = w%3) computes the
+// 16x16 tile rows [16m,16m+16) of matrix g (0=r,1=u,2=cand).
+__global__ __launch_bounds__(384) void gru_step_fwd(
+    const bf16_t* __restrict__ h_bf,   // [32][Hpad] current h (bf16)
+    const float* __restrict__ h_prev,  // [B][H] fp32 (= h_all[t-1] or h0)
+    const bf16_t* __restrict__ Upk,    // [ngrp*3*16][Hpad]
+    const bf16_t* __restrict__ xg_t,   // [B][2H] x@W+b at step t
+    const bf16_t* __restrict__ xc_t,   // [B][H]  x@Wx+bx at step t
+    const float* __restrict__ mask_t,  // [B] or nullptr
+    float* __restrict__ h_out,         // [B][H] fp32 (h_all[t])
+    bf16_t* __restrict__ h_bf_out,     // [32][Hpad] bf16 (next step's A)
+    bf16_t* __restrict__ saved_t,      // [B][3H] (r,u,pxl) at step t
+    int B, int H, int Hpad) {
+  __shared__ float pre[3][32][JB + 1];
+
+  const int wg = blockIdx.x;
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int m = wave / 3;
+  const int g = wave % 3;
+  const int j0 = wg * JB;
+
+  // MFMA phase
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const bf16_t* brow = Upk + (long)(wg * 3 + g) * JB * Hpad;
+  for (int k = 0; k < Hpad; k += 32) {
+    bf16x8 a = frag_a_rowmajor(h_bf, 16 * m, Hpad, k);
+    bf16x8 b = frag_bt_rowmajor(brow, 0, Hpad, k);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  {
+    const int lane = threadIdx.x & (NATS_WAVE - 1);
+    const int col = lane & 15;
+    const int rbase = 16 * m + (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) pre[g][rbase + i][col] = acc[i];
+  }
+  __syncthreads();
+
+  // pointwise phase over B x JB outputs
+  for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+    const int b = idx / JB;
+    const int c = idx % JB;
+    const int j = j0 + c;
+    if (j >= H) continue;
+    const float hp = h_prev[(long)b * H + j];
+    const float pr = pre[0][b][c] + (float)xg_t[(long)b * 2 * H + j];
+    const float pu = pre[1][b][c] + (float)xg_t[(long)b * 2 * H + H + j];
+    const float px = pre[2][b][c];
+    const float r = nats_sigmoid(pr);
+    const float u = nats_sigmoid(pu);
+    const float hbar = tanhf(px * r + (float)xc_t[(long)b * H + j]);
+    float hnew = u * hp + (1.f - u) * hbar;
+    if (mask_t != nullptr) {
+      const float mm = mask_t[b];
+      hnew = mm * hnew + (1.f - mm) * hp;
+    }
+    h_out[(long)b * H + j] = hnew;
+    h_bf_out[(long)b * Hpad + j] = (bf16_t)hnew;
+    saved_t[(long)b * 3 * H + j] = (bf16_t)r;
+    saved_t[(long)b * 3 * H + H + j] = (bf16_t)u;
+    saved_t[(long)b * 3 * H + 2 * H + j] = (bf16_t)px;
+  }
+}
+
+// ---------------- backward pointwise ----------------
+// grid-stride over B*H. Produces the packed step K-operand dstep
+// ([32][K3pad] bf16, layout [dpr | dpu | dpxl]) for the recurrent GEMM,
+// the passthrough term ddirect, and dpre_all[t] ([B][4H] bf16:
+// [dpr|dpu|dpx|dpxl]) for the time-batched weight/input GEMMs.
+__global__ void gru_step_bwd_pointwise(
+    const float* __restrict__ dh_buf,   // [B][H] recurrent dh (from t+1)
+    const float* __restrict__ dh_out_t, // [B][H] upstream grad at t
+    const bf16_t* __restrict__ saved_t, // [B][3H] (r,u,pxl)
+    const bf16_t* __restrict__ xc_t,    // [B][H]
+    const float* __restrict__ h_prev,   // [B][H]
+    const float* __restrict__ mask_t,   // [B] or nullptr
+    bf16_t* __restrict__ dstep,         // [32][K3pad]
+    float* __restrict__ ddirect,        // [B][H]
+    bf16_t* __restrict__ dpre_t,        // [B][4H]
+    int B, int H, int K3pad) {
+  const long total = (long)B * H;
+  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int b = idx / H;
+    const int j = idx % H;
+    const float dh = dh_buf[idx] + dh_out_t[idx];
+    const float r = (float)saved_t[(long)b * 3 * H + j];
+    const float u = (float)saved_t[(long)b * 3 * H + H + j];
+    const float px = (float)saved_t[(long)b * 3 * H + 2 * H + j];
+    const float hbar = tanhf(px * r + (float)xc_t[idx]);
+    const float hp = h_prev[idx];
+    const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
+    const float du = dh * mm * (hp - hbar);
+    const float dhbar = dh * mm * (1.f - u);
+    const float dpx = dhbar * (1.f - hbar * hbar);  // = dxc
+    const float dpxl = dpx * r;
+    const float dr = dpx * px;
+    const float dpr = dr * r * (1.f - r);
+    const float dpu = du * u * (1.f - u);
+    ddirect[idx] = dh * (mm * u + (1.f - mm));
+    dstep[(long)b * K3pad + j] = (bf16_t)dpr;
+    dstep[(long)b * K3pad + H + j] = (bf16_t)dpu;
+    dstep[(long)b * K3pad + 2 * H + j] = (bf16_t)dpxl;
+    dpre_t[(long)b * 4 * H + j] = (bf16_t)dpr;
+    dpre_t[(long)b * 4 * H + H + j] = (bf16_t)dpu;
+    dpre_t[(long)b * 4 * H + 2 * H + j] = (bf16_t)dpx;
+    dpre_t[(long)b * 4 * H + 3 * H + j] = (bf16_t)dpxl;
+  }
+}
+
+// ---------------- backward recurrent GEMM ----------------
+// dh_prev[b, i] = ddirect[b, i] + sum_k dstep[b, k] * Ubwd[i, k]
+// Ubwd = [U | Ux] rows (H x 3H, padded to K3pad). Workgroup owns 16
+// i-columns; 6 waves = 2 row-tiles x 3 K-splits, LDS reduce.
+__global__ __launch_bounds__(384) void gru_step_bwd_gemm(
+    const bf16_t* __restrict__ dstep,  // [32][K3pad]
+    const bf16_t* __restrict__ Ubwd,   // [H][K3pad] (row i contiguous)
+    const float* __restrict__ ddirect, // [B][H]
+    float* __restrict__ dh_buf,        // [B][H] output (overwritten)
+    int B, int H, int K3pad) {
+  __shared__ float part[3][32][JB + 1];
+
+  const int wg = blockIdx.x;
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int m = wave / 3;
+  const int ks = wave % 3;
+  const int i0 = wg * JB;
+
+  // K ranges per split: thirds of K3pad rounded to 32
+  const int kchunk = ((K3pad / 3 + 31) / 32) * 32;
+  const int kbeg = ks * kchunk;
+  const int kend = min(K3pad, (ks + 1) * kchunk);
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int k = kbeg; k < kend; k += 32) {
+    bf16x8 a = frag_a_rowmajor(dstep, 16 * m, K3pad, k);
+    bf16x8 b = frag_bt_rowmajor(Ubwd, i0, K3pad, k);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  {
+    const int lane = threadIdx.x & (NATS_WAVE - 1);
+    const int col = lane & 15;
+    const int rbase = 16 * m + (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) part[ks][rbase + i][col] = acc[i];
+  }
+  __syncthreads();
+
+  for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+    const int b = idx / JB;
+    const int c = idx % JB;
+    const int i = i0 + c;
+    if (i >= H) continue;
+    dh_buf[(long)b * H + i] = ddirect[(long)b * H + i] + part[0][b][c] +
+                              part[1][b][c] + part[2][b][c];
+  }
+}
+
+}  // namespace
+
+// ---------------- host drivers ----------------
+
+static inline int cdiv(int a, int b) { return (a + b - 1) / b; }
+
+// Forward scan over T steps. Returns (h_all fp32 (T,B,H), saved bf16
+// (T,B,3H)).
+std::vector<torch::Tensor> gru_scan_fwd(torch::Tensor xg, torch::Tensor xc,
+                                        c10::optional<torch::Tensor> mask,
+                                        torch::Tensor Upk,
+                                        c10::optional<torch::Tensor> h0) {
+  TORCH_CHECK(xg.is_cuda() && xg.dtype() == torch::kBFloat16 &&
+              xg.is_contiguous());
+  TORCH_CHECK(xc.is_cuda() && xc.dtype() == torch::kBFloat16 &&
+              xc.is_contiguous());
+  const int T = xg.size(0), B = xg.size(1), H = xc.size(2);
+  TORCH_CHECK(xg.size(2) == 2 * H);
+  TORCH_CHECK(B <= 32, "gru_scan: batch per step must be <= 32 (got ", B,
+              "); use more DP ranks or smaller micro-batches");
+  const int Hpad = Upk.size(1);
+  const int ngrp = cdiv(H, JB);
+  TORCH_CHECK(Upk.size(0) == ngrp * 3 * JB && Upk.is_contiguous());
+
+  auto optsF = xg.options().dtype(torch::kFloat32);
+  auto optsB = xg.options();
+  auto h_all = torch::empty({T, B, H}, optsF);
+  auto saved = torch::empty({T, B, 3 * H}, optsB);
+  // ping-pong bf16 h buffers: step t READS buf[t%2] (state t-1) and WRITES
+  // buf[(t+1)%2] (state t) — a single buffer would race across workgroups
+  // (every WG reads ALL columns while owning only its own 16).
+  auto h_bf = torch::zeros({2, 32, Hpad}, optsB);
+  torch::Tensor hprev0;
+  if (h0.has_value()) {
+    hprev0 = h0->contiguous().to(torch::kFloat32);
+    h_bf[0].slice(0, 0, B).slice(1, 0, H).copy_(hprev0.to(torch::kBFloat16));
+  } else {
+    hprev0 = torch::zeros({B, H}, optsF);
+  }
+  const float* mask_p = nullptr;
+  torch::Tensor mask_c;
+  if (mask.has_value()) {
+    mask_c = mask->contiguous().to(torch::kFloat32);
+    mask_p = mask_c.data_ptr<float>();
+  }
+
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  const bf16_t* xg_p = (const bf16_t*)xg.data_ptr();
+  const bf16_t* xc_p = (const bf16_t*)xc.data_ptr();
+  float* h_all_p = h_all.data_ptr<float>();
+  bf16_t* saved_p = (bf16_t*)saved.data_ptr();
+  bf16_t* hbf_p = (bf16_t*)h_bf.data_ptr();
+  const bf16_t* Upk_p = (const bf16_t*)Upk.data_ptr();
+  const float* h0_p = hprev0.data_ptr<float>();
+
+  const long hbuf_stride = (long)32 * Hpad;
+  for (int t = 0; t < T; ++t) {
+    const float* hprev = (t == 0) ? h0_p : (h_all_p + (long)(t - 1) * B * H);
+    hipLaunchKernelGGL(gru_step_fwd, dim3(ngrp), dim3(384), 0, stream,
+                       hbf_p + (t % 2) * hbuf_stride, hprev, Upk_p,
+                       xg_p + (long)t * B * 2 * H, xc_p + (long)t * B * H,
+                       mask_p ? mask_p + (long)t * B : nullptr,
+                       h_all_p + (long)t * B * H,
+                       hbf_p + ((t + 1) % 2) * hbuf_stride,
+                       saved_p + (long)t * B * 3 * H, B, H, Hpad);
+  }
+  HIP_CHECK(hipGetLastError());
+  return {h_all, saved};
+}
+
+// Backward scan (reverse time). Returns (dpre_all bf16 (T,B,4H), dh0 fp32
+// (B,H)).
+std::vector<torch::Tensor> gru_scan_bwd(torch::Tensor dh_out,
+                                        torch::Tensor h_all,
+                                        torch::Tensor saved, torch::Tensor xc,
+                                        c10::optional<torch::Tensor> mask,
+                                        torch::Tensor Ubwd,
+                                        c10::optional<torch::Tensor> h0) {
+  TORCH_CHECK(dh_out.is_cuda() && dh_out.dtype() == torch::kFloat32 &&
+              dh_out.is_contiguous());
+  const int T = dh_out.size(0), B = dh_out.size(1), H = dh_out.size(2);
+  const int K3pad = Ubwd.size(1);
+  const int ngrp = cdiv(H, JB);
+  TORCH_CHECK(Ubwd.size(0) == ngrp * JB && Ubwd.is_contiguous());
+  TORCH_CHECK(saved.dtype() == torch::kBFloat16 && xc.dtype() == torch::kBFloat16);
+
+  auto optsF = dh_out.options();
+  auto optsB = dh_out.options().dtype(torch::kBFloat16);
+  auto dpre_all = torch::empty({T, B, 4 * H}, optsB);
+  auto dh_buf = torch::zeros({B, H}, optsF);
+  auto ddirect = torch::empty({B, H}, optsF);
+  auto dstep = torch::zeros({32, K3pad}, optsB);
+  torch::Tensor hprev0 = h0.has_value()
+                             ? h0->contiguous().to(torch::kFloat32)
+                             : torch::zeros({B, H}, optsF);
+  const float* mask_p = nullptr;
+  torch::Tensor mask_c;
+  if (mask.has_value()) {
+    mask_c = mask->contiguous().to(torch::kFloat32);
+    mask_p = mask_c.data_ptr<float>();
+  }
+
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  const float* dh_out_p = dh_out.data_ptr<float>();
+  const float* h_all_p = h_all.data_ptr<float>();
+  const bf16_t* saved_p = (const bf16_t*)saved.data_ptr();
+  const bf16_t* xc_p = (const bf16_t*)xc.data_ptr();
+  bf16_t* dpre_p = (bf16_t*)dpre_all.data_ptr();
+  float* dh_buf_p = dh_buf.data_ptr<float>();
+  float* ddirect_p = ddirect.data_ptr<float>();
+  bf16_t* dstep_p = (bf16_t*)dstep.data_ptr();
+  const bf16_t* Ubwd_p = (const bf16_t*)Ubwd.data_ptr();
+
+  const int pw_blocks = min(1024, (int)((((long)B * H) + 255) / 256));
+  for (int t = T - 1; t >= 0; --t) {
+    const float* hprev = (t == 0) ? hprev0.data_ptr<float>()
+                                  : (h_all_p + (long)(t - 1) * B * H);
+    hipLaunchKernelGGL(gru_step_bwd_pointwise, dim3(pw_blocks), dim3(256), 0,
+                       stream, dh_buf_p, dh_out_p + (long)t * B * H,
+                       saved_p + (long)t * B * 3 * H, xc_p + (long)t * B * H,
+                       hprev, mask_p ? mask_p + (long)t * B : nullptr, dstep_p,
+                       ddirect_p, dpre_p + (long)t * B * 4 * H, B, H, K3pad);
+    hipLaunchKernelGGL(gru_step_bwd_gemm, dim3(ngrp), dim3(384), 0, stream,
+                       dstep_p, Ubwd_p, ddirect_p, dh_buf_p, B, H, K3pad);
+  }
+  HIP_CHECK(hipGetLastError());
+  return {dpre_all, dh_buf};
+}
