@@ -37,20 +37,19 @@ def clip_grads_global_norm(params, clip_c):
     clip_c/sqrt(g2) iff g2 > clip_c**2. Returns sqrt(g2) (pre-clip norm).
     """
     grads = [p.grad for p in params if p.grad is not None]
-    if clip_c is None or clip_c <= 0 or not grads:
-        if not grads:
-            return 0.0
-        g2 = sum(float(g.float().pow(2).sum()) for g in grads)
-        return math.sqrt(g2)
-    g2 = torch.zeros((), dtype=torch.float32, device=grads[0].device)
-    for g in grads:
-        g2 = g2 + g.float().pow(2).sum()
+    if not grads:
+        return 0.0
+    # fused multi-tensor path (one norm kernel set + one scale pass)
+    norms = torch._foreach_norm(grads)
+    g2 = (torch.stack(norms).float() ** 2).sum()
+    if clip_c is None or clip_c <= 0:
+        return torch.sqrt(g2)
     scale = torch.where(g2 > clip_c * clip_c,
                         clip_c / torch.sqrt(g2),
                         torch.ones_like(g2))
-    for g in grads:
-        g.mul_(scale.to(g.dtype))
-    return float(torch.sqrt(g2))
+    torch._foreach_mul_(grads, scale)
+    # returned lazily (no device sync); float() it only when logging
+    return torch.sqrt(g2)
 
 
 class _RefOptimizer:

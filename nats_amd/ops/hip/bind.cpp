@@ -17,6 +17,27 @@ std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
 torch::Tensor softmax_ce_bwd(torch::Tensor logits, torch::Tensor targets,
                              torch::Tensor stats, torch::Tensor dnll);
 torch::Tensor mfma_gemm_bt(torch::Tensor A, torch::Tensor Bt);
+std::vector<torch::Tensor> cond_gru_fwd(
+    torch::Tensor yg, torch::Tensor yc, c10::optional<torch::Tensor> mask,
+    torch::Tensor init_state, torch::Tensor ctx_bf,
+    c10::optional<torch::Tensor> ctx_mask, torch::Tensor pctx,
+    torch::Tensor Upk2, torch::Tensor W1pk, torch::Tensor WattPk,
+    torch::Tensor b1, torch::Tensor bx1, torch::Tensor Uatt, double catt,
+    torch::Tensor Dwei, torch::Tensor Wcon, torch::Tensor Ucon,
+    c10::optional<torch::Tensor> accC0, c10::optional<torch::Tensor> accA0);
+std::vector<torch::Tensor> cond_gru_bwd(
+    torch::Tensor dh2_all, c10::optional<torch::Tensor> dctxs_all,
+    c10::optional<torch::Tensor> dalphas_all,
+    c10::optional<torch::Tensor> daccC_f, c10::optional<torch::Tensor> daccA_f,
+    torch::Tensor yc, torch::Tensor h1_all, torch::Tensor h2_all,
+    torch::Tensor ctxs_all, torch::Tensor alphas_all, torch::Tensor saved2,
+    torch::Tensor saved1, torch::Tensor pstate_all, torch::Tensor ctxpre_all,
+    torch::Tensor accA_used, torch::Tensor accC_used, torch::Tensor ctx_bf,
+    torch::Tensor pctx, torch::Tensor init_state,
+    c10::optional<torch::Tensor> mask, torch::Tensor U1cat,
+    torch::Tensor W1cat, torch::Tensor U2cat, torch::Tensor WattB,
+    torch::Tensor bx1, torch::Tensor Dwei, torch::Tensor Uatt,
+    torch::Tensor Ucon, torch::Tensor Wcon);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gru_scan_fwd", &gru_scan_fwd, "fused GRU scan forward");
@@ -24,4 +45,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_ce_fwd", &softmax_ce_fwd, "fused softmax+CE forward");
   m.def("softmax_ce_bwd", &softmax_ce_bwd, "fused softmax+CE backward");
   m.def("mfma_gemm_bt", &mfma_gemm_bt, "MFMA layout self-test GEMM");
+  m.def("cond_gru_fwd", &cond_gru_fwd, "fused cond-GRU decoder forward");
+  m.def("cond_gru_bwd", &cond_gru_bwd, "fused cond-GRU decoder backward");
 }

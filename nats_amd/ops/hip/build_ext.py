@@ -14,6 +14,7 @@ BUILD_DIR = os.path.join(HERE, "build")
 SOURCES = [
     os.path.join(HERE, "bind.cpp"),
     os.path.join(HERE, "gru_scan.hip"),
+    os.path.join(HERE, "cond_gru.hip"),
     os.path.join(HERE, "softmax_ce.hip"),
     os.path.join(HERE, "mfma_test.hip"),
 ]

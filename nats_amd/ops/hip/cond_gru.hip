@@ -1,0 +1,740 @@
+// Fused conditional-GRU decoder (training scan + one-step) for gfx950.
+//
+// Implements the reference's gru_cond_layer step (nats.py:498-572; kernel
+// rows K10-K16 in SURVEY §2.4) as five fused kernels per timestep driven
+// by a C++ time loop (no python in the scan):
+//   1. GRU_2          — reuses nats_gru_step_fwd (identical cell math)
+//   2. pstate GEMM    — h1 @ W_att (small MFMA GEMM)
+//   3. attention      — e-scores + distraction-over-weights + masked
+//                       softmax + acc_alpha update, one WG per batch row
+//   4. context        — weighted sum over source + distraction gate over
+//                       content vectors + acc_ctx update
+//   5. GRU_1          — 4-output-group MFMA ([h1|ctx] packed operand) +
+//                       gates/mask pointwise
+// The backward pass is the exact reverse chain with the running-sum
+// accumulators' (acc_ctx/acc_alpha) gradients threaded backwards
+// (SURVEY §7 "hard parts" (ii)); weight gradients that factor over time
+// (dU_1, dW_1, dW_att, dWc_att, dU_con, ...) are computed as single
+// time-batched GEMMs in python from the per-step buffers saved here.
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include <algorithm>
+
+#include "common.h"
+#include "gru_kernels.h"
+
+namespace {
+
+constexpr int JB = 16;
+
+// ---------------- small GEMM: pstate = h1 @ W_att ----------------
+// A = [32][ldK] bf16 (zero-padded), Bt = [Npad][ldK] bf16; C f32 [B][N].
+__global__ void cond_small_gemm_bt(const bf16_t* __restrict__ A,
+                                   const bf16_t* __restrict__ Bt,
+                                   float* __restrict__ C, int B, int N,
+                                   int ldK, int Kpad) {
+  const int m0 = blockIdx.x * 16;
+  const int n0 = blockIdx.y * 16;
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int k = 0; k < Kpad; k += 32) {
+    bf16x8 a = frag_a_rowmajor(A, m0, ldK, k);
+    bf16x8 b = frag_bt_rowmajor(Bt, n0, ldK, k);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  const int lane = threadIdx.x & (NATS_WAVE - 1);
+  const int col = n0 + (lane & 15);
+  const int rbase = m0 + (lane >> 4) * 4;
+  if (col >= N) return;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int row = rbase + i;
+    if (row < B) C[(long)row * N + col] = acc[i];
+  }
+}
+
+// ---------------- attention scores + softmax (one WG per b) ----------
+__global__ __launch_bounds__(256) void cond_attn_fwd(
+    const float* __restrict__ pctx,      // [Ts][B][A]
+    const float* __restrict__ pstate_t,  // [B][A]
+    float* __restrict__ accA,            // [B][Ts] in/out
+    float* __restrict__ accA_used_t,     // [B][Ts] out (pre-update copy)
+    const float* __restrict__ Dwei,      // [A]
+    const float* __restrict__ Uatt,      // [A]
+    float catt,
+    const float* __restrict__ ctx_mask,  // [Ts][B] or null
+    const float* __restrict__ mask_t,    // [B] or null
+    float* __restrict__ e_buf,           // [Ts][B] scratch
+    float* __restrict__ alphas_t,        // [B][Ts] out
+    int B, int Ts, int A) {
+  const int b = blockIdx.x;
+  __shared__ float red[256 / NATS_WAVE];
+  __shared__ float bcast;
+
+  // pass 1: scores + running max
+  float lmax = -INFINITY;
+  for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
+    const float accAu = accA[(long)b * Ts + s];
+    accA_used_t[(long)b * Ts + s] = accAu;
+    const float* prow = pctx + ((long)s * B + b) * A;
+    float e = catt;
+    for (int i = 0; i < A; ++i) {
+      e += tanhf(prow[i] + pstate_t[(long)b * A + i] + accAu * Dwei[i]) *
+           Uatt[i];
+    }
+    e_buf[(long)s * B + b] = e;
+    lmax = fmaxf(lmax, e);
+  }
+#pragma unroll
+  for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
+    lmax = fmaxf(lmax, __shfl_down(lmax, off));
+  if ((threadIdx.x & (NATS_WAVE - 1)) == 0)
+    red[threadIdx.x / NATS_WAVE] = lmax;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float M = red[0];
+    for (int w = 1; w < (int)blockDim.x / NATS_WAVE; ++w)
+      M = fmaxf(M, red[w]);
+    bcast = M;
+  }
+  __syncthreads();
+  const float M = bcast;
+
+  // pass 2: exp * mask + sum
+  float lsum = 0.f;
+  for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
+    float a = __expf(e_buf[(long)s * B + b] - M);
+    if (ctx_mask != nullptr) a *= ctx_mask[(long)s * B + b];
+    e_buf[(long)s * B + b] = a;
+    lsum += a;
+  }
+#pragma unroll
+  for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
+    lsum += __shfl_down(lsum, off);
+  if ((threadIdx.x & (NATS_WAVE - 1)) == 0)
+    red[threadIdx.x / NATS_WAVE] = lsum;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float S = 0.f;
+    for (int w = 0; w < (int)blockDim.x / NATS_WAVE; ++w) S += red[w];
+    bcast = S;
+  }
+  __syncthreads();
+  const float inv = 1.f / bcast;
+
+  // pass 3: normalise + acc_alpha update (nats.py:540, 570)
+  const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
+  for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
+    const float al = e_buf[(long)s * B + b] * inv;
+    alphas_t[(long)b * Ts + s] = al;
+    accA[(long)b * Ts + s] += mm * al;
+  }
+}
+
+// ---------------- weighted context + distraction gate ----------------
+__global__ void cond_attn_ctx_fwd(
+    const bf16_t* __restrict__ ctx_bf,   // [Ts][B][C]
+    const float* __restrict__ alphas_t,  // [B][Ts]
+    const float* __restrict__ Ucon,      // [C]
+    const float* __restrict__ Wcon,      // [C]
+    float* __restrict__ accC,            // [B][C] in/out
+    bf16_t* __restrict__ accC_used_t,    // [B][C]
+    bf16_t* __restrict__ ctxpre_t,       // [B][C] (pre-gate sum)
+    float* __restrict__ ctxs_t,          // [B][C] (gated output)
+    bf16_t* __restrict__ hc_bf,          // [32][K1] GRU_1 operand
+    int ctx_off,                         // = Hpad (ctx block offset in K1)
+    int ldK1, const float* __restrict__ mask_t, int B, int Ts, int C) {
+  const int b = blockIdx.x;
+  const int c = blockIdx.y * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float sum = 0.f;
+  for (int s = 0; s < Ts; ++s) {
+    sum += (float)ctx_bf[((long)s * B + b) * C + c] *
+           alphas_t[(long)b * Ts + s];
+  }
+  ctxpre_t[(long)b * C + c] = (bf16_t)sum;
+  const float accCv = accC[(long)b * C + c];
+  accC_used_t[(long)b * C + c] = (bf16_t)accCv;
+  // distraction over input content vectors (nats.py:545-546)
+  const float g = tanhf(Ucon[c] * sum + accCv * Wcon[c]);
+  ctxs_t[(long)b * C + c] = g;
+  hc_bf[(long)b * ldK1 + ctx_off + c] = (bf16_t)g;
+  const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
+  accC[(long)b * C + c] = accCv + mm * g;
+}
+
+// ---------------- GRU_1 forward ----------------
+// 8 waves: wave w -> (m = w/4, g = w%4) with output groups
+// g0 = r2 (K = h1|ctx), g1 = u2 (K = h1|ctx), g2 = pxa (h1@Ux_1),
+// g3 = pxb (ctx@Wx_1) — K-selectivity comes from zero blocks in W1pk.
+__global__ __launch_bounds__(512) void cond_gru1_step_fwd(
+    const bf16_t* __restrict__ hc_bf,  // [32][K1] = [h1 | ctx_t] bf16
+    const float* __restrict__ h1_t,    // [B][H] fp32
+    const bf16_t* __restrict__ W1pk,   // [ngrp*4*16][K1]
+    const float* __restrict__ b1,      // [2H]
+    const float* __restrict__ bx1,     // [H]
+    const float* __restrict__ mask_t,  // [B] or null
+    float* __restrict__ h2_t,          // [B][H] out
+    bf16_t* __restrict__ h2bf_out,     // [32][Hpad] out
+    int ld_h2bf,
+    bf16_t* __restrict__ saved1_t,     // [B][4H] (r2,u2,pxa,hbar)
+    int B, int H, int K1) {
+  __shared__ float pre[4][32][JB + 1];
+
+  const int wg = blockIdx.x;
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int m = wave / 4;
+  const int g = wave % 4;
+  const int j0 = wg * JB;
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const bf16_t* brow = W1pk + (long)(wg * 4 + g) * JB * K1;
+  for (int k = 0; k < K1; k += 32) {
+    bf16x8 a = frag_a_rowmajor(hc_bf, 16 * m, K1, k);
+    bf16x8 b = frag_bt_rowmajor(brow, 0, K1, k);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  {
+    const int lane = threadIdx.x & (NATS_WAVE - 1);
+    const int col = lane & 15;
+    const int rbase = 16 * m + (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) pre[g][rbase + i][col] = acc[i];
+  }
+  __syncthreads();
+
+  for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+    const int b = idx / JB;
+    const int c = idx % JB;
+    const int j = j0 + c;
+    if (j >= H) continue;
+    const float r2 = nats_sigmoid(pre[0][b][c] + b1[j]);
+    const float u2 = nats_sigmoid(pre[1][b][c] + b1[H + j]);
+    const float pxa = pre[2][b][c];
+    const float pxb = pre[3][b][c];
+    const float hbar = tanhf((pxa + bx1[j]) * r2 + pxb);
+    const float h1v = h1_t[(long)b * H + j];
+    float h2 = u2 * h1v + (1.f - u2) * hbar;
+    if (mask_t != nullptr) {
+      const float mm = mask_t[b];
+      h2 = mm * h2 + (1.f - mm) * h1v;
+    }
+    h2_t[(long)b * H + j] = h2;
+    h2bf_out[(long)b * ld_h2bf + j] = (bf16_t)h2;
+    saved1_t[(long)b * 4 * H + j] = (bf16_t)r2;
+    saved1_t[(long)b * 4 * H + H + j] = (bf16_t)u2;
+    saved1_t[(long)b * 4 * H + 2 * H + j] = (bf16_t)pxa;
+    saved1_t[(long)b * 4 * H + 3 * H + j] = (bf16_t)hbar;
+  }
+}
+
+// ---------------- backward kernels ----------------
+
+__global__ void cond_gru1_bwd_pointwise(
+    const float* __restrict__ dh_carry,   // [B][H]
+    const float* __restrict__ dh2_all_t,  // [B][H] or null
+    const bf16_t* __restrict__ saved1_t,  // [B][4H]
+    const float* __restrict__ h1_all_t,   // [B][H]
+    const float* __restrict__ bx1,        // [H]
+    const float* __restrict__ mask_t,     // [B] or null
+    bf16_t* __restrict__ dstep1,          // [32][K3H] [dpr2|dpu2|dpxa_lin]
+    bf16_t* __restrict__ dstepC,          // [32][K3H] [dpr2|dpu2|dpx2]
+    int ldK3,
+    float* __restrict__ ddirect_h1,       // [B][H]
+    bf16_t* __restrict__ dpre1_t,         // [B][4H]
+    int B, int H) {
+  const long total = (long)B * H;
+  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int b = idx / H;
+    const int j = idx % H;
+    float dh2 = dh_carry[idx];
+    if (dh2_all_t != nullptr) dh2 += dh2_all_t[idx];
+    const float r2 = (float)saved1_t[(long)b * 4 * H + j];
+    const float u2 = (float)saved1_t[(long)b * 4 * H + H + j];
+    const float pxa = (float)saved1_t[(long)b * 4 * H + 2 * H + j];
+    const float hbar = (float)saved1_t[(long)b * 4 * H + 3 * H + j];
+    const float h1v = h1_all_t[idx];
+    const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
+    const float du2 = dh2 * mm * (h1v - hbar);
+    const float dhbar = dh2 * mm * (1.f - u2);
+    const float dpx2 = dhbar * (1.f - hbar * hbar);
+    const float dpxa_lin = dpx2 * r2;
+    const float dr2 = dpx2 * (pxa + bx1[j]);
+    const float dpr2 = dr2 * r2 * (1.f - r2);
+    const float dpu2 = du2 * u2 * (1.f - u2);
+    ddirect_h1[idx] = dh2 * (mm * u2 + (1.f - mm));
+    dstep1[(long)b * ldK3 + j] = (bf16_t)dpr2;
+    dstep1[(long)b * ldK3 + H + j] = (bf16_t)dpu2;
+    dstep1[(long)b * ldK3 + 2 * H + j] = (bf16_t)dpxa_lin;
+    dstepC[(long)b * ldK3 + j] = (bf16_t)dpr2;
+    dstepC[(long)b * ldK3 + H + j] = (bf16_t)dpu2;
+    dstepC[(long)b * ldK3 + 2 * H + j] = (bf16_t)dpx2;
+    dpre1_t[(long)b * 4 * H + j] = (bf16_t)dpr2;
+    dpre1_t[(long)b * 4 * H + H + j] = (bf16_t)dpu2;
+    dpre1_t[(long)b * 4 * H + 2 * H + j] = (bf16_t)dpx2;
+    dpre1_t[(long)b * 4 * H + 3 * H + j] = (bf16_t)dpxa_lin;
+  }
+}
+
+__global__ void cond_dctx_dir(const float* __restrict__ dctxs_t,
+                              const float* __restrict__ daccC,
+                              const float* __restrict__ mask_t,
+                              float* __restrict__ dctx_dir, int B, int C) {
+  const long total = (long)B * C;
+  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int b = idx / C;
+    const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
+    float v = mm * daccC[idx];
+    if (dctxs_t != nullptr) v += dctxs_t[idx];
+    dctx_dir[idx] = v;
+  }
+}
+
+__global__ void cond_gate_bwd(const float* __restrict__ dctx_buf,
+                              const float* __restrict__ ctxs_t,  // gated val
+                              const float* __restrict__ Ucon,
+                              const float* __restrict__ Wcon,
+                              float* __restrict__ daccC,          // in/out
+                              float* __restrict__ dctxpre_f32,    // [B][C]
+                              bf16_t* __restrict__ dctxpre_all_t, // [B][C]
+                              bf16_t* __restrict__ dgate_all_t,   // [B][C]
+                              int B, int C) {
+  const long total = (long)B * C;
+  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int c = idx % C;
+    const float g = ctxs_t[idx];
+    const float dg = dctx_buf[idx] * (1.f - g * g);
+    const float dpre = dg * Ucon[c];
+    dctxpre_f32[idx] = dpre;
+    dctxpre_all_t[idx] = (bf16_t)dpre;
+    dgate_all_t[idx] = (bf16_t)dg;
+    daccC[idx] += dg * Wcon[c];
+  }
+}
+
+// attention backward, stage A (one WG per b): dalpha via the C-dot,
+// softmax backward, per-(s,i) dpc products into scratch, acc_alpha carry.
+__global__ __launch_bounds__(256) void cond_attn_bwd_a(
+    const bf16_t* __restrict__ ctx_bf,       // [Ts][B][C]
+    const float* __restrict__ dctxpre_f32,   // [B][C]
+    const float* __restrict__ alphas_t,      // [B][Ts]
+    const float* __restrict__ pctx,          // [Ts][B][A]
+    const float* __restrict__ pstate_t,      // [B][A]
+    const float* __restrict__ accA_used_t,   // [B][Ts]
+    const float* __restrict__ Dwei, const float* __restrict__ Uatt,
+    const float* __restrict__ mask_t,        // [B] or null
+    const float* __restrict__ dalphas_t,     // [B][Ts] or null
+    float* __restrict__ daccA,               // [B][Ts] in/out
+    float* __restrict__ dal_buf,             // [Ts][B] scratch
+    float* __restrict__ dpctx_acc,           // [Ts][B][A] (+=)
+    float* __restrict__ dpc_buf,             // [B][Ts][Apad]
+    float* __restrict__ depc_buf,            // [B][Ts][Apad]
+    float* __restrict__ gdcatt,              // [1] (atomic)
+    int B, int Ts, int A, int Apad, int C) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* sm_dpre = (float*)smem_raw;  // [C]
+  __shared__ float red[256 / NATS_WAVE];
+  __shared__ float bcast;
+
+  const int b = blockIdx.x;
+  for (int c = threadIdx.x; c < C; c += blockDim.x)
+    sm_dpre[c] = dctxpre_f32[(long)b * C + c];
+  __syncthreads();
+
+  const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
+
+  // pass 1: dalpha + dot(alpha, dalpha)
+  float dot = 0.f;
+  for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
+    const bf16_t* crow = ctx_bf + ((long)s * B + b) * C;
+    float dal = 0.f;
+    int c = 0;
+    for (; c + 8 <= C; c += 8) {
+      bf16x8 v = *(const bf16x8*)(crow + c);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) dal += (float)v[i] * sm_dpre[c + i];
+    }
+    for (; c < C; ++c) dal += (float)crow[c] * sm_dpre[c];
+    dal += mm * daccA[(long)b * Ts + s];
+    if (dalphas_t != nullptr) dal += dalphas_t[(long)b * Ts + s];
+    dal_buf[(long)s * B + b] = dal;
+    dot += alphas_t[(long)b * Ts + s] * dal;
+  }
+#pragma unroll
+  for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
+    dot += __shfl_down(dot, off);
+  if ((threadIdx.x & (NATS_WAVE - 1)) == 0) red[threadIdx.x / NATS_WAVE] = dot;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float D = 0.f;
+    for (int w = 0; w < (int)blockDim.x / NATS_WAVE; ++w) D += red[w];
+    bcast = D;
+  }
+  __syncthreads();
+  const float D = bcast;
+
+  // pass 2a: de -> per-(s,i) products
+  float dcatt = 0.f;
+  for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
+    const float al = alphas_t[(long)b * Ts + s];
+    const float de = al * (dal_buf[(long)s * B + b] - D);
+    const float accAu = accA_used_t[(long)b * Ts + s];
+    const float* prow = pctx + ((long)s * B + b) * A;
+    float* dprow = dpctx_acc + ((long)s * B + b) * A;
+    float* dpcrow = dpc_buf + ((long)b * Ts + s) * Apad;
+    float* depcrow = depc_buf + ((long)b * Ts + s) * Apad;
+    float daccA_add = 0.f;
+    for (int i = 0; i < A; ++i) {
+      const float pc =
+          tanhf(prow[i] + pstate_t[(long)b * A + i] + accAu * Dwei[i]);
+      const float dpc = de * (1.f - pc * pc) * Uatt[i];
+      dprow[i] += dpc;
+      dpcrow[i] = dpc;
+      depcrow[i] = de * pc;
+      daccA_add += dpc * Dwei[i];
+    }
+    daccA[(long)b * Ts + s] += daccA_add;
+    dcatt += de;
+  }
+#pragma unroll
+  for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
+    dcatt += __shfl_down(dcatt, off);
+  if ((threadIdx.x & (NATS_WAVE - 1)) == 0)
+    red[threadIdx.x / NATS_WAVE] = dcatt;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float S = 0.f;
+    for (int w = 0; w < (int)blockDim.x / NATS_WAVE; ++w) S += red[w];
+    atomicAdd(gdcatt, S);
+  }
+}
+
+// attention backward, stage B (one WG per b): reduce the per-(s,i)
+// scratch over s -> dpstate, dD_wei, dU_att.
+__global__ __launch_bounds__(256) void cond_attn_bwd_b(
+    const float* __restrict__ dpc_buf,      // [B][Ts][Apad]
+    const float* __restrict__ depc_buf,     // [B][Ts][Apad]
+    const float* __restrict__ accA_used_t,  // [B][Ts]
+    float* __restrict__ dpstate_all_t,      // [B][A]
+    bf16_t* __restrict__ dstep_att,         // [32][Apad]
+    float* __restrict__ gdDwei,             // [A] (atomic)
+    float* __restrict__ gdUatt,             // [A] (atomic)
+    int B, int Ts, int A, int Apad) {
+  const int b = blockIdx.x;
+  for (int i = threadIdx.x; i < A; i += blockDim.x) {
+    float sps = 0.f, sdw = 0.f, sua = 0.f;
+    for (int s = 0; s < Ts; ++s) {
+      const float d = dpc_buf[((long)b * Ts + s) * Apad + i];
+      sps += d;
+      sdw += d * accA_used_t[(long)b * Ts + s];
+      sua += depc_buf[((long)b * Ts + s) * Apad + i];
+    }
+    dpstate_all_t[(long)b * A + i] = sps;
+    dstep_att[(long)b * Apad + i] = (bf16_t)sps;
+    atomicAdd(&gdDwei[i], sdw);
+    atomicAdd(&gdUatt[i], sua);
+  }
+}
+
+inline int cdiv_i(int a, int b) { return (a + b - 1) / b; }
+
+}  // namespace
+
+// ================= host: forward =================
+std::vector<torch::Tensor> cond_gru_fwd(
+    torch::Tensor yg, torch::Tensor yc, c10::optional<torch::Tensor> mask,
+    torch::Tensor init_state, torch::Tensor ctx_bf,
+    c10::optional<torch::Tensor> ctx_mask, torch::Tensor pctx,
+    torch::Tensor Upk2, torch::Tensor W1pk, torch::Tensor WattPk,
+    torch::Tensor b1, torch::Tensor bx1, torch::Tensor Uatt, double catt,
+    torch::Tensor Dwei, torch::Tensor Wcon, torch::Tensor Ucon,
+    c10::optional<torch::Tensor> accC0, c10::optional<torch::Tensor> accA0) {
+  const int T = yg.size(0), B = yg.size(1);
+  const int H = yc.size(2);
+  const int Ts = ctx_bf.size(0), C = ctx_bf.size(2);
+  const int A = Uatt.size(0);
+  TORCH_CHECK(B <= 32, "cond_gru: batch must be <= 32");
+  TORCH_CHECK(yg.dtype() == torch::kBFloat16 && yg.is_contiguous());
+  TORCH_CHECK(ctx_bf.dtype() == torch::kBFloat16 && ctx_bf.is_contiguous());
+  TORCH_CHECK(pctx.dtype() == torch::kFloat32 && pctx.is_contiguous());
+  const int Hpad = Upk2.size(1);
+  const int K1 = W1pk.size(1);
+  const int ngrpH = cdiv_i(H, JB);
+  TORCH_CHECK(Upk2.size(0) == ngrpH * 3 * JB);
+  TORCH_CHECK(W1pk.size(0) == ngrpH * 4 * JB);
+
+  auto optsF = yg.options().dtype(torch::kFloat32);
+  auto optsB = yg.options();
+  auto h2_all = torch::empty({T, B, H}, optsF);
+  auto h1_all = torch::empty({T, B, H}, optsF);
+  auto ctxs_all = torch::empty({T, B, C}, optsF);
+  auto alphas_all = torch::empty({T, B, Ts}, optsF);
+  auto saved2 = torch::empty({T, B, 3 * H}, optsB);
+  auto saved1 = torch::empty({T, B, 4 * H}, optsB);
+  auto pstate_all = torch::empty({T, B, A}, optsF);
+  auto ctxpre_all = torch::empty({T, B, C}, optsB);
+  auto accA_used = torch::empty({T, B, Ts}, optsF);
+  auto accC_used = torch::empty({T, B, C}, optsB);
+  auto accA = accA0.has_value() ? accA0->contiguous().to(torch::kFloat32)
+                                : torch::zeros({B, Ts}, optsF);
+  auto accC = accC0.has_value() ? accC0->contiguous().to(torch::kFloat32)
+                                : torch::zeros({B, C}, optsF);
+  auto h2bf = torch::zeros({2, 32, Hpad}, optsB);
+  auto hc_bf = torch::zeros({32, K1}, optsB);
+  auto e_buf = torch::empty({Ts, B}, optsF);
+  auto init_f = init_state.contiguous().to(torch::kFloat32);
+  h2bf[0].slice(0, 0, B).slice(1, 0, H).copy_(init_f.to(torch::kBFloat16));
+
+  const float* mask_p = nullptr;
+  torch::Tensor mask_c;
+  if (mask.has_value()) {
+    mask_c = mask->contiguous().to(torch::kFloat32);
+    mask_p = mask_c.data_ptr<float>();
+  }
+  const float* cmask_p = nullptr;
+  torch::Tensor cmask_c;
+  if (ctx_mask.has_value()) {
+    cmask_c = ctx_mask->contiguous().to(torch::kFloat32);
+    cmask_p = cmask_c.data_ptr<float>();
+  }
+
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  const int Apad16 = cdiv_i(A, 16) * 16;
+  const long hbstride = (long)32 * Hpad;
+  bf16_t* h2bf_p = (bf16_t*)h2bf.data_ptr();
+
+  for (int t = 0; t < T; ++t) {
+    const float* h2prev = (t == 0)
+                              ? init_f.data_ptr<float>()
+                              : h2_all.data_ptr<float>() + (long)(t - 1) * B * H;
+    const float* mt = mask_p ? mask_p + (long)t * B : nullptr;
+    // 1) GRU_2 -> h1 (bf16 into hc_bf cols [0,H))
+    hipLaunchKernelGGL(nats_gru_step_fwd, dim3(ngrpH), dim3(384), 0, stream,
+                       h2bf_p + (t % 2) * hbstride, h2prev,
+                       (const bf16_t*)Upk2.data_ptr(),
+                       (const bf16_t*)yg.data_ptr() + (long)t * B * 2 * H,
+                       (const bf16_t*)yc.data_ptr() + (long)t * B * H, mt,
+                       h1_all.data_ptr<float>() + (long)t * B * H,
+                       (bf16_t*)hc_bf.data_ptr(), K1,
+                       (bf16_t*)saved2.data_ptr() + (long)t * B * 3 * H, B, H,
+                       Hpad);
+    // 2) pstate = h1 @ W_att
+    hipLaunchKernelGGL(cond_small_gemm_bt, dim3(2, Apad16 / 16), dim3(64), 0,
+                       stream, (const bf16_t*)hc_bf.data_ptr(),
+                       (const bf16_t*)WattPk.data_ptr(),
+                       pstate_all.data_ptr<float>() + (long)t * B * A, B, A,
+                       K1, (int)WattPk.size(1));
+    // 3) attention scores + softmax + acc_alpha
+    hipLaunchKernelGGL(cond_attn_fwd, dim3(B), dim3(256), 0, stream,
+                       pctx.data_ptr<float>(),
+                       pstate_all.data_ptr<float>() + (long)t * B * A,
+                       accA.data_ptr<float>(),
+                       accA_used.data_ptr<float>() + (long)t * B * Ts,
+                       Dwei.data_ptr<float>(), Uatt.data_ptr<float>(),
+                       (float)catt, cmask_p, mt, e_buf.data_ptr<float>(),
+                       alphas_all.data_ptr<float>() + (long)t * B * Ts, B, Ts,
+                       A);
+    // 4) weighted context + gate + acc_ctx (ctx bf16 into hc_bf [Hpad,...))
+    hipLaunchKernelGGL(cond_attn_ctx_fwd, dim3(B, cdiv_i(C, 256)), dim3(256),
+                       0, stream, (const bf16_t*)ctx_bf.data_ptr(),
+                       alphas_all.data_ptr<float>() + (long)t * B * Ts,
+                       Ucon.data_ptr<float>(), Wcon.data_ptr<float>(),
+                       accC.data_ptr<float>(),
+                       (bf16_t*)accC_used.data_ptr() + (long)t * B * C,
+                       (bf16_t*)ctxpre_all.data_ptr() + (long)t * B * C,
+                       ctxs_all.data_ptr<float>() + (long)t * B * C,
+                       (bf16_t*)hc_bf.data_ptr(), Hpad, K1, mt, B, Ts, C);
+    // 5) GRU_1 -> h2
+    hipLaunchKernelGGL(cond_gru1_step_fwd, dim3(ngrpH), dim3(512), 0, stream,
+                       (const bf16_t*)hc_bf.data_ptr(),
+                       h1_all.data_ptr<float>() + (long)t * B * H,
+                       (const bf16_t*)W1pk.data_ptr(), b1.data_ptr<float>(),
+                       bx1.data_ptr<float>(), mt,
+                       h2_all.data_ptr<float>() + (long)t * B * H,
+                       h2bf_p + ((t + 1) % 2) * hbstride, Hpad,
+                       (bf16_t*)saved1.data_ptr() + (long)t * B * 4 * H, B, H,
+                       K1);
+  }
+  HIP_CHECK(hipGetLastError());
+  return {h2_all, ctxs_all, alphas_all, accC, accA, h1_all, saved2, saved1,
+          pstate_all, ctxpre_all, accA_used, accC_used};
+}
+
+// ================= host: backward =================
+std::vector<torch::Tensor> cond_gru_bwd(
+    torch::Tensor dh2_all, c10::optional<torch::Tensor> dctxs_all,
+    c10::optional<torch::Tensor> dalphas_all,
+    c10::optional<torch::Tensor> daccC_f, c10::optional<torch::Tensor> daccA_f,
+    // saved forward state
+    torch::Tensor yc, torch::Tensor h1_all, torch::Tensor h2_all,
+    torch::Tensor ctxs_all, torch::Tensor alphas_all, torch::Tensor saved2,
+    torch::Tensor saved1, torch::Tensor pstate_all, torch::Tensor ctxpre_all,
+    torch::Tensor accA_used, torch::Tensor accC_used, torch::Tensor ctx_bf,
+    torch::Tensor pctx, torch::Tensor init_state,
+    c10::optional<torch::Tensor> mask,
+    // packed weights for the in-loop GEMMs
+    torch::Tensor U1cat,   // [ngrpH*16][K3Hpad] = [U_1|Ux_1]
+    torch::Tensor W1cat,   // [ngrpC*16][K3Hpad] = [W_1|Wx_1]
+    torch::Tensor U2cat,   // [ngrpH*16][K3Hpad] = [U|Ux]
+    torch::Tensor WattB,   // [ngrpH*16][Apad32] = W_att
+    torch::Tensor bx1, torch::Tensor Dwei, torch::Tensor Uatt,
+    torch::Tensor Ucon, torch::Tensor Wcon) {
+  const int T = dh2_all.size(0), B = dh2_all.size(1), H = dh2_all.size(2);
+  const int Ts = ctx_bf.size(0), C = ctx_bf.size(2);
+  const int A = Uatt.size(0);
+  const int K3Hpad = U1cat.size(1);
+  const int Apad32 = WattB.size(1);
+  const int ngrpH = cdiv_i(H, JB);
+  const int ngrpC = cdiv_i(C, JB);
+
+  auto optsF = dh2_all.options().dtype(torch::kFloat32);
+  auto optsB = dh2_all.options().dtype(torch::kBFloat16);
+  auto dpre1_all = torch::empty({T, B, 4 * H}, optsB);
+  auto dpre2_all = torch::empty({T, B, 4 * H}, optsB);
+  auto dctxpre_all = torch::empty({T, B, C}, optsB);
+  auto dgate_all = torch::empty({T, B, C}, optsB);
+  auto dpstate_all = torch::empty({T, B, A}, optsF);
+  auto dpctx_acc = torch::zeros({Ts, B, A}, optsF);
+  auto gdDwei = torch::zeros({A}, optsF);
+  auto gdUatt = torch::zeros({A}, optsF);
+  auto gdcatt = torch::zeros({1}, optsF);
+
+  auto dh_carry = torch::zeros({B, H}, optsF);
+  auto dh1_buf = torch::zeros({B, H}, optsF);
+  auto ddirect_h1 = torch::empty({B, H}, optsF);
+  auto ddirect2 = torch::empty({B, H}, optsF);
+  auto dctx_dir = torch::empty({B, C}, optsF);
+  auto dctx_buf = torch::empty({B, C}, optsF);
+  auto dctxpre_f32 = torch::empty({B, C}, optsF);
+  auto daccA = daccA_f.has_value() ? daccA_f->contiguous().to(torch::kFloat32)
+                                   : torch::zeros({B, Ts}, optsF);
+  auto daccC = daccC_f.has_value() ? daccC_f->contiguous().to(torch::kFloat32)
+                                   : torch::zeros({B, C}, optsF);
+  auto dal_buf = torch::empty({Ts, B}, optsF);
+  const int Apad = Apad32;
+  auto dpc_buf = torch::empty({B, Ts, Apad}, optsF);
+  auto depc_buf = torch::empty({B, Ts, Apad}, optsF);
+  auto dstep1 = torch::zeros({32, K3Hpad}, optsB);
+  auto dstepC = torch::zeros({32, K3Hpad}, optsB);
+  auto dstep2 = torch::zeros({32, K3Hpad}, optsB);
+  auto dstep_att = torch::zeros({32, Apad32}, optsB);
+  auto init_f = init_state.contiguous().to(torch::kFloat32);
+
+  const float* mask_all = nullptr;
+  torch::Tensor mask_c;
+  if (mask.has_value()) {
+    mask_c = mask->contiguous().to(torch::kFloat32);
+    mask_all = mask_c.data_ptr<float>();
+  }
+  torch::Tensor dctxs_c;
+  const float* dctxs_p = nullptr;
+  if (dctxs_all.has_value()) {
+    dctxs_c = dctxs_all->contiguous().to(torch::kFloat32);
+    dctxs_p = dctxs_c.data_ptr<float>();
+  }
+  torch::Tensor dalpha_c;
+  const float* dalpha_p = nullptr;
+  if (dalphas_all.has_value()) {
+    dalpha_c = dalphas_all->contiguous().to(torch::kFloat32);
+    dalpha_p = dalpha_c.data_ptr<float>();
+  }
+
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  auto dh2_c = dh2_all.contiguous().to(torch::kFloat32);
+  const int pwH = (int)std::min<long>(512, (((long)B * H) + 255) / 256);
+  const int pwC = (int)std::min<long>(512, (((long)B * C) + 255) / 256);
+  const size_t smem_a = (size_t)C * sizeof(float);
+
+  for (int t = T - 1; t >= 0; --t) {
+    const float* mt = mask_all ? mask_all + (long)t * B : nullptr;
+    const float* h2prev =
+        (t == 0) ? init_f.data_ptr<float>()
+                 : h2_all.data_ptr<float>() + (long)(t - 1) * B * H;
+    // b1: GRU_1 pointwise
+    hipLaunchKernelGGL(cond_gru1_bwd_pointwise, dim3(pwH), dim3(256), 0,
+                       stream, dh_carry.data_ptr<float>(),
+                       dh2_c.data_ptr<float>() + (long)t * B * H,
+                       (const bf16_t*)saved1.data_ptr() + (long)t * B * 4 * H,
+                       h1_all.data_ptr<float>() + (long)t * B * H,
+                       bx1.data_ptr<float>(), mt,
+                       (bf16_t*)dstep1.data_ptr(), (bf16_t*)dstepC.data_ptr(),
+                       K3Hpad, ddirect_h1.data_ptr<float>(),
+                       (bf16_t*)dpre1_all.data_ptr() + (long)t * B * 4 * H, B,
+                       H);
+    // b0: dctx passthrough (readout grad + acc chain)
+    hipLaunchKernelGGL(cond_dctx_dir, dim3(pwC), dim3(256), 0, stream,
+                       dctxs_p ? dctxs_p + (long)t * B * C : nullptr,
+                       daccC.data_ptr<float>(), mt,
+                       dctx_dir.data_ptr<float>(), B, C);
+    // b3: dctx = dctx_dir + [dpr2|dpu2|dpx2] @ [W_1|Wx_1]^T
+    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpC), dim3(384), 0,
+                       stream, (const bf16_t*)dstepC.data_ptr(),
+                       (const bf16_t*)W1cat.data_ptr(),
+                       dctx_dir.data_ptr<float>(), dctx_buf.data_ptr<float>(),
+                       B, C, K3Hpad);
+    // b4: distraction gate backward
+    hipLaunchKernelGGL(cond_gate_bwd, dim3(pwC), dim3(256), 0, stream,
+                       dctx_buf.data_ptr<float>(),
+                       ctxs_all.data_ptr<float>() + (long)t * B * C,
+                       Ucon.data_ptr<float>(), Wcon.data_ptr<float>(),
+                       daccC.data_ptr<float>(), dctxpre_f32.data_ptr<float>(),
+                       (bf16_t*)dctxpre_all.data_ptr() + (long)t * B * C,
+                       (bf16_t*)dgate_all.data_ptr() + (long)t * B * C, B, C);
+    // b5a: attention backward stage A (dalpha, softmax bwd, dpc products)
+    hipLaunchKernelGGL(cond_attn_bwd_a, dim3(B), dim3(256), smem_a, stream,
+                       (const bf16_t*)ctx_bf.data_ptr(),
+                       dctxpre_f32.data_ptr<float>(),
+                       alphas_all.data_ptr<float>() + (long)t * B * Ts,
+                       pctx.data_ptr<float>(),
+                       pstate_all.data_ptr<float>() + (long)t * B * A,
+                       accA_used.data_ptr<float>() + (long)t * B * Ts,
+                       Dwei.data_ptr<float>(), Uatt.data_ptr<float>(), mt,
+                       dalpha_p ? dalpha_p + (long)t * B * Ts : nullptr,
+                       daccA.data_ptr<float>(), dal_buf.data_ptr<float>(),
+                       dpctx_acc.data_ptr<float>(),
+                       dpc_buf.data_ptr<float>(), depc_buf.data_ptr<float>(),
+                       gdcatt.data_ptr<float>(), B, Ts, A, Apad, C);
+    // b5b: reduce over s -> dpstate / dD_wei / dU_att
+    hipLaunchKernelGGL(cond_attn_bwd_b, dim3(B), dim3(256), 0, stream,
+                       dpc_buf.data_ptr<float>(), depc_buf.data_ptr<float>(),
+                       accA_used.data_ptr<float>() + (long)t * B * Ts,
+                       dpstate_all.data_ptr<float>() + (long)t * B * A,
+                       (bf16_t*)dstep_att.data_ptr(),
+                       gdDwei.data_ptr<float>(), gdUatt.data_ptr<float>(), B,
+                       Ts, A, Apad);
+    // b2: dh1 += [dpr2|dpu2|dpxa_lin] @ [U_1|Ux_1]^T + passthrough
+    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH), dim3(384), 0,
+                       stream, (const bf16_t*)dstep1.data_ptr(),
+                       (const bf16_t*)U1cat.data_ptr(),
+                       ddirect_h1.data_ptr<float>(),
+                       dh1_buf.data_ptr<float>(), B, H, K3Hpad);
+    // b6a: dh1 += dpstate @ W_att^T (in-place add via ddirect aliasing)
+    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH), dim3(384), 0,
+                       stream, (const bf16_t*)dstep_att.data_ptr(),
+                       (const bf16_t*)WattB.data_ptr(),
+                       dh1_buf.data_ptr<float>(), dh1_buf.data_ptr<float>(),
+                       B, H, Apad32);
+    // b8: GRU_2 pointwise (dh1 -> gate preact grads)
+    hipLaunchKernelGGL(nats_gru_step_bwd_pointwise, dim3(pwH), dim3(256), 0,
+                       stream, dh1_buf.data_ptr<float>(), nullptr,
+                       (const bf16_t*)saved2.data_ptr() + (long)t * B * 3 * H,
+                       (const bf16_t*)yc.data_ptr() + (long)t * B * H, h2prev,
+                       mt, (bf16_t*)dstep2.data_ptr(), K3Hpad,
+                       ddirect2.data_ptr<float>(),
+                       (bf16_t*)dpre2_all.data_ptr() + (long)t * B * 4 * H, B,
+                       H);
+    // b9: dh_{t-1} = [dpr1|dpu1|dpxl1] @ [U|Ux]^T + passthrough
+    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH), dim3(384), 0,
+                       stream, (const bf16_t*)dstep2.data_ptr(),
+                       (const bf16_t*)U2cat.data_ptr(),
+                       ddirect2.data_ptr<float>(), dh_carry.data_ptr<float>(),
+                       B, H, K3Hpad);
+  }
+  HIP_CHECK(hipGetLastError());
+  return {dpre1_all, dpre2_all, dctxpre_all, dgate_all, dpstate_all,
+          dpctx_acc, gdDwei, gdUatt, gdcatt, dh_carry, daccC, daccA};
+}

@@ -7,38 +7,39 @@
 // (16x16x32 bf16, fp32 accumulate), with the per-step preactivations
 // exchanged through LDS between the MFMA phase and the pointwise phase.
 //
-// Weight packing (done by the python wrapper, ops/gru.py):
+// Weight packing (python wrapper, ops/gru.py):
 //   Upk  : [ngrp*3*16, Hpad] bf16, row = output slot. Workgroup `wg` owns
 //          output columns j in [wg*16, wg*16+16); its rows are laid out
 //          [r-cols | u-cols | cand-cols] so B-fragments are contiguous
 //          16-byte loads (see common.h layout note).
-//   Hpad : H rounded up to 32; the padded K-region of every operand is
+//   Hpad : H rounded up to 32; padded K-regions of every operand are
 //          zero-filled so MFMA accumulates exact zeros there.
 //
-// State: h_all (T,B,H) fp32 is the master hidden sequence; h_bf
-// ([32][Hpad] bf16, zero-padded rows/cols) mirrors the CURRENT h for the
-// next step's A-fragments.
+// State: h_all (T,B,H) fp32 is the master hidden sequence; a ping-pong
+// pair of [32][Hpad] bf16 buffers mirrors the current h for the next
+// step's A-fragments (two buffers: every WG reads ALL columns while
+// owning 16, so in-place update would race).
 //
 // The backward scan (reverse-time) mirrors the split: a pointwise kernel
 // turns dh_t into gate-preactivation grads (stored to dpre_all for the
 // big time-batched dW/dU GEMMs done by hipBLASLt from python), then an
 // MFMA kernel computes the recurrent term dh_{t-1} = [dpr|dpu|dpxl] @
 // [U|Ux]^T + passthrough. BPTT semantics = tensor.grad through the scan
-// (nats.py:1340).
+// (nats.py:1340). These kernels are shared with the decoder's GRU_2
+// (gru_kernels.h).
 
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 
 #include "common.h"
+#include "gru_kernels.h"
 
-namespace {
-
-constexpr int JB = 16;  // output columns per workgroup
+static constexpr int JB = 16;  // output columns per workgroup
 
 // ---------------- forward step ----------------
 // block: 384 threads = 6 waves; wave w -> (m = w/3, g = w%3) computes the
 // 16x16 tile rows [16m,16m+16) of matrix g (0=r,1=u,2=cand).
-__global__ __launch_bounds__(384) void gru_step_fwd(
+__global__ __launch_bounds__(384) void nats_gru_step_fwd(
     const bf16_t* __restrict__ h_bf,   // [32][Hpad] current h (bf16)
     const float* __restrict__ h_prev,  // [B][H] fp32 (= h_all[t-1] or h0)
     const bf16_t* __restrict__ Upk,    // [ngrp*3*16][Hpad]
@@ -46,7 +47,8 @@ __global__ __launch_bounds__(384) void gru_step_fwd(
     const bf16_t* __restrict__ xc_t,   // [B][H]  x@Wx+bx at step t
     const float* __restrict__ mask_t,  // [B] or nullptr
     float* __restrict__ h_out,         // [B][H] fp32 (h_all[t])
-    bf16_t* __restrict__ h_bf_out,     // [32][Hpad] bf16 (next step's A)
+    bf16_t* __restrict__ h_bf_out,     // bf16 h out, row stride ld_bfout
+    int ld_bfout,
     bf16_t* __restrict__ saved_t,      // [B][3H] (r,u,pxl) at step t
     int B, int H, int Hpad) {
   __shared__ float pre[3][32][JB + 1];
@@ -57,7 +59,6 @@ __global__ __launch_bounds__(384) void gru_step_fwd(
   const int g = wave % 3;
   const int j0 = wg * JB;
 
-  // MFMA phase
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   const bf16_t* brow = Upk + (long)(wg * 3 + g) * JB * Hpad;
   for (int k = 0; k < Hpad; k += 32) {
@@ -74,7 +75,6 @@ __global__ __launch_bounds__(384) void gru_step_fwd(
   }
   __syncthreads();
 
-  // pointwise phase over B x JB outputs
   for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
     const int b = idx / JB;
     const int c = idx % JB;
@@ -93,7 +93,7 @@ __global__ __launch_bounds__(384) void gru_step_fwd(
       hnew = mm * hnew + (1.f - mm) * hp;
     }
     h_out[(long)b * H + j] = hnew;
-    h_bf_out[(long)b * Hpad + j] = (bf16_t)hnew;
+    h_bf_out[(long)b * ld_bfout + j] = (bf16_t)hnew;
     saved_t[(long)b * 3 * H + j] = (bf16_t)r;
     saved_t[(long)b * 3 * H + H + j] = (bf16_t)u;
     saved_t[(long)b * 3 * H + 2 * H + j] = (bf16_t)px;
@@ -101,27 +101,25 @@ __global__ __launch_bounds__(384) void gru_step_fwd(
 }
 
 // ---------------- backward pointwise ----------------
-// grid-stride over B*H. Produces the packed step K-operand dstep
-// ([32][K3pad] bf16, layout [dpr | dpu | dpxl]) for the recurrent GEMM,
-// the passthrough term ddirect, and dpre_all[t] ([B][4H] bf16:
-// [dpr|dpu|dpx|dpxl]) for the time-batched weight/input GEMMs.
-__global__ void gru_step_bwd_pointwise(
+__global__ void nats_gru_step_bwd_pointwise(
     const float* __restrict__ dh_buf,   // [B][H] recurrent dh (from t+1)
-    const float* __restrict__ dh_out_t, // [B][H] upstream grad at t
+    const float* __restrict__ dh_out_t, // [B][H] upstream grad (or null)
     const bf16_t* __restrict__ saved_t, // [B][3H] (r,u,pxl)
     const bf16_t* __restrict__ xc_t,    // [B][H]
     const float* __restrict__ h_prev,   // [B][H]
     const float* __restrict__ mask_t,   // [B] or nullptr
-    bf16_t* __restrict__ dstep,         // [32][K3pad]
+    bf16_t* __restrict__ dstep,         // [32][ld_dstep] [dpr|dpu|dpxl]
+    int ld_dstep,
     float* __restrict__ ddirect,        // [B][H]
-    bf16_t* __restrict__ dpre_t,        // [B][4H]
-    int B, int H, int K3pad) {
+    bf16_t* __restrict__ dpre_t,        // [B][4H] [dpr|dpu|dpx|dpxl]
+    int B, int H) {
   const long total = (long)B * H;
   for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (long)gridDim.x * blockDim.x) {
     const int b = idx / H;
     const int j = idx % H;
-    const float dh = dh_buf[idx] + dh_out_t[idx];
+    float dh = dh_buf[idx];
+    if (dh_out_t != nullptr) dh += dh_out_t[idx];
     const float r = (float)saved_t[(long)b * 3 * H + j];
     const float u = (float)saved_t[(long)b * 3 * H + H + j];
     const float px = (float)saved_t[(long)b * 3 * H + 2 * H + j];
@@ -136,9 +134,9 @@ __global__ void gru_step_bwd_pointwise(
     const float dpr = dr * r * (1.f - r);
     const float dpu = du * u * (1.f - u);
     ddirect[idx] = dh * (mm * u + (1.f - mm));
-    dstep[(long)b * K3pad + j] = (bf16_t)dpr;
-    dstep[(long)b * K3pad + H + j] = (bf16_t)dpu;
-    dstep[(long)b * K3pad + 2 * H + j] = (bf16_t)dpxl;
+    dstep[(long)b * ld_dstep + j] = (bf16_t)dpr;
+    dstep[(long)b * ld_dstep + H + j] = (bf16_t)dpu;
+    dstep[(long)b * ld_dstep + 2 * H + j] = (bf16_t)dpxl;
     dpre_t[(long)b * 4 * H + j] = (bf16_t)dpr;
     dpre_t[(long)b * 4 * H + H + j] = (bf16_t)dpu;
     dpre_t[(long)b * 4 * H + 2 * H + j] = (bf16_t)dpx;
@@ -147,15 +145,16 @@ __global__ void gru_step_bwd_pointwise(
 }
 
 // ---------------- backward recurrent GEMM ----------------
-// dh_prev[b, i] = ddirect[b, i] + sum_k dstep[b, k] * Ubwd[i, k]
-// Ubwd = [U | Ux] rows (H x 3H, padded to K3pad). Workgroup owns 16
-// i-columns; 6 waves = 2 row-tiles x 3 K-splits, LDS reduce.
-__global__ __launch_bounds__(384) void gru_step_bwd_gemm(
-    const bf16_t* __restrict__ dstep,  // [32][K3pad]
-    const bf16_t* __restrict__ Ubwd,   // [H][K3pad] (row i contiguous)
+// out[b, i] = ddirect[b, i] + sum_k dstep[b, k] * Wt[i, k]
+// Generic over the output width (rows of Wt): the encoder/GRU_2 call has
+// Wt = [U|Ux] (H x 3Hpad); the decoder reuses it with Wt = [W_1|Wx_1]
+// (C rows) and Wt = W_att (K = Apad).
+__global__ __launch_bounds__(384) void nats_gru_step_bwd_gemm(
+    const bf16_t* __restrict__ dstep,  // [32][Kpad]
+    const bf16_t* __restrict__ Wt,     // [ngrp*16][Kpad]
     const float* __restrict__ ddirect, // [B][H]
-    float* __restrict__ dh_buf,        // [B][H] output (overwritten)
-    int B, int H, int K3pad) {
+    float* __restrict__ out,           // [B][H] (may alias ddirect)
+    int B, int H, int Kpad) {
   __shared__ float part[3][32][JB + 1];
 
   const int wg = blockIdx.x;
@@ -164,15 +163,14 @@ __global__ __launch_bounds__(384) void gru_step_bwd_gemm(
   const int ks = wave % 3;
   const int i0 = wg * JB;
 
-  // K ranges per split: thirds of K3pad rounded to 32
-  const int kchunk = ((K3pad / 3 + 31) / 32) * 32;
+  const int kchunk = ((Kpad / 3 + 31) / 32) * 32;
   const int kbeg = ks * kchunk;
-  const int kend = min(K3pad, (ks + 1) * kchunk);
+  const int kend = min(Kpad, (ks + 1) * kchunk);
 
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   for (int k = kbeg; k < kend; k += 32) {
-    bf16x8 a = frag_a_rowmajor(dstep, 16 * m, K3pad, k);
-    bf16x8 b = frag_bt_rowmajor(Ubwd, i0, K3pad, k);
+    bf16x8 a = frag_a_rowmajor(dstep, 16 * m, Kpad, k);
+    bf16x8 b = frag_bt_rowmajor(Wt, i0, Kpad, k);
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
   }
   {
@@ -189,19 +187,15 @@ __global__ __launch_bounds__(384) void gru_step_bwd_gemm(
     const int c = idx % JB;
     const int i = i0 + c;
     if (i >= H) continue;
-    dh_buf[(long)b * H + i] = ddirect[(long)b * H + i] + part[0][b][c] +
-                              part[1][b][c] + part[2][b][c];
+    out[(long)b * H + i] = ddirect[(long)b * H + i] + part[0][b][c] +
+                           part[1][b][c] + part[2][b][c];
   }
 }
-
-}  // namespace
 
 // ---------------- host drivers ----------------
 
 static inline int cdiv(int a, int b) { return (a + b - 1) / b; }
 
-// Forward scan over T steps. Returns (h_all fp32 (T,B,H), saved bf16
-// (T,B,3H)).
 std::vector<torch::Tensor> gru_scan_fwd(torch::Tensor xg, torch::Tensor xc,
                                         c10::optional<torch::Tensor> mask,
                                         torch::Tensor Upk,
@@ -222,9 +216,6 @@ std::vector<torch::Tensor> gru_scan_fwd(torch::Tensor xg, torch::Tensor xc,
   auto optsB = xg.options();
   auto h_all = torch::empty({T, B, H}, optsF);
   auto saved = torch::empty({T, B, 3 * H}, optsB);
-  // ping-pong bf16 h buffers: step t READS buf[t%2] (state t-1) and WRITES
-  // buf[(t+1)%2] (state t) — a single buffer would race across workgroups
-  // (every WG reads ALL columns while owning only its own 16).
   auto h_bf = torch::zeros({2, 32, Hpad}, optsB);
   torch::Tensor hprev0;
   if (h0.has_value()) {
@@ -252,20 +243,18 @@ std::vector<torch::Tensor> gru_scan_fwd(torch::Tensor xg, torch::Tensor xc,
   const long hbuf_stride = (long)32 * Hpad;
   for (int t = 0; t < T; ++t) {
     const float* hprev = (t == 0) ? h0_p : (h_all_p + (long)(t - 1) * B * H);
-    hipLaunchKernelGGL(gru_step_fwd, dim3(ngrp), dim3(384), 0, stream,
+    hipLaunchKernelGGL(nats_gru_step_fwd, dim3(ngrp), dim3(384), 0, stream,
                        hbf_p + (t % 2) * hbuf_stride, hprev, Upk_p,
                        xg_p + (long)t * B * 2 * H, xc_p + (long)t * B * H,
                        mask_p ? mask_p + (long)t * B : nullptr,
                        h_all_p + (long)t * B * H,
-                       hbf_p + ((t + 1) % 2) * hbuf_stride,
+                       hbf_p + ((t + 1) % 2) * hbuf_stride, Hpad,
                        saved_p + (long)t * B * 3 * H, B, H, Hpad);
   }
   HIP_CHECK(hipGetLastError());
   return {h_all, saved};
 }
 
-// Backward scan (reverse time). Returns (dpre_all bf16 (T,B,4H), dh0 fp32
-// (B,H)).
 std::vector<torch::Tensor> gru_scan_bwd(torch::Tensor dh_out,
                                         torch::Tensor h_all,
                                         torch::Tensor saved, torch::Tensor xc,
@@ -278,7 +267,8 @@ std::vector<torch::Tensor> gru_scan_bwd(torch::Tensor dh_out,
   const int K3pad = Ubwd.size(1);
   const int ngrp = cdiv(H, JB);
   TORCH_CHECK(Ubwd.size(0) == ngrp * JB && Ubwd.is_contiguous());
-  TORCH_CHECK(saved.dtype() == torch::kBFloat16 && xc.dtype() == torch::kBFloat16);
+  TORCH_CHECK(saved.dtype() == torch::kBFloat16 &&
+              xc.dtype() == torch::kBFloat16);
 
   auto optsF = dh_out.options();
   auto optsB = dh_out.options().dtype(torch::kBFloat16);
@@ -307,17 +297,18 @@ std::vector<torch::Tensor> gru_scan_bwd(torch::Tensor dh_out,
   bf16_t* dstep_p = (bf16_t*)dstep.data_ptr();
   const bf16_t* Ubwd_p = (const bf16_t*)Ubwd.data_ptr();
 
-  const int pw_blocks = min(1024, (int)((((long)B * H) + 255) / 256));
+  const int pw_blocks = (int)std::min<long>(1024, (((long)B * H) + 255) / 256);
   for (int t = T - 1; t >= 0; --t) {
     const float* hprev = (t == 0) ? hprev0.data_ptr<float>()
                                   : (h_all_p + (long)(t - 1) * B * H);
-    hipLaunchKernelGGL(gru_step_bwd_pointwise, dim3(pw_blocks), dim3(256), 0,
-                       stream, dh_buf_p, dh_out_p + (long)t * B * H,
+    hipLaunchKernelGGL(nats_gru_step_bwd_pointwise, dim3(pw_blocks), dim3(256),
+                       0, stream, dh_buf_p, dh_out_p + (long)t * B * H,
                        saved_p + (long)t * B * 3 * H, xc_p + (long)t * B * H,
                        hprev, mask_p ? mask_p + (long)t * B : nullptr, dstep_p,
-                       ddirect_p, dpre_p + (long)t * B * 4 * H, B, H, K3pad);
-    hipLaunchKernelGGL(gru_step_bwd_gemm, dim3(ngrp), dim3(384), 0, stream,
-                       dstep_p, Ubwd_p, ddirect_p, dh_buf_p, B, H, K3pad);
+                       K3pad, ddirect_p, dpre_p + (long)t * B * 4 * H, B, H);
+    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrp), dim3(384), 0,
+                       stream, dstep_p, Ubwd_p, ddirect_p, dh_buf_p, B, H,
+                       K3pad);
   }
   HIP_CHECK(hipGetLastError());
   return {dpre_all, dh_buf};
