@@ -40,7 +40,9 @@ __global__ void cond_small_gemm_bt(const bf16_t* __restrict__ A,
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   for (int k = 0; k < Kpad; k += 32) {
     bf16x8 a = frag_a_rowmajor(A, m0, ldK, k);
-    bf16x8 b = frag_bt_rowmajor(Bt, n0, ldK, k);
+    // NB: Bt's row stride is Kpad (the packed weight width), NOT the A
+    // operand's ldK — they differ in the decoder ([h1|ctx] operand).
+    bf16x8 b = frag_bt_rowmajor(Bt, n0, Kpad, k);
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
   }
   const int lane = threadIdx.x & (NATS_WAVE - 1);

@@ -159,6 +159,13 @@ def _cond_inputs(T=7, B=5, H=32, Ts=9, A=12, E=10, seed=5, with_masks=True):
     C = 2 * H
     opts = default_options(dim_word=E, dim=H, dim_att=A, n_words=50)
     model = NatsModel(opts, seed=seed)
+    # inflate the 0.01-scale attention/distraction weights so alpha is far
+    # from uniform and the gate far from linear — otherwise a broken
+    # pstate/gate path can hide inside a near-uniform attention pattern
+    with torch.no_grad():
+        for k in ("decoder_W_att", "decoder_Wc_att", "decoder_U_att",
+                  "decoder_D_wei", "decoder_W_con", "decoder_U_con"):
+            model.P[k].mul_(40.0)
     yg = torch.randn(T, B, 2 * H, generator=g)
     yc = torch.randn(T, B, H, generator=g)
     init = torch.randn(B, H, generator=g) * 0.1
