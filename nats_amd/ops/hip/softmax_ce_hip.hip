@@ -53,16 +53,18 @@ __global__ void softmax_ce_fwd_kernel(const bf16_t* __restrict__ logits,
     s += __expf(x - m);
   }
 
-  // wave reduce (max, sum) then LDS reduce across waves
+  // wave reduce (max, sum) then LDS reduce across waves.
+  // Guards: a thread that saw no elements has (m=-inf, s=0); merging two
+  // such pairs must not evaluate exp(-inf - -inf) = NaN.
   __shared__ float sm[BLOCK / NATS_WAVE], ss[BLOCK / NATS_WAVE];
 #pragma unroll
   for (int off = NATS_WAVE / 2; off > 0; off >>= 1) {
     const float om = __shfl_down(m, off);
     const float os = __shfl_down(s, off);
     if (om > m) {
-      s = s * __expf(m - om) + os;
+      s = (s == 0.f) ? os : (s * __expf(m - om) + os);
       m = om;
-    } else {
+    } else if (os != 0.f) {
       s += os * __expf(om - m);
     }
   }
@@ -77,9 +79,9 @@ __global__ void softmax_ce_fwd_kernel(const bf16_t* __restrict__ logits,
 #pragma unroll
     for (int w = 1; w < BLOCK / NATS_WAVE; ++w) {
       if (sm[w] > M) {
-        S = S * __expf(M - sm[w]) + ss[w];
+        S = (S == 0.f) ? ss[w] : (S * __expf(M - sm[w]) + ss[w]);
         M = sm[w];
-      } else {
+      } else if (ss[w] != 0.f) {
         S += ss[w] * __expf(sm[w] - M);
       }
     }

@@ -80,11 +80,13 @@ def test_gru_scan_backward(ext):
         assert rel < 0.06, (name, float(rel))
 
 
-def test_softmax_ce(ext):
+@pytest.mark.parametrize("V", [120, 3001, 30000])
+def test_softmax_ce(ext, V):
+    """V=120 regresses the small-vocab reduce (idle threads hold -inf)."""
     from nats_amd.ops import eager
     from nats_amd.ops.softmax_ce import softmax_xent_hip
     torch.manual_seed(1)
-    N, V = 37, 3001
+    N = 37
     logits = (5 * torch.randn(N, V)).to(torch.bfloat16)
     targets = torch.randint(0, V, (N,))
     ref_l = logits.float().clone().requires_grad_(True)
@@ -220,9 +222,10 @@ def test_cond_gru_backward(ext):
         pctx = ins[3] @ P["decoder_Wc_att"] + P["decoder_b_att"]
         h2s, ctxs, alphas, accC, accA = fn(ins[0], ins[1], m_, ins[2],
                                            ins[3], cm, pctx, P)
-        torch.manual_seed(0)
-        w1 = torch.randn_like(h2s.float())
-        w2 = torch.randn_like(ctxs.float())
+        # loss weights drawn on CPU so both devices optimise the SAME loss
+        g2 = torch.Generator().manual_seed(0)
+        w1 = torch.randn(h2s.shape, generator=g2).to(dev)
+        w2 = torch.randn(ctxs.shape, generator=g2).to(dev)
         ((h2s.float() * w1).sum() + (ctxs.float() * w2).sum()).backward()
         grads = {k: P[k].grad.cpu().float() for k in keys}
         gins = {n: t.grad.cpu().float()
