@@ -94,16 +94,35 @@ def main():
 
     amp = torch.autocast("cuda", dtype=torch.bfloat16) if use_cuda else None
 
+    breakdown = os.environ.get("NATS_BENCH_BREAKDOWN")
+    bd = {"fwd": 0.0, "bwd": 0.0, "opt": 0.0, "n": 0}
+
     def step():
         opt.zero_grad()
+        if breakdown:
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
         if amp is not None:
             with amp:
                 cost = model(x, x_mask, y, y_mask).mean()
         else:
             cost = model(x, x_mask, y, y_mask).mean()
+        if breakdown:
+            torch.cuda.synchronize()
+            t1 = time.perf_counter()
         cost.backward()
         dp.finish()
+        if breakdown:
+            torch.cuda.synchronize()
+            t2 = time.perf_counter()
         opt.step()
+        if breakdown:
+            torch.cuda.synchronize()
+            t3 = time.perf_counter()
+            bd["fwd"] += t1 - t0
+            bd["bwd"] += t2 - t1
+            bd["opt"] += t3 - t2
+            bd["n"] += 1
         return cost
 
     def barrier_sync():
@@ -132,6 +151,11 @@ def main():
     tokens_per_step = world * batch * ((cfg["src"] + 1) + (cfg["tgt"] + 1))
     toks_per_sec = tokens_per_step * args.steps / elapsed
     ms_per_step = 1e3 * elapsed / args.steps
+
+    if breakdown and bd["n"] and rank == 0:
+        print("BREAKDOWN ms/step: fwd=%.2f bwd=%.2f opt=%.2f" % (
+            1e3 * bd["fwd"] / bd["n"], 1e3 * bd["bwd"] / bd["n"],
+            1e3 * bd["opt"] / bd["n"]), file=sys.stderr)
 
     if rank == 0:
         result = {
