@@ -133,6 +133,7 @@ def main():
 
     for _ in range(args.warmup):
         step()
+    bd.update(fwd=0.0, bwd=0.0, opt=0.0, n=0)  # exclude warmup
     barrier_sync()
     t0 = time.perf_counter()
     last_cost = None
