@@ -39,13 +39,8 @@ __global__ void cond_small_gemm_bt(const bf16_t* __restrict__ A,
   const int m0 = blockIdx.x * 16;
   const int n0 = blockIdx.y * 16;
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  for (int k = 0; k < Kpad; k += 32) {
-    bf16x8 a = frag_a_rowmajor(A, m0, ldK, k);
-    // NB: Bt's row stride is Kpad (the packed weight width), NOT the A
-    // operand's ldK — they differ in the decoder ([h1|ctx] operand).
-    bf16x8 b = frag_bt_rowmajor(Bt, n0, Kpad, k);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
-  }
+  // NB: Bt's row stride is Kpad (the packed weight width), NOT ldK.
+  NATS_MFMA_KLOOP(acc, A, m0, ldK, Bt, n0, Kpad, 0, Kpad);
   const int lane = threadIdx.x & (NATS_WAVE - 1);
   const int col = n0 + (lane & 15);
   const int rbase = m0 + (lane >> 4) * 4;
@@ -57,38 +52,54 @@ __global__ void cond_small_gemm_bt(const bf16_t* __restrict__ A,
   }
 }
 
-// ---------------- attention scores + softmax (one WG per b) ----------
-__global__ __launch_bounds__(256) void cond_attn_fwd(
+// ---------------- attention e-scores (s-parallel) ----------------
+// grid (B, ceil(Ts/256)); thread = one source position s.
+__global__ __launch_bounds__(256) void cond_attn_escore(
     const float* __restrict__ pctx,      // [Ts][B][A]
     const float* __restrict__ pstate_t,  // [B][A]
-    float* __restrict__ accA,            // [B][Ts] in/out
-    float* __restrict__ accA_used_t,     // [B][Ts] out (pre-update copy)
+    const float* __restrict__ accA,      // [B][Ts] (pre-update)
     const float* __restrict__ Dwei,      // [A]
     const float* __restrict__ Uatt,      // [A]
     float catt,
+    float* __restrict__ e_buf,           // [Ts][B]
+    int B, int Ts, int A) {
+  const int b = blockIdx.x;
+  const int s = blockIdx.y * blockDim.x + threadIdx.x;
+  if (s >= Ts) return;
+  const float accAu = accA[(long)b * Ts + s];
+  const float* prow = pctx + ((long)s * B + b) * A;
+  const float* srow = pstate_t + (long)b * A;
+  float e = catt;
+  int i = 0;
+  const int A4 = A & ~3;
+  for (; i < A4; i += 4) {
+    const float4 p = *(const float4*)(prow + i);
+    e += tanhf(p.x + srow[i] + accAu * Dwei[i]) * Uatt[i];
+    e += tanhf(p.y + srow[i + 1] + accAu * Dwei[i + 1]) * Uatt[i + 1];
+    e += tanhf(p.z + srow[i + 2] + accAu * Dwei[i + 2]) * Uatt[i + 2];
+    e += tanhf(p.w + srow[i + 3] + accAu * Dwei[i + 3]) * Uatt[i + 3];
+  }
+  for (; i < A; ++i)
+    e += tanhf(prow[i] + srow[i] + accAu * Dwei[i]) * Uatt[i];
+  e_buf[(long)s * B + b] = e;
+}
+
+// ---------------- softmax finish + acc_alpha update (one WG per b) ----
+__global__ __launch_bounds__(256) void cond_attn_softmax(
+    float* __restrict__ accA,            // [B][Ts] in/out
+    float* __restrict__ accA_used_t,     // [B][Ts] out (pre-update copy)
     const float* __restrict__ ctx_mask,  // [Ts][B] or null
     const float* __restrict__ mask_t,    // [B] or null
-    float* __restrict__ e_buf,           // [Ts][B] scratch
+    const float* __restrict__ e_buf,     // [Ts][B]
     float* __restrict__ alphas_t,        // [B][Ts] out
-    int B, int Ts, int A) {
+    int B, int Ts) {
   const int b = blockIdx.x;
   __shared__ float red[256 / NATS_WAVE];
   __shared__ float bcast;
 
-  // pass 1: scores + running max
   float lmax = -INFINITY;
-  for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
-    const float accAu = accA[(long)b * Ts + s];
-    accA_used_t[(long)b * Ts + s] = accAu;
-    const float* prow = pctx + ((long)s * B + b) * A;
-    float e = catt;
-    for (int i = 0; i < A; ++i) {
-      e += tanhf(prow[i] + pstate_t[(long)b * A + i] + accAu * Dwei[i]) *
-           Uatt[i];
-    }
-    e_buf[(long)s * B + b] = e;
-    lmax = fmaxf(lmax, e);
-  }
+  for (int s = threadIdx.x; s < Ts; s += blockDim.x)
+    lmax = fmaxf(lmax, e_buf[(long)s * B + b]);
 #pragma unroll
   for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
     lmax = fmaxf(lmax, __shfl_down(lmax, off));
@@ -104,12 +115,11 @@ __global__ __launch_bounds__(256) void cond_attn_fwd(
   __syncthreads();
   const float M = bcast;
 
-  // pass 2: exp * mask + sum
   float lsum = 0.f;
   for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
     float a = __expf(e_buf[(long)s * B + b] - M);
     if (ctx_mask != nullptr) a *= ctx_mask[(long)s * B + b];
-    e_buf[(long)s * B + b] = a;
+    alphas_t[(long)b * Ts + s] = a;  // unnormalised, fixed below
     lsum += a;
   }
 #pragma unroll
@@ -126,36 +136,61 @@ __global__ __launch_bounds__(256) void cond_attn_fwd(
   __syncthreads();
   const float inv = 1.f / bcast;
 
-  // pass 3: normalise + acc_alpha update (nats.py:540, 570)
   const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
   for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
-    const float al = e_buf[(long)s * B + b] * inv;
+    const float al = alphas_t[(long)b * Ts + s] * inv;
     alphas_t[(long)b * Ts + s] = al;
+    accA_used_t[(long)b * Ts + s] = accA[(long)b * Ts + s];
     accA[(long)b * Ts + s] += mm * al;
   }
 }
 
-// ---------------- weighted context + distraction gate ----------------
-__global__ void cond_attn_ctx_fwd(
+// ---------------- weighted context: s-chunked partial sums ----------
+// grid (B, ceil(C/256), SCH); f32 atomicAdd into ctxpre_f32 (zeroed per
+// step). Unrolled by 4 over s for memory-level parallelism.
+__global__ void cond_attn_ctx_partial(
     const bf16_t* __restrict__ ctx_bf,   // [Ts][B][C]
     const float* __restrict__ alphas_t,  // [B][Ts]
-    const float* __restrict__ Ucon,      // [C]
-    const float* __restrict__ Wcon,      // [C]
-    float* __restrict__ accC,            // [B][C] in/out
-    bf16_t* __restrict__ accC_used_t,    // [B][C]
-    bf16_t* __restrict__ ctxpre_t,       // [B][C] (pre-gate sum)
-    float* __restrict__ ctxs_t,          // [B][C] (gated output)
-    bf16_t* __restrict__ hc_bf,          // [32][K1] GRU_1 operand
-    int ctx_off,                         // = Hpad (ctx block offset in K1)
-    int ldK1, const float* __restrict__ mask_t, int B, int Ts, int C) {
+    float* __restrict__ ctxpre_f32,      // [B][C] (accumulated)
+    int B, int Ts, int C, int SCH) {
   const int b = blockIdx.x;
   const int c = blockIdx.y * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float sum = 0.f;
-  for (int s = 0; s < Ts; ++s) {
-    sum += (float)ctx_bf[((long)s * B + b) * C + c] *
-           alphas_t[(long)b * Ts + s];
+  const int chunk = (Ts + SCH - 1) / SCH;
+  const int sbeg = blockIdx.z * chunk;
+  const int send = min(Ts, sbeg + chunk);
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+  int s = sbeg;
+  for (; s + 4 <= send; s += 4) {
+    s0 += (float)ctx_bf[((long)(s + 0) * B + b) * C + c] *
+          alphas_t[(long)b * Ts + s + 0];
+    s1 += (float)ctx_bf[((long)(s + 1) * B + b) * C + c] *
+          alphas_t[(long)b * Ts + s + 1];
+    s2 += (float)ctx_bf[((long)(s + 2) * B + b) * C + c] *
+          alphas_t[(long)b * Ts + s + 2];
+    s3 += (float)ctx_bf[((long)(s + 3) * B + b) * C + c] *
+          alphas_t[(long)b * Ts + s + 3];
   }
+  for (; s < send; ++s)
+    s0 += (float)ctx_bf[((long)s * B + b) * C + c] *
+          alphas_t[(long)b * Ts + s];
+  atomicAdd(&ctxpre_f32[(long)b * C + c], s0 + s1 + s2 + s3);
+}
+
+// ---------------- distraction gate + acc_ctx update ----------------
+__global__ void cond_attn_gate_fwd(
+    const float* __restrict__ ctxpre_f32,  // [B][C]
+    const float* __restrict__ Ucon, const float* __restrict__ Wcon,
+    float* __restrict__ accC,            // [B][C] in/out
+    bf16_t* __restrict__ accC_used_t,    // [B][C]
+    bf16_t* __restrict__ ctxpre_t,       // [B][C] (saved pre-gate sum)
+    float* __restrict__ ctxs_t,          // [B][C] (gated output)
+    bf16_t* __restrict__ hc_bf,          // [32][K1] GRU_1 operand
+    int ctx_off, int ldK1, const float* __restrict__ mask_t, int B, int C) {
+  const int b = blockIdx.x;
+  const int c = blockIdx.y * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float sum = ctxpre_f32[(long)b * C + c];
   ctxpre_t[(long)b * C + c] = (bf16_t)sum;
   const float accCv = accC[(long)b * C + c];
   accC_used_t[(long)b * C + c] = (bf16_t)accCv;
@@ -193,11 +228,7 @@ __global__ __launch_bounds__(512) void cond_gru1_step_fwd(
 
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   const bf16_t* brow = W1pk + (long)(wg * 4 + g) * JB * K1;
-  for (int k = 0; k < K1; k += 32) {
-    bf16x8 a = frag_a_rowmajor(hc_bf, 16 * m, K1, k);
-    bf16x8 b = frag_bt_rowmajor(brow, 0, K1, k);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
-  }
+  NATS_MFMA_KLOOP(acc, hc_bf, 16 * m, K1, brow, 0, K1, 0, K1);
   {
     const int lane = threadIdx.x & (NATS_WAVE - 1);
     const int col = lane & 15;
@@ -319,127 +350,149 @@ __global__ void cond_gate_bwd(const float* __restrict__ dctx_buf,
   }
 }
 
-// attention backward, stage A (one WG per b): dalpha via the C-dot,
-// softmax backward, per-(s,i) dpc products into scratch, acc_alpha carry.
-__global__ __launch_bounds__(256) void cond_attn_bwd_a(
-    const bf16_t* __restrict__ ctx_bf,       // [Ts][B][C]
-    const float* __restrict__ dctxpre_f32,   // [B][C]
-    const float* __restrict__ alphas_t,      // [B][Ts]
-    const float* __restrict__ pctx,          // [Ts][B][A]
-    const float* __restrict__ pstate_t,      // [B][A]
-    const float* __restrict__ accA_used_t,   // [B][Ts]
-    const float* __restrict__ Dwei, const float* __restrict__ Uatt,
-    const float* __restrict__ mask_t,        // [B] or null
-    const float* __restrict__ dalphas_t,     // [B][Ts] or null
-    float* __restrict__ daccA,               // [B][Ts] in/out
-    float* __restrict__ dal_buf,             // [Ts][B] scratch
-    float* __restrict__ dpctx_acc,           // [Ts][B][A] (+=)
-    float* __restrict__ dpc_buf,             // [B][Ts][Apad]
-    float* __restrict__ depc_buf,            // [B][Ts][Apad]
-    float* __restrict__ gdcatt,              // [1] (atomic)
-    int B, int Ts, int A, int Apad, int C) {
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* sm_dpre = (float*)smem_raw;  // [C]
-  __shared__ float red[256 / NATS_WAVE];
-  __shared__ float bcast;
-
+// attention backward, stage 1 (one WG per (b,s)): dalpha[s,b] =
+// ctx[s,b,:] . dctxpre[b,:] (+ acc-chain and upstream alpha grads) and the
+// softmax-backward dot partial, accumulated by atomics into dot_buf.
+__global__ __launch_bounds__(256) void cond_attn_bwd_dalpha(
+    const bf16_t* __restrict__ ctx_bf,      // [Ts][B][C]
+    const float* __restrict__ dctxpre_f32,  // [B][C]
+    const float* __restrict__ alphas_t,     // [B][Ts]
+    const float* __restrict__ daccA,        // [B][Ts]
+    const float* __restrict__ mask_t,       // [B] or null
+    const float* __restrict__ dalphas_t,    // [B][Ts] or null
+    float* __restrict__ dal_buf,            // [Ts][B]
+    float* __restrict__ dot_buf,            // [B] (zeroed per step)
+    int B, int Ts, int C) {
   const int b = blockIdx.x;
-  for (int c = threadIdx.x; c < C; c += blockDim.x)
-    sm_dpre[c] = dctxpre_f32[(long)b * C + c];
-  __syncthreads();
-
-  const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
-
-  // pass 1: dalpha + dot(alpha, dalpha)
-  float dot = 0.f;
-  for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
-    const bf16_t* crow = ctx_bf + ((long)s * B + b) * C;
-    float dal = 0.f;
-    int c = 0;
-    for (; c + 8 <= C; c += 8) {
-      bf16x8 v = *(const bf16x8*)(crow + c);
+  const int s = blockIdx.y;
+  __shared__ float red[256 / NATS_WAVE];
+  const bf16_t* crow = ctx_bf + ((long)s * B + b) * C;
+  const float* drow = dctxpre_f32 + (long)b * C;
+  float part = 0.f;
+  const int C8 = C & ~7;
+  for (int c = threadIdx.x * 8; c < C8; c += blockDim.x * 8) {
+    bf16x8 v = *(const bf16x8*)(crow + c);
+    const float4 d0 = *(const float4*)(drow + c);
+    const float4 d1 = *(const float4*)(drow + c + 4);
+    part += (float)v[0] * d0.x + (float)v[1] * d0.y + (float)v[2] * d0.z +
+            (float)v[3] * d0.w + (float)v[4] * d1.x + (float)v[5] * d1.y +
+            (float)v[6] * d1.z + (float)v[7] * d1.w;
+  }
+  for (int c = C8 + threadIdx.x; c < C; c += blockDim.x) {
+    part += (float)crow[c] * drow[c];
+  }
 #pragma unroll
-      for (int i = 0; i < 8; ++i) dal += (float)v[i] * sm_dpre[c + i];
-    }
-    for (; c < C; ++c) dal += (float)crow[c] * sm_dpre[c];
+  for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
+    part += __shfl_down(part, off);
+  if ((threadIdx.x & (NATS_WAVE - 1)) == 0)
+    red[threadIdx.x / NATS_WAVE] = part;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float dal = 0.f;
+    for (int w = 0; w < (int)blockDim.x / NATS_WAVE; ++w) dal += red[w];
+    const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
     dal += mm * daccA[(long)b * Ts + s];
     if (dalphas_t != nullptr) dal += dalphas_t[(long)b * Ts + s];
     dal_buf[(long)s * B + b] = dal;
-    dot += alphas_t[(long)b * Ts + s] * dal;
+    atomicAdd(&dot_buf[b], alphas_t[(long)b * Ts + s] * dal);
   }
-#pragma unroll
-  for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
-    dot += __shfl_down(dot, off);
-  if ((threadIdx.x & (NATS_WAVE - 1)) == 0) red[threadIdx.x / NATS_WAVE] = dot;
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    float D = 0.f;
-    for (int w = 0; w < (int)blockDim.x / NATS_WAVE; ++w) D += red[w];
-    bcast = D;
-  }
-  __syncthreads();
-  const float D = bcast;
+}
 
-  // pass 2a: de -> per-(s,i) products
-  float dcatt = 0.f;
-  for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
+// attention backward, stage 2 (grid (B, ceil(Ts/256))): softmax backward,
+// per-(s,i) dpc with wave-reduced accumulation of dpstate/dD_wei/dU_att
+// (no per-(s,i) scratch), and the acc_alpha gradient carry.
+__global__ __launch_bounds__(256) void cond_attn_bwd_scatter(
+    const float* __restrict__ alphas_t,     // [B][Ts]
+    const float* __restrict__ dal_buf,      // [Ts][B]
+    const float* __restrict__ dot_buf,      // [B]
+    const float* __restrict__ pctx,         // [Ts][B][A]
+    const float* __restrict__ pstate_t,     // [B][A]
+    const float* __restrict__ accA_used_t,  // [B][Ts]
+    const float* __restrict__ Dwei, const float* __restrict__ Uatt,
+    float* __restrict__ daccA,              // [B][Ts] in/out
+    float* __restrict__ dpctx_acc,          // [Ts][B][A] (+=)
+    float* __restrict__ dpstate_t,          // [B][A] (zeroed, atomic +=)
+    float* __restrict__ gdDwei,             // [A] (atomic)
+    float* __restrict__ gdUatt,             // [A] (atomic)
+    float* __restrict__ gdcatt,             // [1] (atomic)
+    int B, int Ts, int A) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* sm_dps = (float*)smem_raw;            // [A]
+  float* sm_ddw = sm_dps + A;                  // [A]
+  float* sm_dua = sm_ddw + A;                  // [A]
+  __shared__ float red[256 / NATS_WAVE];
+  const int b = blockIdx.x;
+  const int s = blockIdx.y * blockDim.x + threadIdx.x;
+  const bool live = (s < Ts);
+  for (int i = threadIdx.x; i < 3 * A; i += blockDim.x) sm_dps[i] = 0.f;
+  __syncthreads();
+
+  float de = 0.f, accAu = 0.f;
+  const float* prow = nullptr;
+  float* dprow = nullptr;
+  if (live) {
     const float al = alphas_t[(long)b * Ts + s];
-    const float de = al * (dal_buf[(long)s * B + b] - D);
-    const float accAu = accA_used_t[(long)b * Ts + s];
-    const float* prow = pctx + ((long)s * B + b) * A;
-    float* dprow = dpctx_acc + ((long)s * B + b) * A;
-    float* dpcrow = dpc_buf + ((long)b * Ts + s) * Apad;
-    float* depcrow = depc_buf + ((long)b * Ts + s) * Apad;
-    float daccA_add = 0.f;
-    for (int i = 0; i < A; ++i) {
-      const float pc =
-          tanhf(prow[i] + pstate_t[(long)b * A + i] + accAu * Dwei[i]);
-      const float dpc = de * (1.f - pc * pc) * Uatt[i];
-      dprow[i] += dpc;
-      dpcrow[i] = dpc;
-      depcrow[i] = de * pc;
-      daccA_add += dpc * Dwei[i];
-    }
-    daccA[(long)b * Ts + s] += daccA_add;
-    dcatt += de;
+    de = al * (dal_buf[(long)s * B + b] - dot_buf[b]);
+    accAu = accA_used_t[(long)b * Ts + s];
+    prow = pctx + ((long)s * B + b) * A;
+    dprow = dpctx_acc + ((long)s * B + b) * A;
   }
+  const float* srow = pstate_t + (long)b * A;
+  const int lane = threadIdx.x & (NATS_WAVE - 1);
+  float daccA_add = 0.f;
+  for (int i = 0; i < A; ++i) {
+    float dpc = 0.f, v2 = 0.f, v3 = 0.f;
+    if (live) {
+      const float pc = tanhf(prow[i] + srow[i] + accAu * Dwei[i]);
+      dpc = de * (1.f - pc * pc) * Uatt[i];
+      dprow[i] += dpc;
+      daccA_add += dpc * Dwei[i];
+      v2 = dpc * accAu;
+      v3 = de * pc;
+    }
+    float v1 = dpc;
 #pragma unroll
-  for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
-    dcatt += __shfl_down(dcatt, off);
-  if ((threadIdx.x & (NATS_WAVE - 1)) == 0)
-    red[threadIdx.x / NATS_WAVE] = dcatt;
+    for (int off = NATS_WAVE / 2; off > 0; off >>= 1) {
+      v1 += __shfl_down(v1, off);
+      v2 += __shfl_down(v2, off);
+      v3 += __shfl_down(v3, off);
+    }
+    if (lane == 0) {
+      atomicAdd(&sm_dps[i], v1);
+      atomicAdd(&sm_ddw[i], v2);
+      atomicAdd(&sm_dua[i], v3);
+    }
+  }
+  if (live) daccA[(long)b * Ts + s] += daccA_add;
+
+  // dcatt = sum de
+  float dc = de;
+#pragma unroll
+  for (int off = NATS_WAVE / 2; off > 0; off >>= 1) dc += __shfl_down(dc, off);
+  if ((threadIdx.x & (NATS_WAVE - 1)) == 0) red[threadIdx.x / NATS_WAVE] = dc;
   __syncthreads();
   if (threadIdx.x == 0) {
     float S = 0.f;
     for (int w = 0; w < (int)blockDim.x / NATS_WAVE; ++w) S += red[w];
     atomicAdd(gdcatt, S);
   }
+  for (int i = threadIdx.x; i < A; i += blockDim.x) {
+    atomicAdd(&dpstate_t[(long)b * A + i], sm_dps[i]);
+    atomicAdd(&gdDwei[i], sm_ddw[i]);
+    atomicAdd(&gdUatt[i], sm_dua[i]);
+  }
 }
 
-// attention backward, stage B (one WG per b): reduce the per-(s,i)
-// scratch over s -> dpstate, dD_wei, dU_att.
-__global__ __launch_bounds__(256) void cond_attn_bwd_b(
-    const float* __restrict__ dpc_buf,      // [B][Ts][Apad]
-    const float* __restrict__ depc_buf,     // [B][Ts][Apad]
-    const float* __restrict__ accA_used_t,  // [B][Ts]
-    float* __restrict__ dpstate_all_t,      // [B][A]
-    bf16_t* __restrict__ dstep_att,         // [32][Apad]
-    float* __restrict__ gdDwei,             // [A] (atomic)
-    float* __restrict__ gdUatt,             // [A] (atomic)
-    int B, int Ts, int A, int Apad) {
-  const int b = blockIdx.x;
-  for (int i = threadIdx.x; i < A; i += blockDim.x) {
-    float sps = 0.f, sdw = 0.f, sua = 0.f;
-    for (int s = 0; s < Ts; ++s) {
-      const float d = dpc_buf[((long)b * Ts + s) * Apad + i];
-      sps += d;
-      sdw += d * accA_used_t[(long)b * Ts + s];
-      sua += depc_buf[((long)b * Ts + s) * Apad + i];
-    }
-    dpstate_all_t[(long)b * A + i] = sps;
-    dstep_att[(long)b * Apad + i] = (bf16_t)sps;
-    atomicAdd(&gdDwei[i], sdw);
-    atomicAdd(&gdUatt[i], sua);
+// bf16 copy of dpstate into the padded GEMM operand
+__global__ void cond_dpstate_cast(const float* __restrict__ dpstate_t,
+                                  bf16_t* __restrict__ dstep_att, int Apad,
+                                  int B, int A) {
+  const long total = (long)B * A;
+  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int b = idx / A;
+    const int i = idx % A;
+    dstep_att[(long)b * Apad + i] = (bf16_t)dpstate_t[idx];
   }
 }
 
@@ -489,6 +542,8 @@ std::vector<torch::Tensor> cond_gru_fwd(
   auto h2bf = torch::zeros({2, 32, Hpad}, optsB);
   auto hc_bf = torch::zeros({32, K1}, optsB);
   auto e_buf = torch::empty({Ts, B}, optsF);
+  auto ctxpre_f32 = torch::empty({B, C}, optsF);
+  const int SCH = std::max(1, std::min(8, Ts / 64));
   auto init_f = init_state.contiguous().to(torch::kFloat32);
   h2bf[0].slice(0, 0, B).slice(1, 0, H).copy_(init_f.to(torch::kBFloat16));
 
@@ -531,26 +586,35 @@ std::vector<torch::Tensor> cond_gru_fwd(
                        (const bf16_t*)WattPk.data_ptr(),
                        pstate_all.data_ptr<float>() + (long)t * B * A, B, A,
                        K1, (int)WattPk.size(1));
-    // 3) attention scores + softmax + acc_alpha
-    hipLaunchKernelGGL(cond_attn_fwd, dim3(B), dim3(256), 0, stream,
-                       pctx.data_ptr<float>(),
+    // 3) attention scores (s-parallel) + softmax finish + acc_alpha
+    hipLaunchKernelGGL(cond_attn_escore, dim3(B, cdiv_i(Ts, 256)), dim3(256),
+                       0, stream, pctx.data_ptr<float>(),
                        pstate_all.data_ptr<float>() + (long)t * B * A,
+                       accA.data_ptr<float>(), Dwei.data_ptr<float>(),
+                       Uatt.data_ptr<float>(), (float)catt,
+                       e_buf.data_ptr<float>(), B, Ts, A);
+    hipLaunchKernelGGL(cond_attn_softmax, dim3(B), dim3(256), 0, stream,
                        accA.data_ptr<float>(),
                        accA_used.data_ptr<float>() + (long)t * B * Ts,
-                       Dwei.data_ptr<float>(), Uatt.data_ptr<float>(),
-                       (float)catt, cmask_p, mt, e_buf.data_ptr<float>(),
-                       alphas_all.data_ptr<float>() + (long)t * B * Ts, B, Ts,
-                       A);
-    // 4) weighted context + gate + acc_ctx (ctx bf16 into hc_bf [Hpad,...))
-    hipLaunchKernelGGL(cond_attn_ctx_fwd, dim3(B, cdiv_i(C, 256)), dim3(256),
-                       0, stream, (const bf16_t*)ctx_bf.data_ptr(),
+                       cmask_p, mt, e_buf.data_ptr<float>(),
+                       alphas_all.data_ptr<float>() + (long)t * B * Ts, B,
+                       Ts);
+    // 4) weighted context (s-chunked partials) + gate + acc_ctx
+    HIP_CHECK(hipMemsetAsync(ctxpre_f32.data_ptr<float>(), 0,
+                             (size_t)B * C * sizeof(float), stream));
+    hipLaunchKernelGGL(cond_attn_ctx_partial,
+                       dim3(B, cdiv_i(C, 256), SCH), dim3(256), 0, stream,
+                       (const bf16_t*)ctx_bf.data_ptr(),
                        alphas_all.data_ptr<float>() + (long)t * B * Ts,
+                       ctxpre_f32.data_ptr<float>(), B, Ts, C, SCH);
+    hipLaunchKernelGGL(cond_attn_gate_fwd, dim3(B, cdiv_i(C, 256)), dim3(256),
+                       0, stream, ctxpre_f32.data_ptr<float>(),
                        Ucon.data_ptr<float>(), Wcon.data_ptr<float>(),
                        accC.data_ptr<float>(),
                        (bf16_t*)accC_used.data_ptr() + (long)t * B * C,
                        (bf16_t*)ctxpre_all.data_ptr() + (long)t * B * C,
                        ctxs_all.data_ptr<float>() + (long)t * B * C,
-                       (bf16_t*)hc_bf.data_ptr(), Hpad, K1, mt, B, Ts, C);
+                       (bf16_t*)hc_bf.data_ptr(), Hpad, K1, mt, B, C);
     // 5) GRU_1 -> h2
     hipLaunchKernelGGL(cond_gru1_step_fwd, dim3(ngrpH), dim3(512), 0, stream,
                        (const bf16_t*)hc_bf.data_ptr(),
@@ -618,9 +682,8 @@ std::vector<torch::Tensor> cond_gru_bwd(
   auto daccC = daccC_f.has_value() ? daccC_f->contiguous().to(torch::kFloat32)
                                    : torch::zeros({B, C}, optsF);
   auto dal_buf = torch::empty({Ts, B}, optsF);
+  auto dot_buf = torch::empty({B}, optsF);
   const int Apad = Apad32;
-  auto dpc_buf = torch::empty({B, Ts, Apad}, optsF);
-  auto depc_buf = torch::empty({B, Ts, Apad}, optsF);
   auto dstep1 = torch::zeros({32, K3Hpad}, optsB);
   auto dstepC = torch::zeros({32, K3Hpad}, optsB);
   auto dstep2 = torch::zeros({32, K3Hpad}, optsB);
@@ -687,28 +750,37 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        daccC.data_ptr<float>(), dctxpre_f32.data_ptr<float>(),
                        (bf16_t*)dctxpre_all.data_ptr() + (long)t * B * C,
                        (bf16_t*)dgate_all.data_ptr() + (long)t * B * C, B, C);
-    // b5a: attention backward stage A (dalpha, softmax bwd, dpc products)
-    hipLaunchKernelGGL(cond_attn_bwd_a, dim3(B), dim3(256), smem_a, stream,
-                       (const bf16_t*)ctx_bf.data_ptr(),
+    // b5: attention backward — s-parallel dalpha + softmax-bwd dot, then
+    // the scatter stage with wave-reduced dpstate/dD_wei/dU_att
+    HIP_CHECK(hipMemsetAsync(dot_buf.data_ptr<float>(), 0,
+                             (size_t)B * sizeof(float), stream));
+    HIP_CHECK(hipMemsetAsync(
+        dpstate_all.data_ptr<float>() + (long)t * B * A, 0,
+        (size_t)B * A * sizeof(float), stream));
+    hipLaunchKernelGGL(cond_attn_bwd_dalpha, dim3(B, Ts), dim3(256), 0,
+                       stream, (const bf16_t*)ctx_bf.data_ptr(),
                        dctxpre_f32.data_ptr<float>(),
                        alphas_all.data_ptr<float>() + (long)t * B * Ts,
+                       daccA.data_ptr<float>(), mt,
+                       dalpha_p ? dalpha_p + (long)t * B * Ts : nullptr,
+                       dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
+                       B, Ts, C);
+    hipLaunchKernelGGL(cond_attn_bwd_scatter, dim3(B, cdiv_i(Ts, 256)),
+                       dim3(256), 3 * A * sizeof(float), stream,
+                       alphas_all.data_ptr<float>() + (long)t * B * Ts,
+                       dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
                        pctx.data_ptr<float>(),
                        pstate_all.data_ptr<float>() + (long)t * B * A,
                        accA_used.data_ptr<float>() + (long)t * B * Ts,
-                       Dwei.data_ptr<float>(), Uatt.data_ptr<float>(), mt,
-                       dalpha_p ? dalpha_p + (long)t * B * Ts : nullptr,
-                       daccA.data_ptr<float>(), dal_buf.data_ptr<float>(),
-                       dpctx_acc.data_ptr<float>(),
-                       dpc_buf.data_ptr<float>(), depc_buf.data_ptr<float>(),
-                       gdcatt.data_ptr<float>(), B, Ts, A, Apad, C);
-    // b5b: reduce over s -> dpstate / dD_wei / dU_att
-    hipLaunchKernelGGL(cond_attn_bwd_b, dim3(B), dim3(256), 0, stream,
-                       dpc_buf.data_ptr<float>(), depc_buf.data_ptr<float>(),
-                       accA_used.data_ptr<float>() + (long)t * B * Ts,
+                       Dwei.data_ptr<float>(), Uatt.data_ptr<float>(),
+                       daccA.data_ptr<float>(), dpctx_acc.data_ptr<float>(),
                        dpstate_all.data_ptr<float>() + (long)t * B * A,
-                       (bf16_t*)dstep_att.data_ptr(),
-                       gdDwei.data_ptr<float>(), gdUatt.data_ptr<float>(), B,
-                       Ts, A, Apad);
+                       gdDwei.data_ptr<float>(), gdUatt.data_ptr<float>(),
+                       gdcatt.data_ptr<float>(), B, Ts, A);
+    hipLaunchKernelGGL(cond_dpstate_cast, dim3(cdiv_i(B * A, 256)), dim3(256),
+                       0, stream,
+                       dpstate_all.data_ptr<float>() + (long)t * B * A,
+                       (bf16_t*)dstep_att.data_ptr(), Apad32, B, A);
     // b2: dh1 += [dpr2|dpu2|dpxa_lin] @ [U_1|Ux_1]^T + passthrough
     hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH), dim3(384), 0,
                        stream, (const bf16_t*)dstep1.data_ptr(),

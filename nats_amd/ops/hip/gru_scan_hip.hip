@@ -62,11 +62,7 @@ __global__ __launch_bounds__(384) void nats_gru_step_fwd(
 
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   const bf16_t* brow = Upk + (long)(wg * 3 + g) * JB * Hpad;
-  for (int k = 0; k < Hpad; k += 32) {
-    bf16x8 a = frag_a_rowmajor(h_bf, 16 * m, Hpad, k);
-    bf16x8 b = frag_bt_rowmajor(brow, 0, Hpad, k);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
-  }
+  NATS_MFMA_KLOOP(acc, h_bf, 16 * m, Hpad, brow, 0, Hpad, 0, Hpad);
   {
     const int lane = threadIdx.x & (NATS_WAVE - 1);
     const int col = lane & 15;
@@ -169,11 +165,7 @@ __global__ __launch_bounds__(384) void nats_gru_step_bwd_gemm(
   const int kend = min(Kpad, (ks + 1) * kchunk);
 
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  for (int k = kbeg; k < kend; k += 32) {
-    bf16x8 a = frag_a_rowmajor(dstep, 16 * m, Kpad, k);
-    bf16x8 b = frag_bt_rowmajor(Wt, i0, Kpad, k);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
-  }
+  NATS_MFMA_KLOOP(acc, dstep, 16 * m, Kpad, Wt, i0, Kpad, kbeg, kend);
   {
     const int lane = threadIdx.x & (NATS_WAVE - 1);
     const int col = lane & 15;
