@@ -67,6 +67,17 @@ def gru_scan(x_gates, x_cand, mask, U, Ux, h0=None):
     return eager.gru_scan(x_gates, x_cand, mask, U, Ux, h0)
 
 
+def gru_scan_bidir(xg0, xc0, mask0, U0, Ux0, xg1, xc1, mask1, U1, Ux1):
+    """Both encoder directions (independent scans) in one fused launch
+    sequence on GPU; two eager scans on CPU. Returns (h_fwd, h_bwd)."""
+    if _use_hip(xg0, U0):
+        from .gru import gru_scan_bidir_hip
+        return gru_scan_bidir_hip(xg0, xc0, mask0, U0, Ux0, xg1, xc1, mask1,
+                                  U1, Ux1)
+    return (eager.gru_scan(xg0, xc0, mask0, U0, Ux0),
+            eager.gru_scan(xg1, xc1, mask1, U1, Ux1))
+
+
 def cond_gru_scan(y_gates, y_cand, mask, init_state, ctx, ctx_mask, pctx, P):
     """Conditional-GRU decoder scan (training). Returns
     (h2s, ctxs, alphas, acc_ctx, acc_alpha).

@@ -51,6 +51,7 @@ class GRUScanFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, xg, xc, mask, U, Ux, h0):
         ext = _hip_ext()
+        xg_dtype = xg.dtype
         xg = xg.to(torch.bfloat16).contiguous()
         xc = xc.to(torch.bfloat16).contiguous()
         Upk = pack_fwd_weights(U, Ux)
@@ -58,6 +59,7 @@ class GRUScanFn(torch.autograd.Function):
         ctx.save_for_backward(xc, h_all, saved, U, Ux,
                               mask if mask is not None else torch.empty(0),
                               h0 if h0 is not None else torch.empty(0))
+        ctx.xg_dtype = xg_dtype
         return h_all
 
     @staticmethod
@@ -83,9 +85,7 @@ class GRUScanFn(torch.autograd.Function):
         dU = (flat_h.t() @ dpre_all[..., :2 * H].reshape(T * B, 2 * H)).float()
         dUx = (flat_h.t() @ dpre_all[..., 3 * H:].reshape(T * B, H)).float()
         need_h0 = ctx.needs_input_grad[5]
-        return (dxg.to(U.dtype) if U.dtype != torch.bfloat16 else dxg,
-                dxc.to(U.dtype) if U.dtype != torch.bfloat16 else dxc,
-                None,
+        return (dxg.to(ctx.xg_dtype), dxc.to(ctx.xg_dtype), None,
                 dU.to(U.dtype), dUx.to(Ux.dtype),
                 dh0 if need_h0 else None)
 
@@ -94,3 +94,64 @@ def gru_scan_hip(x_gates, x_cand, mask, U, Ux, h0=None):
     if mask is not None:
         mask = mask.float().contiguous()
     return GRUScanFn.apply(x_gates, x_cand, mask, U, Ux, h0)
+
+
+class BidirGRUScanFn(torch.autograd.Function):
+    """Both encoder directions in one fused launch sequence (h0 = 0)."""
+
+    @staticmethod
+    def forward(ctx, xg0, xc0, mask0, U0, Ux0, xg1, xc1, mask1, U1, Ux1):
+        ext = _hip_ext()
+        ctx.xg_dtype = xg0.dtype
+        xg0 = xg0.to(torch.bfloat16).contiguous()
+        xc0 = xc0.to(torch.bfloat16).contiguous()
+        xg1 = xg1.to(torch.bfloat16).contiguous()
+        xc1 = xc1.to(torch.bfloat16).contiguous()
+        h_all0, saved0, h_all1, saved1 = ext.gru_scan_fwd_bidir(
+            xg0, xc0, mask0, pack_fwd_weights(U0, Ux0),
+            xg1, xc1, mask1, pack_fwd_weights(U1, Ux1))
+        e = torch.empty(0)
+        ctx.save_for_backward(xc0, h_all0, saved0, U0, Ux0,
+                              mask0 if mask0 is not None else e,
+                              xc1, h_all1, saved1, U1, Ux1,
+                              mask1 if mask1 is not None else e)
+        return h_all0, h_all1
+
+    @staticmethod
+    def backward(ctx, dh0_out, dh1_out):
+        ext = _hip_ext()
+        (xc0, h_all0, saved0, U0, Ux0, mask0,
+         xc1, h_all1, saved1, U1, Ux1, mask1) = ctx.saved_tensors
+        mask0 = mask0 if mask0.numel() else None
+        mask1 = mask1 if mask1.numel() else None
+        T, B, H = h_all0.shape
+        dpre0, dh00, dpre1, dh01 = ext.gru_scan_bwd_bidir(
+            dh0_out.contiguous().float(), h_all0, saved0, xc0, mask0,
+            pack_bwd_weights(U0, Ux0),
+            dh1_out.contiguous().float(), h_all1, saved1, xc1, mask1,
+            pack_bwd_weights(U1, Ux1))
+
+        def wgrads(h_all, dpre):
+            h_prev = torch.cat([torch.zeros_like(h_all[:1]), h_all[:-1]], 0)
+            fh = h_prev.reshape(T * B, H).to(torch.bfloat16)
+            dU = (fh.t() @ dpre[..., :2 * H].reshape(T * B, 2 * H)).float()
+            dUx = (fh.t() @ dpre[..., 3 * H:].reshape(T * B, H)).float()
+            return dU, dUx
+
+        dU0, dUx0 = wgrads(h_all0, dpre0)
+        dU1, dUx1 = wgrads(h_all1, dpre1)
+        xt = ctx.xg_dtype
+        pt = U0.dtype
+        return (dpre0[..., :2 * H].to(xt), dpre0[..., 2 * H:3 * H].to(xt),
+                None, dU0.to(pt), dUx0.to(pt),
+                dpre1[..., :2 * H].to(xt), dpre1[..., 2 * H:3 * H].to(xt),
+                None, dU1.to(pt), dUx1.to(pt))
+
+
+def gru_scan_bidir_hip(xg0, xc0, mask0, U0, Ux0, xg1, xc1, mask1, U1, Ux1):
+    if mask0 is not None:
+        mask0 = mask0.float().contiguous()
+    if mask1 is not None:
+        mask1 = mask1.float().contiguous()
+    return BidirGRUScanFn.apply(xg0, xc0, mask0, U0, Ux0, xg1, xc1, mask1,
+                                U1, Ux1)

@@ -12,6 +12,16 @@ std::vector<torch::Tensor> gru_scan_bwd(torch::Tensor dh_out,
                                         c10::optional<torch::Tensor> mask,
                                         torch::Tensor Ubwd,
                                         c10::optional<torch::Tensor> h0);
+std::vector<torch::Tensor> gru_scan_fwd_bidir(
+    torch::Tensor xg0, torch::Tensor xc0, c10::optional<torch::Tensor> mask0,
+    torch::Tensor Upk0, torch::Tensor xg1, torch::Tensor xc1,
+    c10::optional<torch::Tensor> mask1, torch::Tensor Upk1);
+std::vector<torch::Tensor> gru_scan_bwd_bidir(
+    torch::Tensor dh_out0, torch::Tensor h_all0, torch::Tensor saved0,
+    torch::Tensor xc0, c10::optional<torch::Tensor> mask0,
+    torch::Tensor Ubwd0, torch::Tensor dh_out1, torch::Tensor h_all1,
+    torch::Tensor saved1, torch::Tensor xc1,
+    c10::optional<torch::Tensor> mask1, torch::Tensor Ubwd1);
 std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
                                           torch::Tensor targets);
 torch::Tensor softmax_ce_bwd(torch::Tensor logits, torch::Tensor targets,
@@ -42,6 +52,10 @@ std::vector<torch::Tensor> cond_gru_bwd(
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gru_scan_fwd", &gru_scan_fwd, "fused GRU scan forward");
   m.def("gru_scan_bwd", &gru_scan_bwd, "fused GRU scan backward");
+  m.def("gru_scan_fwd_bidir", &gru_scan_fwd_bidir,
+        "bidirectional fused GRU scan forward");
+  m.def("gru_scan_bwd_bidir", &gru_scan_bwd_bidir,
+        "bidirectional fused GRU scan backward");
   m.def("softmax_ce_fwd", &softmax_ce_fwd, "fused softmax+CE forward");
   m.def("softmax_ce_bwd", &softmax_ce_bwd, "fused softmax+CE backward");
   m.def("mfma_gemm_bt", &mfma_gemm_bt, "MFMA layout self-test GEMM");

@@ -140,6 +140,126 @@ __global__ void nats_gru_step_bwd_pointwise(
   }
 }
 
+// ---------------- bidirectional forward step ----------------
+__device__ __forceinline__ void gru_fwd_body(const GruFwdArgs& a,
+                                             int ld_bfout, int B, int H,
+                                             int Hpad, int wg) {
+  __shared__ float pre[3][32][JB + 1];
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int m = wave / 3;
+  const int g = wave % 3;
+  const int j0 = wg * JB;
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const bf16_t* brow = a.Upk + (long)(wg * 3 + g) * JB * Hpad;
+  NATS_MFMA_KLOOP(acc, a.h_bf, 16 * m, Hpad, brow, 0, Hpad, 0, Hpad);
+  {
+    const int lane = threadIdx.x & (NATS_WAVE - 1);
+    const int col = lane & 15;
+    const int rbase = 16 * m + (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) pre[g][rbase + i][col] = acc[i];
+  }
+  __syncthreads();
+
+  for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+    const int b = idx / JB;
+    const int c = idx % JB;
+    const int j = j0 + c;
+    if (j >= H) continue;
+    const float hp = a.h_prev[(long)b * H + j];
+    const float pr = pre[0][b][c] + (float)a.xg_t[(long)b * 2 * H + j];
+    const float pu = pre[1][b][c] + (float)a.xg_t[(long)b * 2 * H + H + j];
+    const float px = pre[2][b][c];
+    const float r = nats_sigmoid(pr);
+    const float u = nats_sigmoid(pu);
+    const float hbar = tanhf(px * r + (float)a.xc_t[(long)b * H + j]);
+    float hnew = u * hp + (1.f - u) * hbar;
+    if (a.mask_t != nullptr) {
+      const float mm = a.mask_t[b];
+      hnew = mm * hnew + (1.f - mm) * hp;
+    }
+    a.h_out[(long)b * H + j] = hnew;
+    a.h_bf_out[(long)b * ld_bfout + j] = (bf16_t)hnew;
+    a.saved_t[(long)b * 3 * H + j] = (bf16_t)r;
+    a.saved_t[(long)b * 3 * H + H + j] = (bf16_t)u;
+    a.saved_t[(long)b * 3 * H + 2 * H + j] = (bf16_t)px;
+  }
+}
+
+__global__ __launch_bounds__(384) void nats_gru_step_fwd_bidir(
+    GruFwdArgs a0, GruFwdArgs a1, int ld_bfout, int B, int H, int Hpad) {
+  gru_fwd_body(blockIdx.y == 0 ? a0 : a1, ld_bfout, B, H, Hpad, blockIdx.x);
+}
+
+// ---------------- fused backward step (gemm of t+1's dstep + pointwise)
+__device__ __forceinline__ void gru_bwd_fused_body(const GruBwdArgs& a,
+                                                   const bf16_t* Ubwd, int B,
+                                                   int H, int Kpad, int wg) {
+  __shared__ float part[3][32][JB + 1];
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int m = wave / 3;
+  const int ks = wave % 3;
+  const int i0 = wg * JB;
+
+  const int kchunk = ((Kpad / 3 + 31) / 32) * 32;
+  const int kbeg = ks * kchunk;
+  const int kend = min(Kpad, (ks + 1) * kchunk);
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  NATS_MFMA_KLOOP(acc, a.dstep_in, 16 * m, Kpad, Ubwd, i0, Kpad, kbeg, kend);
+  {
+    const int lane = threadIdx.x & (NATS_WAVE - 1);
+    const int col = lane & 15;
+    const int rbase = 16 * m + (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) part[ks][rbase + i][col] = acc[i];
+  }
+  __syncthreads();
+
+  for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+    const int b = idx / JB;
+    const int c = idx % JB;
+    const int j = i0 + c;
+    if (j >= H) continue;
+    const long bj = (long)b * H + j;
+    // dh at step t for column j (recurrent + passthrough + upstream)
+    float dh = a.ddirect_in[bj] + part[0][b][c] + part[1][b][c] +
+               part[2][b][c];
+    if (a.dh_out_t != nullptr) dh += a.dh_out_t[bj];
+    const float r = (float)a.saved_t[(long)b * 3 * H + j];
+    const float u = (float)a.saved_t[(long)b * 3 * H + H + j];
+    const float px = (float)a.saved_t[(long)b * 3 * H + 2 * H + j];
+    const float hbar = tanhf(px * r + (float)a.xc_t[bj]);
+    const float hp = a.h_prev[bj];
+    const float mm = (a.mask_t != nullptr) ? a.mask_t[b] : 1.f;
+    const float du = dh * mm * (hp - hbar);
+    const float dhbar = dh * mm * (1.f - u);
+    const float dpx = dhbar * (1.f - hbar * hbar);
+    const float dpxl = dpx * r;
+    const float dr = dpx * px;
+    const float dpr = dr * r * (1.f - r);
+    const float dpu = du * u * (1.f - u);
+    a.ddirect_out[bj] = dh * (mm * u + (1.f - mm));
+    a.dstep_out[(long)b * Kpad + j] = (bf16_t)dpr;
+    a.dstep_out[(long)b * Kpad + H + j] = (bf16_t)dpu;
+    a.dstep_out[(long)b * Kpad + 2 * H + j] = (bf16_t)dpxl;
+    a.dpre_t[(long)b * 4 * H + j] = (bf16_t)dpr;
+    a.dpre_t[(long)b * 4 * H + H + j] = (bf16_t)dpu;
+    a.dpre_t[(long)b * 4 * H + 2 * H + j] = (bf16_t)dpx;
+    a.dpre_t[(long)b * 4 * H + 3 * H + j] = (bf16_t)dpxl;
+  }
+}
+
+__global__ __launch_bounds__(384) void nats_gru_step_bwd_fused_bidir(
+    GruBwdArgs a0, GruBwdArgs a1, const bf16_t* Ubwd0, const bf16_t* Ubwd1,
+    int B, int H, int Kpad) {
+  if (blockIdx.y == 0)
+    gru_bwd_fused_body(a0, Ubwd0, B, H, Kpad, blockIdx.x);
+  else
+    gru_bwd_fused_body(a1, Ubwd1, B, H, Kpad, blockIdx.x);
+}
+
 // ---------------- backward recurrent GEMM ----------------
 // out[b, i] = ddirect[b, i] + sum_k dstep[b, k] * Wt[i, k]
 // Generic over the output width (rows of Wt): the encoder/GRU_2 call has
@@ -245,6 +365,153 @@ std::vector<torch::Tensor> gru_scan_fwd(torch::Tensor xg, torch::Tensor xc,
   }
   HIP_CHECK(hipGetLastError());
   return {h_all, saved};
+}
+
+// Bidirectional encoder forward: one launch per timestep covers both
+// directions (grid.y = direction). h0 = 0 (nats.py:360).
+std::vector<torch::Tensor> gru_scan_fwd_bidir(
+    torch::Tensor xg0, torch::Tensor xc0, c10::optional<torch::Tensor> mask0,
+    torch::Tensor Upk0, torch::Tensor xg1, torch::Tensor xc1,
+    c10::optional<torch::Tensor> mask1, torch::Tensor Upk1) {
+  const int T = xg0.size(0), B = xg0.size(1), H = xc0.size(2);
+  TORCH_CHECK(B <= 32, "gru_scan: batch per step must be <= 32");
+  const int Hpad = Upk0.size(1);
+  const int ngrp = cdiv(H, JB);
+
+  auto optsF = xg0.options().dtype(torch::kFloat32);
+  auto optsB = xg0.options();
+  auto h_all0 = torch::empty({T, B, H}, optsF);
+  auto h_all1 = torch::empty({T, B, H}, optsF);
+  auto saved0 = torch::empty({T, B, 3 * H}, optsB);
+  auto saved1 = torch::empty({T, B, 3 * H}, optsB);
+  auto h_bf = torch::zeros({2, 2, 32, Hpad}, optsB);  // [dir][pingpong]
+  auto h00 = torch::zeros({B, H}, optsF);
+
+  const float* m0 = nullptr;
+  const float* m1 = nullptr;
+  torch::Tensor m0c, m1c;
+  if (mask0.has_value()) {
+    m0c = mask0->contiguous().to(torch::kFloat32);
+    m0 = m0c.data_ptr<float>();
+  }
+  if (mask1.has_value()) {
+    m1c = mask1->contiguous().to(torch::kFloat32);
+    m1 = m1c.data_ptr<float>();
+  }
+
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  const long hb = (long)32 * Hpad;
+  bf16_t* hbf = (bf16_t*)h_bf.data_ptr();
+  for (int t = 0; t < T; ++t) {
+    GruFwdArgs a0{
+        hbf + 0 * 2 * hb + (t % 2) * hb,
+        (t == 0) ? h00.data_ptr<float>()
+                 : h_all0.data_ptr<float>() + (long)(t - 1) * B * H,
+        (const bf16_t*)Upk0.data_ptr(),
+        (const bf16_t*)xg0.data_ptr() + (long)t * B * 2 * H,
+        (const bf16_t*)xc0.data_ptr() + (long)t * B * H,
+        m0 ? m0 + (long)t * B : nullptr,
+        h_all0.data_ptr<float>() + (long)t * B * H,
+        hbf + 0 * 2 * hb + ((t + 1) % 2) * hb,
+        (bf16_t*)saved0.data_ptr() + (long)t * B * 3 * H};
+    GruFwdArgs a1{
+        hbf + 1 * 2 * hb + (t % 2) * hb,
+        (t == 0) ? h00.data_ptr<float>()
+                 : h_all1.data_ptr<float>() + (long)(t - 1) * B * H,
+        (const bf16_t*)Upk1.data_ptr(),
+        (const bf16_t*)xg1.data_ptr() + (long)t * B * 2 * H,
+        (const bf16_t*)xc1.data_ptr() + (long)t * B * H,
+        m1 ? m1 + (long)t * B : nullptr,
+        h_all1.data_ptr<float>() + (long)t * B * H,
+        hbf + 1 * 2 * hb + ((t + 1) % 2) * hb,
+        (bf16_t*)saved1.data_ptr() + (long)t * B * 3 * H};
+    hipLaunchKernelGGL(nats_gru_step_fwd_bidir, dim3(ngrp, 2), dim3(384), 0,
+                       stream, a0, a1, Hpad, B, H, Hpad);
+  }
+  HIP_CHECK(hipGetLastError());
+  return {h_all0, saved0, h_all1, saved1};
+}
+
+// Bidirectional fused backward: one launch per timestep (both directions,
+// gemm-of-previous-dstep + pointwise fused, SURVEY §2.4 backward of K3).
+std::vector<torch::Tensor> gru_scan_bwd_bidir(
+    torch::Tensor dh_out0, torch::Tensor h_all0, torch::Tensor saved0,
+    torch::Tensor xc0, c10::optional<torch::Tensor> mask0,
+    torch::Tensor Ubwd0, torch::Tensor dh_out1, torch::Tensor h_all1,
+    torch::Tensor saved1, torch::Tensor xc1,
+    c10::optional<torch::Tensor> mask1, torch::Tensor Ubwd1) {
+  const int T = dh_out0.size(0), B = dh_out0.size(1), H = dh_out0.size(2);
+  const int K3pad = Ubwd0.size(1);
+  const int ngrp = cdiv(H, JB);
+
+  auto optsF = dh_out0.options();
+  auto optsB = dh_out0.options().dtype(torch::kBFloat16);
+  auto dpre0 = torch::empty({T, B, 4 * H}, optsB);
+  auto dpre1 = torch::empty({T, B, 4 * H}, optsB);
+  auto ddir = torch::zeros({2, B, H}, optsF);
+  auto dstep = torch::zeros({2, 2, 32, K3pad}, optsB);  // [dir][pingpong]
+  auto h00 = torch::zeros({B, H}, optsF);
+  auto dh0_0 = torch::empty({B, H}, optsF);
+  auto dh0_1 = torch::empty({B, H}, optsF);
+
+  const float* m0 = nullptr;
+  const float* m1 = nullptr;
+  torch::Tensor m0c, m1c;
+  if (mask0.has_value()) {
+    m0c = mask0->contiguous().to(torch::kFloat32);
+    m0 = m0c.data_ptr<float>();
+  }
+  if (mask1.has_value()) {
+    m1c = mask1->contiguous().to(torch::kFloat32);
+    m1 = m1c.data_ptr<float>();
+  }
+
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  const long ds = (long)32 * K3pad;
+  bf16_t* dsp = (bf16_t*)dstep.data_ptr();
+  float* dd = ddir.data_ptr<float>();
+  auto dh0c = dh_out0.contiguous().to(torch::kFloat32);
+  auto dh1c = dh_out1.contiguous().to(torch::kFloat32);
+  for (int t = T - 1; t >= 0; --t) {
+    // parity: step t reads dstep[(t+1)%2], writes dstep[t%2]
+    GruBwdArgs a0{
+        dsp + 0 * 2 * ds + ((t + 1) % 2) * ds,
+        dd + 0,
+        dh0c.data_ptr<float>() + (long)t * B * H,
+        (const bf16_t*)saved0.data_ptr() + (long)t * B * 3 * H,
+        (const bf16_t*)xc0.data_ptr() + (long)t * B * H,
+        (t == 0) ? h00.data_ptr<float>()
+                 : h_all0.data_ptr<float>() + (long)(t - 1) * B * H,
+        m0 ? m0 + (long)t * B : nullptr,
+        dsp + 0 * 2 * ds + (t % 2) * ds,
+        dd + 0,
+        (bf16_t*)dpre0.data_ptr() + (long)t * B * 4 * H};
+    GruBwdArgs a1{
+        dsp + 1 * 2 * ds + ((t + 1) % 2) * ds,
+        dd + (long)B * H,
+        dh1c.data_ptr<float>() + (long)t * B * H,
+        (const bf16_t*)saved1.data_ptr() + (long)t * B * 3 * H,
+        (const bf16_t*)xc1.data_ptr() + (long)t * B * H,
+        (t == 0) ? h00.data_ptr<float>()
+                 : h_all1.data_ptr<float>() + (long)(t - 1) * B * H,
+        m1 ? m1 + (long)t * B : nullptr,
+        dsp + 1 * 2 * ds + (t % 2) * ds,
+        dd + (long)B * H,
+        (bf16_t*)dpre1.data_ptr() + (long)t * B * 4 * H};
+    hipLaunchKernelGGL(nats_gru_step_bwd_fused_bidir, dim3(ngrp, 2),
+                       dim3(384), 0, stream, a0, a1,
+                       (const bf16_t*)Ubwd0.data_ptr(),
+                       (const bf16_t*)Ubwd1.data_ptr(), B, H, K3pad);
+  }
+  // dh0 = ddirect(0) + dstep(0) @ Ubwd^T (one plain gemm per direction)
+  hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrp), dim3(384), 0, stream,
+                     dsp + 0 * 2 * ds + 0 * ds, (const bf16_t*)Ubwd0.data_ptr(),
+                     dd + 0, dh0_0.data_ptr<float>(), B, H, K3pad);
+  hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrp), dim3(384), 0, stream,
+                     dsp + 1 * 2 * ds + 0 * ds, (const bf16_t*)Ubwd1.data_ptr(),
+                     dd + (long)B * H, dh0_1.data_ptr<float>(), B, H, K3pad);
+  HIP_CHECK(hipGetLastError());
+  return {dpre0, dh0_0, dpre1, dh0_1};
 }
 
 std::vector<torch::Tensor> gru_scan_bwd(torch::Tensor dh_out,
