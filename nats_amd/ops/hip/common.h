@@ -57,23 +57,33 @@ __device__ __forceinline__ void store_cd_rowmajor(float* base, const f32x4& d,
   for (int i = 0; i < 4; ++i) base[(long)(rbase + i) * ld + col] = d[i];
 }
 
-// 1-deep software-pipelined MFMA K-loop: issue next iteration's fragment
-// loads before the current MFMA so L2 latency hides under the matrix op.
+// 2-deep software-pipelined MFMA K-loop: two iterations' fragment loads
+// stay in flight ahead of each MFMA so L2 latency (~300cy for the 16
+// scattered 16B lines per fragment) hides under the matrix ops.
 // (The plain loop measured ~13.9us/step on gfx950 — latency-bound.)
 #define NATS_MFMA_KLOOP(ACC, APTR, AROW, ALD, BPTR, BROW, BLD, KBEG, KEND)   \
   do {                                                                       \
     int _k = (KBEG);                                                         \
-    if (_k < (KEND)) {                                                       \
+    const int _ke = (KEND);                                                  \
+    if (_k + 32 >= _ke) {                                                    \
+      if (_k < _ke) {                                                        \
+        bf16x8 _a0 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
+        bf16x8 _b0 = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k);            \
+        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0);\
+      }                                                                      \
+    } else {                                                                 \
       bf16x8 _a0 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);               \
       bf16x8 _b0 = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k);              \
-      for (_k += 32; _k < (KEND); _k += 32) {                                \
-        bf16x8 _a1 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
-        bf16x8 _b1 = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k);            \
+      bf16x8 _a1 = frag_a_rowmajor((APTR), (AROW), (ALD), _k + 32);          \
+      bf16x8 _b1 = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k + 32);         \
+      for (_k += 64; _k < _ke; _k += 32) {                                   \
+        bf16x8 _a2 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
+        bf16x8 _b2 = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k);            \
         ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0);\
-        _a0 = _a1;                                                           \
-        _b0 = _b1;                                                           \
+        _a0 = _a1; _b0 = _b1; _a1 = _a2; _b1 = _b2;                          \
       }                                                                      \
       ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0); \
+      ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a1, _b1, ACC, 0, 0, 0); \
     }                                                                        \
   } while (0)
 

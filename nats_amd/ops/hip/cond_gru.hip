@@ -349,7 +349,7 @@ __global__ void cond_gate_bwd(const float* __restrict__ dctx_buf,
   }
 }
 
-// attention backward, stage 1 (one WG per (b,s)): dalpha[s,b] =
+// attention backward, stage 1 (one WAVE per (b,s)): dalpha[s,b] =
 // ctx[s,b,:] . dctxpre[b,:] (+ acc-chain and upstream alpha grads) and the
 // softmax-backward dot partial, accumulated by atomics into dot_buf.
 __global__ __launch_bounds__(256) void cond_attn_bwd_dalpha(
@@ -363,13 +363,15 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_dalpha(
     float* __restrict__ dot_buf,            // [B] (zeroed per step)
     int B, int Ts, int C) {
   const int b = blockIdx.x;
-  const int s = blockIdx.y;
-  __shared__ float red[256 / NATS_WAVE];
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int lane = threadIdx.x & (NATS_WAVE - 1);
+  const int s = blockIdx.y * 4 + wave;
+  if (s >= Ts) return;
   const bf16_t* crow = ctx_bf + ((long)s * B + b) * C;
   const float* drow = dctxpre_f32 + (long)b * C;
   float part = 0.f;
   const int C8 = C & ~7;
-  for (int c = threadIdx.x * 8; c < C8; c += blockDim.x * 8) {
+  for (int c = lane * 8; c < C8; c += NATS_WAVE * 8) {
     bf16x8 v = *(const bf16x8*)(crow + c);
     const float4 d0 = *(const float4*)(drow + c);
     const float4 d1 = *(const float4*)(drow + c + 4);
@@ -377,18 +379,14 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_dalpha(
             (float)v[3] * d0.w + (float)v[4] * d1.x + (float)v[5] * d1.y +
             (float)v[6] * d1.z + (float)v[7] * d1.w;
   }
-  for (int c = C8 + threadIdx.x; c < C; c += blockDim.x) {
+  for (int c = C8 + lane; c < C; c += NATS_WAVE) {
     part += (float)crow[c] * drow[c];
   }
 #pragma unroll
   for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
     part += __shfl_down(part, off);
-  if ((threadIdx.x & (NATS_WAVE - 1)) == 0)
-    red[threadIdx.x / NATS_WAVE] = part;
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    float dal = 0.f;
-    for (int w = 0; w < (int)blockDim.x / NATS_WAVE; ++w) dal += red[w];
+  if (lane == 0) {
+    float dal = part;
     const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
     dal += mm * daccA[(long)b * Ts + s];
     if (dalphas_t != nullptr) dal += dalphas_t[(long)b * Ts + s];
@@ -398,8 +396,8 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_dalpha(
 }
 
 // attention backward, stage 2 (grid (B, ceil(Ts/256))): softmax backward,
-// per-(s,i) dpc with wave-reduced accumulation of dpstate/dD_wei/dU_att
-// (no per-(s,i) scratch), and the acc_alpha gradient carry.
+// per-(s,i) dpc into dpctx_acc + acc_alpha carry; tanh values stored to
+// pc_buf [B][A][Ts] (coalesced in s) for the stage-3 reduction.
 __global__ __launch_bounds__(256) void cond_attn_bwd_scatter(
     const float* __restrict__ alphas_t,     // [B][Ts]
     const float* __restrict__ dal_buf,      // [Ts][B]
@@ -410,62 +408,58 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_scatter(
     const float* __restrict__ Dwei, const float* __restrict__ Uatt,
     float* __restrict__ daccA,              // [B][Ts] in/out
     float* __restrict__ dpctx_acc,          // [Ts][B][A] (+=)
-    float* __restrict__ dpstate_t,          // [B][A] (zeroed, atomic +=)
+    bf16_t* __restrict__ pc_buf,            // [B][A][Ts]
+    int B, int Ts, int A) {
+  const int b = blockIdx.x;
+  const int s = blockIdx.y * blockDim.x + threadIdx.x;
+  if (s >= Ts) return;
+  const float al = alphas_t[(long)b * Ts + s];
+  const float de = al * (dal_buf[(long)s * B + b] - dot_buf[b]);
+  const float accAu = accA_used_t[(long)b * Ts + s];
+  const float* prow = pctx + ((long)s * B + b) * A;
+  const float* srow = pstate_t + (long)b * A;
+  float* dprow = dpctx_acc + ((long)s * B + b) * A;
+  float daccA_add = 0.f;
+  for (int i = 0; i < A; ++i) {
+    const float pc = tanhf(prow[i] + srow[i] + accAu * Dwei[i]);
+    const float dpc = de * (1.f - pc * pc) * Uatt[i];
+    dprow[i] += dpc;
+    daccA_add += dpc * Dwei[i];
+    pc_buf[((long)b * A + i) * Ts + s] = (bf16_t)pc;
+  }
+  daccA[(long)b * Ts + s] += daccA_add;
+}
+
+// attention backward, stage 3 (one WG per b): reduce over s (contiguous
+// pc_buf rows) -> dpstate / dD_wei / dU_att / dc_att. de and accA_used
+// staged in LDS once per block.
+__global__ __launch_bounds__(256) void cond_attn_bwd_reduce(
+    const float* __restrict__ alphas_t,     // [B][Ts]
+    const float* __restrict__ dal_buf,      // [Ts][B]
+    const float* __restrict__ dot_buf,      // [B]
+    const float* __restrict__ accA_used_t,  // [B][Ts]
+    const bf16_t* __restrict__ pc_buf,      // [B][A][Ts]
+    const float* __restrict__ Uatt,
+    float* __restrict__ dpstate_t,          // [B][A]
+    bf16_t* __restrict__ dstep_att,         // [32][Apad]
     float* __restrict__ gdDwei,             // [A] (atomic)
     float* __restrict__ gdUatt,             // [A] (atomic)
     float* __restrict__ gdcatt,             // [1] (atomic)
-    int B, int Ts, int A) {
+    int B, int Ts, int A, int Apad) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* sm_dps = (float*)smem_raw;            // [A]
-  float* sm_ddw = sm_dps + A;                  // [A]
-  float* sm_dua = sm_ddw + A;                  // [A]
+  float* sm_de = (float*)smem_raw;      // [Ts]
+  float* sm_au = sm_de + Ts;            // [Ts]
   __shared__ float red[256 / NATS_WAVE];
   const int b = blockIdx.x;
-  const int s = blockIdx.y * blockDim.x + threadIdx.x;
-  const bool live = (s < Ts);
-  for (int i = threadIdx.x; i < 3 * A; i += blockDim.x) sm_dps[i] = 0.f;
-  __syncthreads();
-
-  float de = 0.f, accAu = 0.f;
-  const float* prow = nullptr;
-  float* dprow = nullptr;
-  if (live) {
-    const float al = alphas_t[(long)b * Ts + s];
-    de = al * (dal_buf[(long)s * B + b] - dot_buf[b]);
-    accAu = accA_used_t[(long)b * Ts + s];
-    prow = pctx + ((long)s * B + b) * A;
-    dprow = dpctx_acc + ((long)s * B + b) * A;
+  const float dot = dot_buf[b];
+  float dc = 0.f;
+  for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
+    const float de = alphas_t[(long)b * Ts + s] *
+                     (dal_buf[(long)s * B + b] - dot);
+    sm_de[s] = de;
+    sm_au[s] = accA_used_t[(long)b * Ts + s];
+    dc += de;
   }
-  const float* srow = pstate_t + (long)b * A;
-  const int lane = threadIdx.x & (NATS_WAVE - 1);
-  float daccA_add = 0.f;
-  for (int i = 0; i < A; ++i) {
-    float dpc = 0.f, v2 = 0.f, v3 = 0.f;
-    if (live) {
-      const float pc = tanhf(prow[i] + srow[i] + accAu * Dwei[i]);
-      dpc = de * (1.f - pc * pc) * Uatt[i];
-      dprow[i] += dpc;
-      daccA_add += dpc * Dwei[i];
-      v2 = dpc * accAu;
-      v3 = de * pc;
-    }
-    float v1 = dpc;
-#pragma unroll
-    for (int off = NATS_WAVE / 2; off > 0; off >>= 1) {
-      v1 += __shfl_down(v1, off);
-      v2 += __shfl_down(v2, off);
-      v3 += __shfl_down(v3, off);
-    }
-    if (lane == 0) {
-      atomicAdd(&sm_dps[i], v1);
-      atomicAdd(&sm_ddw[i], v2);
-      atomicAdd(&sm_dua[i], v3);
-    }
-  }
-  if (live) daccA[(long)b * Ts + s] += daccA_add;
-
-  // dcatt = sum de
-  float dc = de;
 #pragma unroll
   for (int off = NATS_WAVE / 2; off > 0; off >>= 1) dc += __shfl_down(dc, off);
   if ((threadIdx.x & (NATS_WAVE - 1)) == 0) red[threadIdx.x / NATS_WAVE] = dc;
@@ -476,22 +470,21 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_scatter(
     atomicAdd(gdcatt, S);
   }
   for (int i = threadIdx.x; i < A; i += blockDim.x) {
-    atomicAdd(&dpstate_t[(long)b * A + i], sm_dps[i]);
-    atomicAdd(&gdDwei[i], sm_ddw[i]);
-    atomicAdd(&gdUatt[i], sm_dua[i]);
-  }
-}
-
-// bf16 copy of dpstate into the padded GEMM operand
-__global__ void cond_dpstate_cast(const float* __restrict__ dpstate_t,
-                                  bf16_t* __restrict__ dstep_att, int Apad,
-                                  int B, int A) {
-  const long total = (long)B * A;
-  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    const int b = idx / A;
-    const int i = idx % A;
-    dstep_att[(long)b * Apad + i] = (bf16_t)dpstate_t[idx];
+    const bf16_t* prow = pc_buf + ((long)b * A + i) * Ts;
+    const float ua = Uatt[i];
+    float sps = 0.f, sdw = 0.f, sua = 0.f;
+    for (int s = 0; s < Ts; ++s) {
+      const float pc = (float)prow[s];
+      const float de = sm_de[s];
+      const float dpc = de * (1.f - pc * pc) * ua;
+      sps += dpc;
+      sdw += dpc * sm_au[s];
+      sua += de * pc;
+    }
+    dpstate_t[(long)b * A + i] = sps;
+    dstep_att[(long)b * Apad + i] = (bf16_t)sps;
+    atomicAdd(&gdDwei[i], sdw);
+    atomicAdd(&gdUatt[i], sua);
   }
 }
 
@@ -682,6 +675,8 @@ std::vector<torch::Tensor> cond_gru_bwd(
                                    : torch::zeros({B, C}, optsF);
   auto dal_buf = torch::empty({Ts, B}, optsF);
   auto dot_buf = torch::empty({B}, optsF);
+  auto pc_buf = torch::empty({B, A, Ts},
+                             dh2_all.options().dtype(torch::kBFloat16));
   const int Apad = Apad32;
   auto dstep1 = torch::zeros({32, K3Hpad}, optsB);
   auto dstepC = torch::zeros({32, K3Hpad}, optsB);
@@ -712,7 +707,6 @@ std::vector<torch::Tensor> cond_gru_bwd(
   auto dh2_c = dh2_all.contiguous().to(torch::kFloat32);
   const int pwH = (int)std::min<long>(512, (((long)B * H) + 255) / 256);
   const int pwC = (int)std::min<long>(512, (((long)B * C) + 255) / 256);
-  const size_t smem_a = (size_t)C * sizeof(float);
 
   for (int t = T - 1; t >= 0; --t) {
     const float* mt = mask_all ? mask_all + (long)t * B : nullptr;
@@ -749,15 +743,14 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        daccC.data_ptr<float>(), dctxpre_f32.data_ptr<float>(),
                        (bf16_t*)dctxpre_all.data_ptr() + (long)t * B * C,
                        (bf16_t*)dgate_all.data_ptr() + (long)t * B * C, B, C);
-    // b5: attention backward — s-parallel dalpha + softmax-bwd dot, then
-    // the scatter stage with wave-reduced dpstate/dD_wei/dU_att
+    // b5: attention backward — wave-per-(b,s) dalpha + softmax-bwd dot,
+    // the (b,s)-parallel scatter (dpctx/daccA/pc_buf), and the per-(b,i)
+    // s-contiguous reduce for dpstate/dD_wei/dU_att/dc_att
     HIP_CHECK(hipMemsetAsync(dot_buf.data_ptr<float>(), 0,
                              (size_t)B * sizeof(float), stream));
-    HIP_CHECK(hipMemsetAsync(
-        dpstate_all.data_ptr<float>() + (long)t * B * A, 0,
-        (size_t)B * A * sizeof(float), stream));
-    hipLaunchKernelGGL(cond_attn_bwd_dalpha, dim3(B, Ts), dim3(256), 0,
-                       stream, (const bf16_t*)ctx_bf.data_ptr(),
+    hipLaunchKernelGGL(cond_attn_bwd_dalpha, dim3(B, cdiv_i(Ts, 4)),
+                       dim3(256), 0, stream,
+                       (const bf16_t*)ctx_bf.data_ptr(),
                        dctxpre_f32.data_ptr<float>(),
                        alphas_all.data_ptr<float>() + (long)t * B * Ts,
                        daccA.data_ptr<float>(), mt,
@@ -765,7 +758,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
                        B, Ts, C);
     hipLaunchKernelGGL(cond_attn_bwd_scatter, dim3(B, cdiv_i(Ts, 256)),
-                       dim3(256), 3 * A * sizeof(float), stream,
+                       dim3(256), 0, stream,
                        alphas_all.data_ptr<float>() + (long)t * B * Ts,
                        dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
                        pctx.data_ptr<float>(),
@@ -773,13 +766,18 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        accA_used.data_ptr<float>() + (long)t * B * Ts,
                        Dwei.data_ptr<float>(), Uatt.data_ptr<float>(),
                        daccA.data_ptr<float>(), dpctx_acc.data_ptr<float>(),
+                       (bf16_t*)pc_buf.data_ptr(), B, Ts, A);
+    hipLaunchKernelGGL(cond_attn_bwd_reduce, dim3(B), dim3(256),
+                       2 * Ts * sizeof(float), stream,
+                       alphas_all.data_ptr<float>() + (long)t * B * Ts,
+                       dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
+                       accA_used.data_ptr<float>() + (long)t * B * Ts,
+                       (const bf16_t*)pc_buf.data_ptr(),
+                       Uatt.data_ptr<float>(),
                        dpstate_all.data_ptr<float>() + (long)t * B * A,
+                       (bf16_t*)dstep_att.data_ptr(),
                        gdDwei.data_ptr<float>(), gdUatt.data_ptr<float>(),
-                       gdcatt.data_ptr<float>(), B, Ts, A);
-    hipLaunchKernelGGL(cond_dpstate_cast, dim3(cdiv_i(B * A, 256)), dim3(256),
-                       0, stream,
-                       dpstate_all.data_ptr<float>() + (long)t * B * A,
-                       (bf16_t*)dstep_att.data_ptr(), Apad32, B, A);
+                       gdcatt.data_ptr<float>(), B, Ts, A, Apad32);
     // b2: dh1 += [dpr2|dpu2|dpxa_lin] @ [U_1|Ux_1]^T + passthrough
     hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH), dim3(384), 0,
                        stream, (const bf16_t*)dstep1.data_ptr(),
