@@ -392,7 +392,30 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_dalpha(
     dal += mm * daccA[(long)b * Ts + s];
     if (dalphas_t != nullptr) dal += dalphas_t[(long)b * Ts + s];
     dal_buf[(long)s * B + b] = dal;
-    atomicAdd(&dot_buf[b], alphas_t[(long)b * Ts + s] * dal);
+    // NOTE: dot(alpha, dal) is computed by cond_attn_bwd_dot — a per-s
+    // atomicAdd here serializes ~Ts adds on one address (~120us/step).
+  }
+}
+
+// softmax-backward dot: dot_buf[b] = sum_s alpha[b,s] * dal[s,b]
+__global__ __launch_bounds__(256) void cond_attn_bwd_dot(
+    const float* __restrict__ alphas_t, const float* __restrict__ dal_buf,
+    float* __restrict__ dot_buf, int B, int Ts) {
+  const int b = blockIdx.x;
+  __shared__ float red[256 / NATS_WAVE];
+  float part = 0.f;
+  for (int s = threadIdx.x; s < Ts; s += blockDim.x)
+    part += alphas_t[(long)b * Ts + s] * dal_buf[(long)s * B + b];
+#pragma unroll
+  for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
+    part += __shfl_down(part, off);
+  if ((threadIdx.x & (NATS_WAVE - 1)) == 0)
+    red[threadIdx.x / NATS_WAVE] = part;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float S = 0.f;
+    for (int w = 0; w < (int)blockDim.x / NATS_WAVE; ++w) S += red[w];
+    dot_buf[b] = S;
   }
 }
 
@@ -409,8 +432,8 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_scatter(
     const float* __restrict__ Dwei, const float* __restrict__ Uatt,
     float* __restrict__ daccA,              // [B][Ts] in/out
     float* __restrict__ dpctx_acc,          // [Ts][B][A] (+=)
-    bf16_t* __restrict__ pc_buf,            // [B][A][Ts]
-    int B, int Ts, int A) {
+    bf16_t* __restrict__ pc_buf,            // [B][A][Tpad8]
+    int B, int Ts, int A, int Tpad8) {
   const int b = blockIdx.x;
   const int s = blockIdx.y * blockDim.x + threadIdx.x;
   if (s >= Ts) return;
@@ -421,12 +444,39 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_scatter(
   const float* srow = pstate_t + (long)b * A;
   float* dprow = dpctx_acc + ((long)s * B + b) * A;
   float daccA_add = 0.f;
-  for (int i = 0; i < A; ++i) {
+  const int A4 = (A % 4 == 0) ? A : 0;  // f32x4 rows need A % 4 == 0
+  int i = 0;
+  for (; i < A4; i += 4) {
+    const float4 p = *(const float4*)(prow + i);
+    const float4 st = *(const float4*)(srow + i);
+    const float4 dw = *(const float4*)(Dwei + i);
+    const float4 ua = *(const float4*)(Uatt + i);
+    float4 dpv = *(const float4*)(dprow + i);
+    const float pc0 = tanhf(p.x + st.x + accAu * dw.x);
+    const float pc1 = tanhf(p.y + st.y + accAu * dw.y);
+    const float pc2 = tanhf(p.z + st.z + accAu * dw.z);
+    const float pc3 = tanhf(p.w + st.w + accAu * dw.w);
+    const float d0 = de * (1.f - pc0 * pc0) * ua.x;
+    const float d1 = de * (1.f - pc1 * pc1) * ua.y;
+    const float d2 = de * (1.f - pc2 * pc2) * ua.z;
+    const float d3 = de * (1.f - pc3 * pc3) * ua.w;
+    dpv.x += d0;
+    dpv.y += d1;
+    dpv.z += d2;
+    dpv.w += d3;
+    *(float4*)(dprow + i) = dpv;
+    daccA_add += d0 * dw.x + d1 * dw.y + d2 * dw.z + d3 * dw.w;
+    pc_buf[((long)b * A + i) * Tpad8 + s] = (bf16_t)pc0;
+    pc_buf[((long)b * A + i + 1) * Tpad8 + s] = (bf16_t)pc1;
+    pc_buf[((long)b * A + i + 2) * Tpad8 + s] = (bf16_t)pc2;
+    pc_buf[((long)b * A + i + 3) * Tpad8 + s] = (bf16_t)pc3;
+  }
+  for (; i < A; ++i) {
     const float pc = tanhf(prow[i] + srow[i] + accAu * Dwei[i]);
     const float dpc = de * (1.f - pc * pc) * Uatt[i];
     dprow[i] += dpc;
     daccA_add += dpc * Dwei[i];
-    pc_buf[((long)b * A + i) * Ts + s] = (bf16_t)pc;
+    pc_buf[((long)b * A + i) * Tpad8 + s] = (bf16_t)pc;
   }
   daccA[(long)b * Ts + s] += daccA_add;
 }
@@ -439,14 +489,14 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_reduce(
     const float* __restrict__ dal_buf,      // [Ts][B]
     const float* __restrict__ dot_buf,      // [B]
     const float* __restrict__ accA_used_t,  // [B][Ts]
-    const bf16_t* __restrict__ pc_buf,      // [B][A][Ts]
+    const bf16_t* __restrict__ pc_buf,      // [B][A][Tpad8]
     const float* __restrict__ Uatt,
     float* __restrict__ dpstate_t,          // [B][A]
     bf16_t* __restrict__ dstep_att,         // [32][Apad]
     float* __restrict__ gdDwei,             // [A] (atomic)
     float* __restrict__ gdUatt,             // [A] (atomic)
     float* __restrict__ gdcatt,             // [1] (atomic)
-    int B, int Ts, int A, int Apad) {
+    int B, int Ts, int A, int Apad, int Tpad8) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   float* sm_de = (float*)smem_raw;      // [Ts]
   float* sm_au = sm_de + Ts;            // [Ts]
@@ -471,10 +521,24 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_reduce(
     atomicAdd(gdcatt, S);
   }
   for (int i = threadIdx.x; i < A; i += blockDim.x) {
-    const bf16_t* prow = pc_buf + ((long)b * A + i) * Ts;
+    const bf16_t* prow = pc_buf + ((long)b * A + i) * Tpad8;
     const float ua = Uatt[i];
     float sps = 0.f, sdw = 0.f, sua = 0.f;
-    for (int s = 0; s < Ts; ++s) {
+    int s = 0;
+    const int T8 = Ts & ~7;
+    for (; s < T8; s += 8) {
+      bf16x8 pv = *(const bf16x8*)(prow + s);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const float pc = (float)pv[k];
+        const float de = sm_de[s + k];
+        const float dpc = de * (1.f - pc * pc) * ua;
+        sps += dpc;
+        sdw += dpc * sm_au[s + k];
+        sua += de * pc;
+      }
+    }
+    for (; s < Ts; ++s) {
       const float pc = (float)prow[s];
       const float de = sm_de[s];
       const float dpc = de * (1.f - pc * pc) * ua;
@@ -676,7 +740,8 @@ std::vector<torch::Tensor> cond_gru_bwd(
                                    : torch::zeros({B, C}, optsF);
   auto dal_buf = torch::empty({Ts, B}, optsF);
   auto dot_buf = torch::empty({B}, optsF);
-  auto pc_buf = torch::empty({B, A, Ts},
+  const int Tpad8 = (Ts + 7) / 8 * 8;
+  auto pc_buf = torch::empty({B, A, Tpad8},
                              dh2_all.options().dtype(torch::kBFloat16));
   const int Apad = Apad32;
   auto dstep1 = torch::zeros({32, K3Hpad}, optsB);
@@ -747,8 +812,6 @@ std::vector<torch::Tensor> cond_gru_bwd(
     // b5: attention backward — wave-per-(b,s) dalpha + softmax-bwd dot,
     // the (b,s)-parallel scatter (dpctx/daccA/pc_buf), and the per-(b,i)
     // s-contiguous reduce for dpstate/dD_wei/dU_att/dc_att
-    HIP_CHECK(hipMemsetAsync(dot_buf.data_ptr<float>(), 0,
-                             (size_t)B * sizeof(float), stream));
     hipLaunchKernelGGL(cond_attn_bwd_dalpha, dim3(B, cdiv_i(Ts, 4)),
                        dim3(256), 0, stream,
                        (const bf16_t*)ctx_bf.data_ptr(),
@@ -758,6 +821,10 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        dalpha_p ? dalpha_p + (long)t * B * Ts : nullptr,
                        dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
                        B, Ts, C);
+    hipLaunchKernelGGL(cond_attn_bwd_dot, dim3(B), dim3(256), 0, stream,
+                       alphas_all.data_ptr<float>() + (long)t * B * Ts,
+                       dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
+                       B, Ts);
     hipLaunchKernelGGL(cond_attn_bwd_scatter, dim3(B, cdiv_i(Ts, 256)),
                        dim3(256), 0, stream,
                        alphas_all.data_ptr<float>() + (long)t * B * Ts,
@@ -767,7 +834,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        accA_used.data_ptr<float>() + (long)t * B * Ts,
                        Dwei.data_ptr<float>(), Uatt.data_ptr<float>(),
                        daccA.data_ptr<float>(), dpctx_acc.data_ptr<float>(),
-                       (bf16_t*)pc_buf.data_ptr(), B, Ts, A);
+                       (bf16_t*)pc_buf.data_ptr(), B, Ts, A, Tpad8);
     hipLaunchKernelGGL(cond_attn_bwd_reduce, dim3(B), dim3(256),
                        2 * Ts * sizeof(float), stream,
                        alphas_all.data_ptr<float>() + (long)t * B * Ts,
@@ -778,7 +845,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        dpstate_all.data_ptr<float>() + (long)t * B * A,
                        (bf16_t*)dstep_att.data_ptr(),
                        gdDwei.data_ptr<float>(), gdUatt.data_ptr<float>(),
-                       gdcatt.data_ptr<float>(), B, Ts, A, Apad32);
+                       gdcatt.data_ptr<float>(), B, Ts, A, Apad32, Tpad8);
     // b2: dh1 += [dpr2|dpu2|dpxa_lin] @ [U_1|Ux_1]^T + passthrough
     hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH), dim3(384), 0,
                        stream, (const bf16_t*)dstep1.data_ptr(),
