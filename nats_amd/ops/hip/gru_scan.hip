@@ -260,6 +260,282 @@ __global__ __launch_bounds__(384) void nats_gru_step_bwd_fused_bidir(
     gru_bwd_fused_body(a1, Ubwd1, B, H, Kpad, blockIdx.x);
 }
 
+// ---------------- persistent bidirectional scans ----------------
+// The per-step launches re-read the packed weights from L2 every step
+// (~14us/step, latency-bound). The persistent variant stages each
+// workgroup's weight slice into LDS ONCE (XOR-swizzled against the
+// 16-way ds_read_b128 bank conflict of 2KB row strides, §6 G4) and walks
+// all T steps inside one launch, exchanging h through global memory with
+// an agent-scope release/acquire grid barrier per step (guide §6 G16:
+// placement-independent, bounded spin with a give-up flag the host
+// checks — a barrier bug aborts instead of hanging the box).
+
+__device__ __forceinline__ bool nats_grid_barrier(unsigned* cnt,
+                                                  unsigned* give_up,
+                                                  unsigned target) {
+  __shared__ unsigned ok_sh;
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // every wave drains
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __hip_atomic_fetch_add(cnt, 1u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+    unsigned ok = 1u, spins = 0u;
+    for (;;) {
+      if (__hip_atomic_load(cnt, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_AGENT) >= target)
+        break;
+      if (__hip_atomic_load(give_up, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_AGENT) != 0u) {
+        ok = 0u;
+        break;
+      }
+      if (++spins > 200000000u) {  // ~seconds; then abort the launch
+        __hip_atomic_store(give_up, 1u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        ok = 0u;
+        break;
+      }
+      __builtin_amdgcn_s_sleep(8);
+    }
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    ok_sh = ok;
+  }
+  __syncthreads();
+  return ok_sh != 0u;
+}
+
+// stage a [rows][Kpad] bf16 slice into LDS with the (row&15)<<4 byte-XOR
+// swizzle; read back with the same XOR (write+read swizzled together).
+__device__ __forceinline__ void stage_weights_lds(bf16_t* lds,
+                                                  const bf16_t* src,
+                                                  int rows, int Kpad) {
+  const long total16 = (long)rows * Kpad * 2 / 16;  // 16B chunks
+  for (long idx = threadIdx.x; idx < total16; idx += blockDim.x) {
+    long byte = idx * 16;
+    const int row = (int)(byte / ((long)Kpad * 2));
+    const long dst = byte ^ (long)((row & 15) << 4);
+    *(uint4*)((char*)lds + dst) = *(const uint4*)((const char*)src + byte);
+  }
+}
+
+__device__ __forceinline__ bf16x8 frag_bt_lds_swz(const bf16_t* lds, int row,
+                                                  int Kpad, int k) {
+  const int lane = threadIdx.x & (NATS_WAVE - 1);
+  const int r = row + (lane & 15);
+  long byte = ((long)r * Kpad + k + (lane >> 4) * 8) * 2;
+  byte ^= (long)((r & 15) << 4);
+  return *(const bf16x8*)((const char*)lds + byte);
+}
+
+// swizzled-LDS MFMA K-loop (B operand from LDS, A from global), 2-deep on A
+#define NATS_MFMA_KLOOP_LDSB(ACC, APTR, AROW, ALD, LDSB, BROW, BLD, KBEG,    \
+                             KEND)                                           \
+  do {                                                                       \
+    int _k = (KBEG);                                                         \
+    const int _ke = (KEND);                                                  \
+    if (_k + 32 >= _ke) {                                                    \
+      if (_k < _ke) {                                                        \
+        bf16x8 _a0 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
+        bf16x8 _b0 = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k);             \
+        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0);\
+      }                                                                      \
+    } else {                                                                 \
+      bf16x8 _a0 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);               \
+      bf16x8 _a1 = frag_a_rowmajor((APTR), (AROW), (ALD), _k + 32);          \
+      for (_k += 64; _k < _ke; _k += 32) {                                   \
+        bf16x8 _a2 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
+        bf16x8 _b0 = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 64);        \
+        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0);\
+        _a0 = _a1; _a1 = _a2;                                                \
+      }                                                                      \
+      {                                                                      \
+        bf16x8 _b0 = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 64);        \
+        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0);\
+        bf16x8 _b1 = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 32);        \
+        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a1, _b1, ACC, 0, 0, 0);\
+      }                                                                      \
+    }                                                                        \
+  } while (0)
+
+struct GruPersistFwd {
+  const bf16_t* xg;    // [T][B][2H]
+  const bf16_t* xc;    // [T][B][H]
+  const float* mask;   // [T][B] or null
+  const bf16_t* Upk;   // [ngrp*3*16][Hpad]
+  float* h_all;        // [T][B][H]
+  bf16_t* h_bf;        // [2][32][Hpad] ping-pong
+  bf16_t* saved;       // [T][B][3H]
+  const float* h0;     // [B][H]
+};
+
+__global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
+    GruPersistFwd p0, GruPersistFwd p1, int T, int B, int H, int Hpad,
+    unsigned* cnt, unsigned* give_up, unsigned nwg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* upk_lds = (bf16_t*)smem;                       // [3*16][Hpad] swz
+  float(*pre)[32][JB + 1] =
+      (float(*)[32][JB + 1])(smem + (long)3 * JB * Hpad * 2);
+
+  const GruPersistFwd& p = (blockIdx.y == 0) ? p0 : p1;
+  const int wg = blockIdx.x;
+  stage_weights_lds(upk_lds, p.Upk + (long)wg * 3 * JB * Hpad, 3 * JB, Hpad);
+  __syncthreads();
+
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int m = wave / 3;
+  const int g = wave % 3;
+  const int j0 = wg * JB;
+  const long hb = (long)32 * Hpad;
+
+  for (int t = 0; t < T; ++t) {
+    const bf16_t* h_bf_in = p.h_bf + (t % 2) * hb;
+    const float* h_prev =
+        (t == 0) ? p.h0 : (p.h_all + (long)(t - 1) * B * H);
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    NATS_MFMA_KLOOP_LDSB(acc, h_bf_in, 16 * m, Hpad, upk_lds, g * JB, Hpad,
+                         0, Hpad);
+    {
+      const int lane = threadIdx.x & (NATS_WAVE - 1);
+      const int col = lane & 15;
+      const int rbase = 16 * m + (lane >> 4) * 4;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) pre[g][rbase + i][col] = acc[i];
+    }
+    __syncthreads();
+
+    const bf16_t* xg_t = p.xg + (long)t * B * 2 * H;
+    const bf16_t* xc_t = p.xc + (long)t * B * H;
+    const float* mask_t = p.mask ? p.mask + (long)t * B : nullptr;
+    float* h_out = p.h_all + (long)t * B * H;
+    bf16_t* h_bf_out = p.h_bf + ((t + 1) % 2) * hb;
+    bf16_t* saved_t = p.saved + (long)t * B * 3 * H;
+    for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+      const int b = idx / JB;
+      const int c = idx % JB;
+      const int j = j0 + c;
+      if (j >= H) continue;
+      const float hp = h_prev[(long)b * H + j];
+      const float pr = pre[0][b][c] + (float)xg_t[(long)b * 2 * H + j];
+      const float pu = pre[1][b][c] + (float)xg_t[(long)b * 2 * H + H + j];
+      const float px = pre[2][b][c];
+      const float r = nats_sigmoid(pr);
+      const float u = nats_sigmoid(pu);
+      const float hbar = tanhf(px * r + (float)xc_t[(long)b * H + j]);
+      float hnew = u * hp + (1.f - u) * hbar;
+      if (mask_t != nullptr) {
+        const float mm = mask_t[b];
+        hnew = mm * hnew + (1.f - mm) * hp;
+      }
+      h_out[(long)b * H + j] = hnew;
+      h_bf_out[(long)b * Hpad + j] = (bf16_t)hnew;
+      saved_t[(long)b * 3 * H + j] = (bf16_t)r;
+      saved_t[(long)b * 3 * H + H + j] = (bf16_t)u;
+      saved_t[(long)b * 3 * H + 2 * H + j] = (bf16_t)px;
+    }
+    if (!nats_grid_barrier(cnt, give_up, nwg * (unsigned)(t + 1))) {
+      // poison output so a barrier give-up surfaces as NaN, never a hang
+      if (threadIdx.x == 0) p.h_all[0] = __builtin_nanf("");
+      return;
+    }
+  }
+}
+
+struct GruPersistBwd {
+  const float* dh_out;  // [T][B][H]
+  const float* h_all;   // [T][B][H]
+  const bf16_t* saved;  // [T][B][3H]
+  const bf16_t* xc;     // [T][B][H]
+  const float* mask;    // [T][B] or null
+  const bf16_t* Ubwd;   // [ngrp*16][K3pad]
+  bf16_t* dstep;        // [2][32][K3pad] ping-pong (zeroed)
+  float* ddirect;       // [B][H] (zeroed)
+  bf16_t* dpre;         // [T][B][4H]
+  const float* h0;      // [B][H]
+};
+
+__global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
+    GruPersistBwd p0, GruPersistBwd p1, int T, int B, int H, int K3pad,
+    unsigned* cnt, unsigned* give_up, unsigned nwg) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* ub_lds = (bf16_t*)smem;  // [16][K3pad] swizzled
+  float(*part)[32][JB + 1] =
+      (float(*)[32][JB + 1])(smem + (long)JB * K3pad * 2);
+
+  const GruPersistBwd& p = (blockIdx.y == 0) ? p0 : p1;
+  const int wg = blockIdx.x;
+  stage_weights_lds(ub_lds, p.Ubwd + (long)wg * JB * K3pad, JB, K3pad);
+  __syncthreads();
+
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int m = wave / 3;
+  const int ks = wave % 3;
+  const int i0 = wg * JB;
+  const int kchunk = ((K3pad / 3 + 31) / 32) * 32;
+  const int kbeg = ks * kchunk;
+  const int kend = min(K3pad, (ks + 1) * kchunk);
+  const long ds = (long)32 * K3pad;
+
+  for (int t = T - 1; t >= 0; --t) {
+    const bf16_t* dstep_in = p.dstep + ((t + 1) % 2) * ds;
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    NATS_MFMA_KLOOP_LDSB(acc, dstep_in, 16 * m, K3pad, ub_lds, 0, K3pad,
+                         kbeg, kend);
+    {
+      const int lane = threadIdx.x & (NATS_WAVE - 1);
+      const int col = lane & 15;
+      const int rbase = 16 * m + (lane >> 4) * 4;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) part[ks][rbase + i][col] = acc[i];
+    }
+    __syncthreads();
+
+    const bf16_t* saved_t = p.saved + (long)t * B * 3 * H;
+    const bf16_t* xc_t = p.xc + (long)t * B * H;
+    const float* h_prev =
+        (t == 0) ? p.h0 : (p.h_all + (long)(t - 1) * B * H);
+    const float* mask_t = p.mask ? p.mask + (long)t * B : nullptr;
+    const float* dh_out_t = p.dh_out + (long)t * B * H;
+    bf16_t* dstep_out = p.dstep + (t % 2) * ds;
+    bf16_t* dpre_t = p.dpre + (long)t * B * 4 * H;
+    for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+      const int b = idx / JB;
+      const int c = idx % JB;
+      const int j = i0 + c;
+      if (j >= H) continue;
+      const long bj = (long)b * H + j;
+      float dh = p.ddirect[bj] + part[0][b][c] + part[1][b][c] +
+                 part[2][b][c] + dh_out_t[bj];
+      const float r = (float)saved_t[(long)b * 3 * H + j];
+      const float u = (float)saved_t[(long)b * 3 * H + H + j];
+      const float px = (float)saved_t[(long)b * 3 * H + 2 * H + j];
+      const float hbar = tanhf(px * r + (float)xc_t[bj]);
+      const float hp = h_prev[bj];
+      const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
+      const float du = dh * mm * (hp - hbar);
+      const float dhbar = dh * mm * (1.f - u);
+      const float dpx = dhbar * (1.f - hbar * hbar);
+      const float dpxl = dpx * r;
+      const float dr = dpx * px;
+      const float dpr = dr * r * (1.f - r);
+      const float dpu = du * u * (1.f - u);
+      p.ddirect[bj] = dh * (mm * u + (1.f - mm));
+      dstep_out[(long)b * K3pad + j] = (bf16_t)dpr;
+      dstep_out[(long)b * K3pad + H + j] = (bf16_t)dpu;
+      dstep_out[(long)b * K3pad + 2 * H + j] = (bf16_t)dpxl;
+      dpre_t[(long)b * 4 * H + j] = (bf16_t)dpr;
+      dpre_t[(long)b * 4 * H + H + j] = (bf16_t)dpu;
+      dpre_t[(long)b * 4 * H + 2 * H + j] = (bf16_t)dpx;
+      dpre_t[(long)b * 4 * H + 3 * H + j] = (bf16_t)dpxl;
+    }
+    if (!nats_grid_barrier(cnt, give_up, nwg * (unsigned)(T - t))) {
+      if (threadIdx.x == 0) p.ddirect[0] = __builtin_nanf("");
+      return;
+    }
+  }
+}
+
 // ---------------- backward recurrent GEMM ----------------
 // out[b, i] = ddirect[b, i] + sum_k dstep[b, k] * Wt[i, k]
 // Generic over the output width (rows of Wt): the encoder/GRU_2 call has
@@ -402,6 +678,46 @@ std::vector<torch::Tensor> gru_scan_fwd_bidir(
   auto stream = at::cuda::getCurrentCUDAStream().stream();
   const long hb = (long)32 * Hpad;
   bf16_t* hbf = (bf16_t*)h_bf.data_ptr();
+
+  // persistent path: weight slices LDS-resident across all T steps
+  const size_t smem_fwd =
+      (size_t)3 * JB * Hpad * 2 + sizeof(float) * 3 * 32 * (JB + 1);
+  const bool persistent = (2 * ngrp <= 192) && (smem_fwd <= 150 * 1024) &&
+                          (getenv("NATS_NO_PERSISTENT") == nullptr);
+  if (persistent) {
+    static bool attr_set = false;
+    if (!attr_set) {
+      HIP_CHECK(hipFuncSetAttribute(
+          (const void*)nats_gru_persistent_fwd,
+          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024));
+      attr_set = true;
+    }
+    auto sync = torch::zeros(
+        {2}, xg0.options().dtype(torch::kInt32));
+    unsigned* cnt = (unsigned*)sync.data_ptr<int>();
+    GruPersistFwd p0{(const bf16_t*)xg0.data_ptr(),
+                     (const bf16_t*)xc0.data_ptr(),
+                     m0,
+                     (const bf16_t*)Upk0.data_ptr(),
+                     h_all0.data_ptr<float>(),
+                     hbf + 0 * 2 * hb,
+                     (bf16_t*)saved0.data_ptr(),
+                     h00.data_ptr<float>()};
+    GruPersistFwd p1{(const bf16_t*)xg1.data_ptr(),
+                     (const bf16_t*)xc1.data_ptr(),
+                     m1,
+                     (const bf16_t*)Upk1.data_ptr(),
+                     h_all1.data_ptr<float>(),
+                     hbf + 1 * 2 * hb,
+                     (bf16_t*)saved1.data_ptr(),
+                     h00.data_ptr<float>()};
+    hipLaunchKernelGGL(nats_gru_persistent_fwd, dim3(ngrp, 2), dim3(384),
+                       smem_fwd, stream, p0, p1, T, B, H, Hpad, cnt, cnt + 1,
+                       (unsigned)(2 * ngrp));
+    HIP_CHECK(hipGetLastError());
+    return {h_all0, saved0, h_all1, saved1};
+  }
+
   for (int t = 0; t < T; ++t) {
     GruFwdArgs a0{
         hbf + 0 * 2 * hb + (t % 2) * hb,
@@ -472,6 +788,57 @@ std::vector<torch::Tensor> gru_scan_bwd_bidir(
   float* dd = ddir.data_ptr<float>();
   auto dh0c = dh_out0.contiguous().to(torch::kFloat32);
   auto dh1c = dh_out1.contiguous().to(torch::kFloat32);
+
+  const size_t smem_bwd =
+      (size_t)JB * K3pad * 2 + sizeof(float) * 3 * 32 * (JB + 1);
+  const bool persistent = (2 * ngrp <= 192) && (smem_bwd <= 150 * 1024) &&
+                          (getenv("NATS_NO_PERSISTENT") == nullptr);
+  if (persistent) {
+    static bool attr_set = false;
+    if (!attr_set) {
+      HIP_CHECK(hipFuncSetAttribute(
+          (const void*)nats_gru_persistent_bwd,
+          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024));
+      attr_set = true;
+    }
+    auto sync = torch::zeros({2}, dh_out0.options().dtype(torch::kInt32));
+    unsigned* cnt = (unsigned*)sync.data_ptr<int>();
+    GruPersistBwd p0{dh0c.data_ptr<float>(),
+                     h_all0.data_ptr<float>(),
+                     (const bf16_t*)saved0.data_ptr(),
+                     (const bf16_t*)xc0.data_ptr(),
+                     m0,
+                     (const bf16_t*)Ubwd0.data_ptr(),
+                     dsp + 0 * 2 * ds,
+                     dd + 0,
+                     (bf16_t*)dpre0.data_ptr(),
+                     h00.data_ptr<float>()};
+    GruPersistBwd p1{dh1c.data_ptr<float>(),
+                     h_all1.data_ptr<float>(),
+                     (const bf16_t*)saved1.data_ptr(),
+                     (const bf16_t*)xc1.data_ptr(),
+                     m1,
+                     (const bf16_t*)Ubwd1.data_ptr(),
+                     dsp + 1 * 2 * ds,
+                     dd + (long)B * H,
+                     (bf16_t*)dpre1.data_ptr(),
+                     h00.data_ptr<float>()};
+    hipLaunchKernelGGL(nats_gru_persistent_bwd, dim3(ngrp, 2), dim3(384),
+                       smem_bwd, stream, p0, p1, T, B, H, K3pad, cnt, cnt + 1,
+                       (unsigned)(2 * ngrp));
+    // final dh0 per direction (dstep(0) lives in ping-pong slot 0)
+    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrp), dim3(384), 0,
+                       stream, dsp + 0 * 2 * ds,
+                       (const bf16_t*)Ubwd0.data_ptr(), dd + 0,
+                       dh0_0.data_ptr<float>(), B, H, K3pad);
+    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrp), dim3(384), 0,
+                       stream, dsp + 1 * 2 * ds,
+                       (const bf16_t*)Ubwd1.data_ptr(), dd + (long)B * H,
+                       dh0_1.data_ptr<float>(), B, H, K3pad);
+    HIP_CHECK(hipGetLastError());
+    return {dpre0, dh0_0, dpre1, dh0_1};
+  }
+
   for (int t = T - 1; t >= 0; --t) {
     // parity: step t reads dstep[(t+1)%2], writes dstep[t%2]
     GruBwdArgs a0{
