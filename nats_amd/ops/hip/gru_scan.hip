@@ -685,13 +685,6 @@ std::vector<torch::Tensor> gru_scan_fwd_bidir(
   const bool persistent = (2 * ngrp <= 192) && (smem_fwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
-    static bool attr_set = false;
-    if (!attr_set) {
-      HIP_CHECK(hipFuncSetAttribute(
-          (const void*)nats_gru_persistent_fwd,
-          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024));
-      attr_set = true;
-    }
     auto sync = torch::zeros(
         {2}, xg0.options().dtype(torch::kInt32));
     unsigned* cnt = (unsigned*)sync.data_ptr<int>();
@@ -794,13 +787,6 @@ std::vector<torch::Tensor> gru_scan_bwd_bidir(
   const bool persistent = (2 * ngrp <= 192) && (smem_bwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
-    static bool attr_set = false;
-    if (!attr_set) {
-      HIP_CHECK(hipFuncSetAttribute(
-          (const void*)nats_gru_persistent_bwd,
-          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024));
-      attr_set = true;
-    }
     auto sync = torch::zeros({2}, dh_out0.options().dtype(torch::kInt32));
     unsigned* cnt = (unsigned*)sync.data_ptr<int>();
     GruPersistBwd p0{dh0c.data_ptr<float>(),
