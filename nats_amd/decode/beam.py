@@ -73,7 +73,7 @@ def distraction_penalties(hyp_dec_alphas, hyp_ctxs, hyp_states_dis,
 @torch.no_grad()
 def gen_sample(model, x, k=1, maxlen=30, stochastic=True, argmax=False,
                use_unk=False, kl_factor=0.0, ctx_factor=0.0, state_factor=0.0,
-               generator=None):
+               generator=None, use_graph=False):
     """Generate one summary by beam search or stochastic sampling.
 
     model: NatsModel; x: (T,1) int64 tensor on the model's device.
@@ -107,13 +107,33 @@ def gen_sample(model, x, k=1, maxlen=30, stochastic=True, argmax=False,
     acc_ctx = torch.zeros((live_k, C), device=device, dtype=ctx0.dtype)
     acc_alpha = torch.zeros((live_k, Ts), device=device, dtype=ctx0.dtype)
 
+    # hipGraph-captured decode step (beam mode on GPU): the whole f_next
+    # kernel chain replays as one graph per step
+    stepper = None
+    if use_graph and not stochastic and device.type == "cuda":
+        from .graph import GraphDecodeStepper
+        stepper = GraphDecodeStepper(model, ctx0, pctx0, k)
+
     for ii in range(maxlen):
-        ctx = ctx0.expand(Ts, live_k, C)
-        pctx = pctx0.expand(Ts, live_k, pctx0.shape[2])
-        probs, w_sample, next_state, dec_alphas, ctxs, acc_ctx, acc_alpha = \
-            model.f_next(next_w, ctx, None, pctx, next_state, acc_ctx,
-                         acc_alpha, generator=generator,
-                         sample_draw=stochastic and not argmax)
+        if stepper is not None:
+            probs_k, h2_k, alpha_k, ctxs_k, accC_k, accA_k = stepper.step(
+                next_w, next_state.float(), acc_ctx.float(),
+                acc_alpha.float())
+            probs = probs_k[:live_k]
+            w_sample = probs.argmax(dim=-1)
+            next_state = h2_k[:live_k]
+            dec_alphas = alpha_k[:live_k]
+            ctxs = ctxs_k[:live_k]
+            acc_ctx = accC_k[:live_k]
+            acc_alpha = accA_k[:live_k]
+        else:
+            ctx = ctx0.expand(Ts, live_k, C)
+            pctx = pctx0.expand(Ts, live_k, pctx0.shape[2])
+            probs, w_sample, next_state, dec_alphas, ctxs, acc_ctx, \
+                acc_alpha = model.f_next(
+                    next_w, ctx, None, pctx, next_state, acc_ctx, acc_alpha,
+                    generator=generator,
+                    sample_draw=stochastic and not argmax)
 
         if stochastic:
             if argmax:

@@ -1,0 +1,78 @@
+"""hipGraph-captured beam-decode step.
+
+The reference's beam loop calls a compiled f_next per step (nats.py:960);
+our f_next is itself a chain of ~15 kernel launches (embedding, fused
+cond-GRU step, readout GEMMs, softmax). Capturing that chain into a
+hipGraph (torch.cuda.CUDAGraph on ROCm) collapses per-step launch
+overhead into one graph replay.
+
+Shapes are static: the stepper is captured for a fixed (beam k, source
+length Ts); the beam keeps k rows alive (dead/padded rows carry copies of
+row 0, harmless — callers slice [:live_k]). Host-side feedback between
+replays (parent gather after rank selection) writes the static input
+buffers with index copies.
+"""
+
+import torch
+
+
+class GraphDecodeStepper:
+    """Replayable f_next for beam search (argmax/beam mode, no sampling)."""
+
+    def __init__(self, model, ctx0, pctx0, k):
+        """ctx0 (Ts,1,C), pctx0 (Ts,1,A) from f_init/project_ctx."""
+        self.model = model
+        self.k = k
+        device = ctx0.device
+        Ts, _, C = ctx0.shape
+        H = model.options["dim"]
+        self.ctx = ctx0.expand(Ts, k, C).contiguous()
+        self.pctx = pctx0.expand(Ts, k, pctx0.shape[2]).contiguous()
+        self.y_in = torch.zeros(k, dtype=torch.int64, device=device)
+        self.state_in = torch.zeros(k, H, device=device)
+        self.accC_in = torch.zeros(k, C, device=device)
+        self.accA_in = torch.zeros(k, Ts, device=device)
+        self.graph = None
+        self.outs = None
+
+    def _run(self):
+        return self.model.f_next(self.y_in, self.ctx, None, self.pctx,
+                                 self.state_in, self.accC_in, self.accA_in,
+                                 sample_draw=False)
+
+    def capture(self):
+        # warmup (allocator + kernels) on a side stream, then capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._run()
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            outs = self._run()
+            # keep stable output storage across replays
+            self.outs = tuple(o.clone() if isinstance(o, torch.Tensor) else o
+                              for o in outs)
+        return self
+
+    def step(self, y, state, acc_ctx, acc_alpha):
+        """Run one decode step for `live` rows (<= k); returns
+        (probs, state, alpha, ctx_t, acc_ctx, acc_alpha) each k rows —
+        caller slices [:live]."""
+        live = y.shape[0]
+        self.y_in[:live] = y
+        if live < self.k:
+            self.y_in[live:] = y[0]
+        self.state_in[:live] = state
+        self.accC_in[:live] = acc_ctx
+        self.accA_in[:live] = acc_alpha
+        if live < self.k:
+            self.state_in[live:] = state[0]
+            self.accC_in[live:] = acc_ctx[0]
+            self.accA_in[live:] = acc_alpha[0]
+        if self.graph is None:
+            self.capture()
+        self.graph.replay()
+        probs, _, h2, alpha, ctx_t, accC, accA = self.outs
+        return probs, h2, alpha, ctx_t, accC, accA

@@ -73,8 +73,8 @@ class CondGRUScanFn(torch.autograd.Function):
         outs = ext.cond_gru_fwd(
             yg, yc, mask, init_state, ctx_bf, ctx_mask, pctx_f, Upk2, W1pk,
             WattPk, vecs["b1"], vecs["bx1"], vecs["Uatt"],
-            float(c_att.float().reshape(-1)[0]), vecs["Dwei"], vecs["Wcon"],
-            vecs["Ucon"], acc_ctx0, acc_alpha0)
+            c_att.float().reshape(-1).contiguous(), vecs["Dwei"],
+            vecs["Wcon"], vecs["Ucon"], acc_ctx0, acc_alpha0)
         (h2_all, ctxs_all, alphas_all, accC, accA, h1_all, saved2, saved1,
          pstate_all, ctxpre_all, accA_used, accC_used) = outs
         fctx.save_for_backward(
@@ -206,7 +206,7 @@ def cond_gru_step_hip(h_prev, x_g, x_c, ctx, ctx_mask, pctx, acc_ctx,
         P["decoder_b_1"].float().contiguous(),
         P["decoder_bx_1"].float().contiguous(),
         P["decoder_U_att"].float().reshape(-1).contiguous(),
-        float(P["decoder_c_att"].float().reshape(-1)[0]),
+        P["decoder_c_att"].float().reshape(-1).contiguous(),
         P["decoder_D_wei"].float().reshape(-1).contiguous(),
         P["decoder_W_con"].float().reshape(-1).contiguous(),
         P["decoder_U_con"].float().reshape(-1).contiguous(),

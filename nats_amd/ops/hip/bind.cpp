@@ -32,8 +32,9 @@ std::vector<torch::Tensor> cond_gru_fwd(
     torch::Tensor init_state, torch::Tensor ctx_bf,
     c10::optional<torch::Tensor> ctx_mask, torch::Tensor pctx,
     torch::Tensor Upk2, torch::Tensor W1pk, torch::Tensor WattPk,
-    torch::Tensor b1, torch::Tensor bx1, torch::Tensor Uatt, double catt,
-    torch::Tensor Dwei, torch::Tensor Wcon, torch::Tensor Ucon,
+    torch::Tensor b1, torch::Tensor bx1, torch::Tensor Uatt,
+    torch::Tensor catt, torch::Tensor Dwei, torch::Tensor Wcon,
+    torch::Tensor Ucon,
     c10::optional<torch::Tensor> accC0, c10::optional<torch::Tensor> accA0);
 std::vector<torch::Tensor> cond_gru_bwd(
     torch::Tensor dh2_all, c10::optional<torch::Tensor> dctxs_all,

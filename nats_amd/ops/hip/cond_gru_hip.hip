@@ -60,7 +60,7 @@ __global__ __launch_bounds__(256) void cond_attn_escore(
     const float* __restrict__ accA,      // [B][Ts] (pre-update)
     const float* __restrict__ Dwei,      // [A]
     const float* __restrict__ Uatt,      // [A]
-    float catt,
+    const float* __restrict__ catt_p,    // [1]
     float* __restrict__ e_buf,           // [Ts][B]
     int B, int Ts, int A) {
   const int b = blockIdx.x;
@@ -69,7 +69,7 @@ __global__ __launch_bounds__(256) void cond_attn_escore(
   const float accAu = accA[(long)b * Ts + s];
   const float* prow = pctx + ((long)s * B + b) * A;
   const float* srow = pstate_t + (long)b * A;
-  float e = catt;
+  float e = catt_p[0];
   int i = 0;
   const int A4 = A & ~3;
   for (; i < A4; i += 4) {
@@ -563,8 +563,9 @@ std::vector<torch::Tensor> cond_gru_fwd(
     torch::Tensor init_state, torch::Tensor ctx_bf,
     c10::optional<torch::Tensor> ctx_mask, torch::Tensor pctx,
     torch::Tensor Upk2, torch::Tensor W1pk, torch::Tensor WattPk,
-    torch::Tensor b1, torch::Tensor bx1, torch::Tensor Uatt, double catt,
-    torch::Tensor Dwei, torch::Tensor Wcon, torch::Tensor Ucon,
+    torch::Tensor b1, torch::Tensor bx1, torch::Tensor Uatt,
+    torch::Tensor catt, torch::Tensor Dwei, torch::Tensor Wcon,
+    torch::Tensor Ucon,
     c10::optional<torch::Tensor> accC0, c10::optional<torch::Tensor> accA0) {
   const int T = yg.size(0), B = yg.size(1);
   const int H = yc.size(2);
@@ -648,7 +649,7 @@ std::vector<torch::Tensor> cond_gru_fwd(
                        0, stream, pctx.data_ptr<float>(),
                        pstate_all.data_ptr<float>() + (long)t * B * A,
                        accA.data_ptr<float>(), Dwei.data_ptr<float>(),
-                       Uatt.data_ptr<float>(), (float)catt,
+                       Uatt.data_ptr<float>(), catt.data_ptr<float>(),
                        e_buf.data_ptr<float>(), B, Ts, A);
     hipLaunchKernelGGL(cond_attn_softmax, dim3(B), dim3(256), 0, stream,
                        accA.data_ptr<float>(),
