@@ -70,6 +70,31 @@ def distraction_penalties(hyp_dec_alphas, hyp_ctxs, hyp_states_dis,
     return alphac_score, ctxs_score, state_score
 
 
+def distraction_penalties_gpu(hist_a, hist_c, hist_s, cur_a, cur_c, cur_s,
+                              kl_factor, ctx_factor, state_factor):
+    """Batched device-side rerank penalties (same math as
+    distraction_penalties, vectorised over (history step, hypothesis) —
+    the host scipy loops cost ~3x decode throughput at beam 10).
+
+    hist_* are (n_hist, live_k, dim) fp32 device tensors; cur_* (live_k,
+    dim). Returns a (live_k,) fp32 tensor of summed penalties."""
+    p = hist_a / hist_a.sum(-1, keepdim=True)
+    q = (cur_a / cur_a.sum(-1, keepdim=True)).unsqueeze(0)
+    logterm = torch.where(p > 0, p * (torch.log(p) - torch.log(q)),
+                          torch.zeros_like(p))
+    a_pen = -kl_factor * logterm.sum(-1).min(dim=0).values
+
+    def cosd(h, c):
+        num = (h * c.unsqueeze(0)).sum(-1)
+        den = h.norm(dim=-1) * c.norm(dim=-1).unsqueeze(0)
+        d = 1.0 - num / den
+        return torch.where(den == 0, torch.zeros_like(d), d)
+
+    c_pen = ctx_factor * cosd(hist_c, cur_c).max(dim=0).values
+    s_pen = state_factor * cosd(hist_s, cur_s).max(dim=0).values
+    return a_pen + c_pen + s_pen
+
+
 @torch.no_grad()
 def gen_sample(model, x, k=1, maxlen=30, stochastic=True, argmax=False,
                use_unk=False, kl_factor=0.0, ctx_factor=0.0, state_factor=0.0,
@@ -111,8 +136,14 @@ def gen_sample(model, x, k=1, maxlen=30, stochastic=True, argmax=False,
     # kernel chain replays as one graph per step
     stepper = None
     if use_graph and not stochastic and device.type == "cuda":
-        from .graph import GraphDecodeStepper
-        stepper = GraphDecodeStepper(model, ctx0, pctx0, k)
+        from .graph import get_stepper
+        stepper = get_stepper(model, ctx0.float(), pctx0.float(), k)
+
+    any_lambda = (kl_factor > 0.0 or ctx_factor > 0.0 or state_factor > 0.0)
+    # device-resident rerank histories (GPU path); the numpy histories are
+    # still kept — they are the returned alignment output
+    gpu_rerank = any_lambda and device.type == "cuda"
+    hist_a_dev = hist_c_dev = hist_s_dev = None
 
     for ii in range(maxlen):
         if stepper is not None:
@@ -155,15 +186,22 @@ def gen_sample(model, x, k=1, maxlen=30, stochastic=True, argmax=False,
         cand_flat = cand_scores.flatten()
         ranks_flat = cand_flat.argsort()[: (k - dead_k)]
 
-        if ii > 0 and (kl_factor > 0.0 or ctx_factor > 0.0 or
-                       state_factor > 0.0):
-            da = dec_alphas.float().cpu().numpy()
-            cs = ctxs.float().cpu().numpy()
-            ns = next_state.float().cpu().numpy()
-            a_s, c_s, s_s = distraction_penalties(
-                hyp_dec_alphas, hyp_ctxs, hyp_states_dis, da, cs, ns,
-                kl_factor, ctx_factor, state_factor)
-            new_cand = cand_scores + a_s[:, None] + c_s[:, None] + s_s[:, None]
+        if ii > 0 and any_lambda:
+            if gpu_rerank:
+                pen = distraction_penalties_gpu(
+                    hist_a_dev, hist_c_dev, hist_s_dev, dec_alphas.float(),
+                    ctxs.float(), next_state.float(), kl_factor, ctx_factor,
+                    state_factor).cpu().numpy()
+                new_cand = cand_scores + pen[:, None]
+            else:
+                da = dec_alphas.float().cpu().numpy()
+                cs = ctxs.float().cpu().numpy()
+                ns = next_state.float().cpu().numpy()
+                a_s, c_s, s_s = distraction_penalties(
+                    hyp_dec_alphas, hyp_ctxs, hyp_states_dis, da, cs, ns,
+                    kl_factor, ctx_factor, state_factor)
+                new_cand = (cand_scores + a_s[:, None] + c_s[:, None] +
+                            s_s[:, None])
             ranks_flat = new_cand.flatten().argsort()[: (k - dead_k)]
 
         voc_size = next_p.shape[1]
@@ -197,7 +235,21 @@ def gen_sample(model, x, k=1, maxlen=30, stochastic=True, argmax=False,
             new_hyp_acc_alpha.append(acc_alpha_np[ti].copy())
             new_hyp_states_dis.append(hyp_states_dis[ti] + [ns_np[ti, :].copy()])
 
+        if gpu_rerank:
+            sel_t = torch.as_tensor(trans_indices.astype(numpy.int64),
+                                    device=device)
+            cur_a = dec_alphas.float()[sel_t].unsqueeze(0)
+            cur_c = ctxs.float()[sel_t].unsqueeze(0)
+            cur_s = next_state.float()[sel_t].unsqueeze(0)
+            if hist_a_dev is None:
+                hist_a_new, hist_c_new, hist_s_new = cur_a, cur_c, cur_s
+            else:
+                hist_a_new = torch.cat([hist_a_dev[:, sel_t], cur_a], dim=0)
+                hist_c_new = torch.cat([hist_c_dev[:, sel_t], cur_c], dim=0)
+                hist_s_new = torch.cat([hist_s_dev[:, sel_t], cur_s], dim=0)
+
         new_live_k = 0
+        live_pos = []
         hyp_samples, hyp_scores_l, hyp_states = [], [], []
         hyp_dec_alphas, hyp_ctxs, hyp_states_dis = [], [], []
         hyp_acc_ctx, hyp_acc_alpha = [], []
@@ -210,6 +262,7 @@ def gen_sample(model, x, k=1, maxlen=30, stochastic=True, argmax=False,
                 dead_k += 1
             else:
                 new_live_k += 1
+                live_pos.append(idx)
                 hyp_samples.append(new_hyp_samples[idx])
                 hyp_scores_l.append(new_hyp_scores[idx])
                 hyp_states.append(new_hyp_states[idx])
@@ -226,6 +279,12 @@ def gen_sample(model, x, k=1, maxlen=30, stochastic=True, argmax=False,
             break
         if dead_k >= k:
             break
+
+        if gpu_rerank:
+            keep = torch.as_tensor(live_pos, dtype=torch.int64, device=device)
+            hist_a_dev = hist_a_new[:, keep]
+            hist_c_dev = hist_c_new[:, keep]
+            hist_s_dev = hist_s_new[:, keep]
 
         next_w = torch.tensor([w[-1] for w in hyp_samples],
                               dtype=torch.int64, device=device)

@@ -16,6 +16,26 @@ buffers with index copies.
 import torch
 
 
+import weakref
+
+_STEPPER_CACHE = weakref.WeakKeyDictionary()
+
+
+def get_stepper(model, ctx0, pctx0, k):
+    """Cached stepper per (model, Ts, k): the captured graph is reused
+    across sentences of the same source length — only the static context
+    buffers are refreshed."""
+    per_model = _STEPPER_CACHE.setdefault(model, {})
+    key = (int(ctx0.shape[0]), int(k))
+    st = per_model.get(key)
+    if st is None:
+        st = GraphDecodeStepper(model, ctx0, pctx0, k)
+        per_model[key] = st
+    else:
+        st.set_context(ctx0, pctx0)
+    return st
+
+
 class GraphDecodeStepper:
     """Replayable f_next for beam search (argmax/beam mode, no sampling)."""
 
@@ -34,6 +54,13 @@ class GraphDecodeStepper:
         self.accA_in = torch.zeros(k, Ts, device=device)
         self.graph = None
         self.outs = None
+
+    def set_context(self, ctx0, pctx0):
+        """Refresh the static context buffers for a new source sequence
+        (same Ts) without recapturing the graph."""
+        Ts, _, C = ctx0.shape
+        self.ctx.copy_(ctx0.expand(Ts, self.k, C))
+        self.pctx.copy_(pctx0.expand(Ts, self.k, pctx0.shape[2]))
 
     def _run(self):
         return self.model.f_next(self.y_in, self.ctx, None, self.pctx,
