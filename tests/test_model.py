@@ -137,3 +137,41 @@ def test_bos_embedding_is_zero(tiny_options):
     out_tok = model.f_next(torch.tensor([3]), ctx, None, pctx, state, z, za,
                            sample_draw=False)
     assert not torch.allclose(out_bos[0], out_tok[0])
+
+
+def test_layer_registry(tiny_options):
+    """Registry parity (layers dict + get_layer, nats.py:106-114)."""
+    import torch
+    from nats_amd.models.registry import get_layer, layers, dropout_layer
+    assert set(layers) == {"ff", "gru", "gru_cond"}
+    model = NatsModel(tiny_options, seed=1)
+    P = {k: v.detach() for k, v in model.P.items()}
+    T, B = 5, 3
+    emb = torch.randn(T, B, tiny_options["dim_word"])
+    out = get_layer("gru")[1](P, emb, tiny_options, prefix="encoder")
+    assert out[0].shape == (T, B, tiny_options["dim"])
+    # ff layer
+    h = torch.randn(B, 2 * tiny_options["dim"])
+    o = get_layer("ff")[1](P, h, tiny_options, prefix="ff_state")
+    assert o.shape == (B, tiny_options["dim"])
+    # dropout: train mode masks, eval mode scales by (1-p)
+    x = torch.ones(4, 4)
+    g = torch.Generator().manual_seed(0)
+    d_train = dropout_layer(x, True, generator=g)
+    assert set(d_train.flatten().tolist()) <= {0.0, 1.0}
+    d_eval = dropout_layer(x, False)
+    assert torch.allclose(d_eval, x * 0.5)
+
+
+def test_registry_cond_layer(tiny_options):
+    import torch
+    from nats_amd.models.registry import get_layer
+    model = NatsModel(tiny_options, seed=2)
+    P = {k: v.detach() for k, v in model.P.items()}
+    T, B, Ts = 4, 2, 6
+    H = tiny_options["dim"]
+    emb = torch.randn(T, B, tiny_options["dim_word"])
+    ctx = torch.randn(Ts, B, 2 * H)
+    outs = get_layer("gru_cond")[1](P, emb, tiny_options, context=ctx)
+    assert outs[0].shape == (T, B, H)
+    assert outs[2].shape == (T, B, Ts)
