@@ -37,9 +37,9 @@ CONFIGS = {
     # BASELINE.json configs[1] — LCSTS shape, 500-dim
     "lcsts": dict(src=120, tgt=30, n_words=4000, dim=500, dim_word=100,
                   dim_att=100, batch=20),
-    # BASELINE.json configs[4] — long-doc stress (single-layer variant)
+    # BASELINE.json configs[4] — long-doc stress: 4-layer stacked bi-GRU
     "longdoc": dict(src=4000, tgt=100, n_words=30000, dim=2048, dim_word=100,
-                    dim_att=100, batch=4),
+                    dim_att=100, batch=4, enc_depth=4),
     # tiny smoke config (CPU-capable)
     "toy": dict(src=40, tgt=12, n_words=500, dim=64, dim_word=32, dim_att=16,
                 batch=8),
@@ -80,7 +80,8 @@ def main():
     opts = default_options(
         dim=cfg["dim"], dim_word=cfg["dim_word"], dim_att=cfg["dim_att"],
         n_words=cfg["n_words"], batch_size=batch, optimizer="adadelta",
-        clip_c=100.0, maxlen=cfg["src"] + 1)
+        clip_c=100.0, maxlen=cfg["src"] + 1,
+        enc_depth=cfg.get("enc_depth", 1))
     model = NatsModel(opts, seed=1234).to(device)
     dp = DataParallelGrads(model.parameters())
     dp.broadcast_params()
@@ -176,6 +177,7 @@ def main():
                 "model": "distraction-gru-seq2seq",
                 "shape": cfg_name,
                 "dim": cfg["dim"], "dim_word": cfg["dim_word"],
+                "enc_depth": cfg.get("enc_depth", 1),
                 "dim_att": cfg["dim_att"], "vocab": cfg["n_words"],
                 "src_len": cfg["src"], "tgt_len": cfg["tgt"],
                 "batch_per_gpu": batch,

@@ -175,3 +175,21 @@ def test_registry_cond_layer(tiny_options):
     outs = get_layer("gru_cond")[1](P, emb, tiny_options, context=ctx)
     assert outs[0].shape == (T, B, H)
     assert outs[2].shape == (T, B, Ts)
+
+
+def test_stacked_encoder():
+    """enc_depth>1 (BASELINE configs[4] stress shape, tiny dims here)."""
+    opts = default_options(dim_word=8, dim=12, dim_att=6, n_words=40,
+                           enc_depth=3)
+    model = NatsModel(opts, seed=1)
+    assert "encoder_l2_Ux" in model.P and "encoder_r_l1_W" in model.P
+    x, x_mask, y, y_mask = _batch(opts, B=2)
+    cost = model(x, x_mask, y, y_mask)
+    assert torch.isfinite(cost).all()
+    cost.mean().backward()
+    assert model.P["encoder_l1_U"].grad is not None
+    # checkpoint roundtrip includes the stacked keys
+    params = model.get_params()
+    m2 = NatsModel(opts, params=params)
+    for k in model.P:
+        torch.testing.assert_close(model.P[k], m2.P[k])
