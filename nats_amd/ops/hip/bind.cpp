@@ -27,6 +27,10 @@ std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
 torch::Tensor softmax_ce_bwd(torch::Tensor logits, torch::Tensor targets,
                              torch::Tensor stats, torch::Tensor dnll);
 torch::Tensor mfma_gemm_bt(torch::Tensor A, torch::Tensor Bt);
+void fused_adadelta_step(torch::Tensor ptrs, torch::Tensor sizes,
+                         torch::Tensor chunk_tensor, torch::Tensor chunk_off,
+                         torch::Tensor g2, double clip_c, double rho,
+                         double eps);
 std::vector<torch::Tensor> cond_gru_fwd(
     torch::Tensor yg, torch::Tensor yc, c10::optional<torch::Tensor> mask,
     torch::Tensor init_state, torch::Tensor ctx_bf,
@@ -60,6 +64,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_ce_fwd", &softmax_ce_fwd, "fused softmax+CE forward");
   m.def("softmax_ce_bwd", &softmax_ce_bwd, "fused softmax+CE backward");
   m.def("mfma_gemm_bt", &mfma_gemm_bt, "MFMA layout self-test GEMM");
+  m.def("fused_adadelta_step", &fused_adadelta_step,
+        "fused multi-tensor global-norm clip + adadelta");
   m.def("cond_gru_fwd", &cond_gru_fwd, "fused cond-GRU decoder forward");
   m.def("cond_gru_bwd", &cond_gru_bwd, "fused cond-GRU decoder backward");
 }

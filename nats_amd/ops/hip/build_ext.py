@@ -17,6 +17,7 @@ SOURCES = [
     os.path.join(HERE, "cond_gru.hip"),
     os.path.join(HERE, "softmax_ce.hip"),
     os.path.join(HERE, "mfma_test.hip"),
+    os.path.join(HERE, "optim.hip"),
 ]
 
 

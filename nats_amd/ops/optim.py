@@ -1,39 +1,62 @@
-"""Fused GPU optimizer step (global-norm clip + adadelta) — wrapper.
-
-Until the HIP kernel lands, this subclass runs the same math through
-torch._foreach ops on GPU (one fused multi-tensor pass per state update);
-the HIP path replaces it transparently when the extension is present.
-Formula identical to engine.optim.Adadelta (nats.py:1145-1173).
-"""
+"""Fused GPU optimizer step: global-norm clip + adadelta in two
+multi-tensor HIP kernels (ops/hip/optim.hip). Falls back to torch._foreach
+when the extension is unavailable. Formulas identical to
+engine.optim.Adadelta + clip_grads_global_norm (nats.py:1145-1173,
+1344-1356)."""
 
 import torch
 
 from ..engine.optim import Adadelta
+from . import _hip_ext
+
+CHUNK = 1 << 16
 
 
 class FusedAdadelta(Adadelta):
+    def __init__(self, *args, **kwargs):
+        super().__init__(*args, **kwargs)
+        self._tables = None
+
+    def _build_tables(self, device):
+        sizes = [p.numel() for _, p in self.named]
+        chunk_tensor, chunk_off = [], []
+        for ti, n in enumerate(sizes):
+            for off in range(0, n, CHUNK):
+                chunk_tensor.append(ti)
+                chunk_off.append(off)
+        self._tables = (
+            torch.tensor(sizes, dtype=torch.int64, device=device),
+            torch.tensor(chunk_tensor, dtype=torch.int32, device=device),
+            torch.tensor(chunk_off, dtype=torch.int64, device=device),
+            torch.zeros(1, dtype=torch.float32, device=device),
+        )
+
     @torch.no_grad()
-    def _update(self):
-        from . import _hip_ext
+    def step(self):
         ext = _hip_ext()
-        ks = [k for k, p in self.named if p.grad is not None]
-        ps = [p for _, p in self.named if p.grad is not None]
-        gs = [p.grad for p in ps]
-        rg2 = [self.state[k]["rg2"] for k in ks]
-        ru2 = [self.state[k]["ru2"] for k in ks]
-        if ext is not None and hasattr(ext, "adadelta_step"):
-            ext.adadelta_step(ps, gs, rg2, ru2, self.rho, self.eps)
-            return
-        rho, eps = self.rho, self.eps
-        torch._foreach_mul_(rg2, rho)
-        g2 = torch._foreach_mul(gs, gs)
-        torch._foreach_add_(rg2, g2, alpha=1.0 - rho)
-        num = torch._foreach_sqrt(torch._foreach_add(ru2, eps))
-        den = torch._foreach_sqrt(torch._foreach_add(rg2, eps))
-        ud = torch._foreach_div(num, den)
-        ud = torch._foreach_mul(ud, gs)
-        torch._foreach_neg_(ud)
-        torch._foreach_mul_(ru2, rho)
-        ud2 = torch._foreach_mul(ud, ud)
-        torch._foreach_add_(ru2, ud2, alpha=1.0 - rho)
-        torch._foreach_add_(ps, ud)
+        ps = [p for _, p in self.named]
+        if ext is None or not ps[0].is_cuda or any(
+                p.grad is None for p in ps):
+            return super().step()
+        self.t += 1
+        device = ps[0].device
+        if self._tables is None:
+            self._build_tables(device)
+        sizes_t, chunk_tensor, chunk_off, g2 = self._tables
+        g2.zero_()
+        rg2 = [self.state[k]["rg2"] for k, _ in self.named]
+        ru2 = [self.state[k]["ru2"] for k, _ in self.named]
+        # pointer table rebuilt each step (grad storages change)
+        ptrs = torch.tensor(
+            [p.data_ptr() for p in ps] +
+            [p.grad.data_ptr() for p in ps] +
+            [t.data_ptr() for t in rg2] +
+            [t.data_ptr() for t in ru2],
+            dtype=torch.int64, device=device)
+        ext.fused_adadelta_step(ptrs, sizes_t, chunk_tensor, chunk_off, g2,
+                                float(self.clip_c), self.rho, self.eps)
+        return torch.sqrt(g2[0])  # pre-clip norm, no host sync
+
+    @torch.no_grad()
+    def _update(self):  # pragma: no cover - only used via super().step()
+        Adadelta._update(self)

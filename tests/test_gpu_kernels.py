@@ -366,3 +366,30 @@ def test_beam_distraction_gpu_runs(ext):
                           use_unk=True, kl_factor=0.6, ctx_factor=0.6,
                           state_factor=0.6)
     assert len(s) >= 1 and all(cc > 0 for cc in c)
+
+
+def test_fused_adadelta_matches_cpu(ext):
+    """fused clip+adadelta HIP kernel vs the CPU reference optimizer."""
+    from nats_amd.engine.optim import Adadelta
+    from nats_amd.ops.optim import FusedAdadelta
+    torch.manual_seed(0)
+    shapes = [(70000,), (123, 45), (7,), (300, 11)]
+    cpu_params = [torch.nn.Parameter(torch.randn(*s)) for s in shapes]
+    gpu_params = [torch.nn.Parameter(p.detach().clone().cuda())
+                  for p in cpu_params]
+    grads = [torch.randn(*s) for s in shapes]
+    cpu_opt = Adadelta([("p%d" % i, p) for i, p in enumerate(cpu_params)],
+                       clip_c=1.0)
+    gpu_opt = FusedAdadelta([("p%d" % i, p)
+                             for i, p in enumerate(gpu_params)], clip_c=1.0)
+    for it in range(3):
+        for p, g in zip(cpu_params, grads):
+            p.grad = (g * (it + 1)).clone()
+        for p, g in zip(gpu_params, grads):
+            p.grad = (g * (it + 1)).clone().cuda()
+        n_cpu = cpu_opt.step()
+        n_gpu = gpu_opt.step()
+        assert abs(float(n_cpu) - float(n_gpu)) < 1e-2 * float(n_cpu)
+    for pc, pg in zip(cpu_params, gpu_params):
+        torch.testing.assert_close(pg.detach().cpu(), pc.detach(),
+                                   rtol=1e-4, atol=1e-6)
