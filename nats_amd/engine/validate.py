@@ -30,9 +30,10 @@ def pred_probs(model, iterator, device=None, verbose=False):
             x, x_mask = x.to(device), x_mask.to(device)
             y, y_mask = y.to(device), y_mask.to(device)
         cost = model(x, x_mask, y, y_mask)
-        probs.extend(float(c) for c in cost.cpu())
-        if not all(math.isfinite(p) for p in probs):
+        batch_costs = [float(c) for c in cost.cpu()]
+        if not all(math.isfinite(p) for p in batch_costs):
             raise FloatingPointError("NaN/Inf in validation cost")
+        probs.extend(batch_costs)
         if verbose:
             print("%d samples computed" % n_done)
     return numpy.array(probs, dtype="float32")

@@ -480,9 +480,9 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_scatter(
   daccA[(long)b * Ts + s] += daccA_add;
 }
 
-// attention backward, stage 3 (one WG per b): reduce over s (contiguous
-// pc_buf rows) -> dpstate / dD_wei / dU_att / dc_att. de and accA_used
-// staged in LDS once per block.
+// attention backward, stage 3 (grid (b, s-chunk)): reduce over an
+// s-range of the contiguous pc_buf rows -> dpstate / dD_wei / dU_att /
+// dc_att by atomic combine. de and accA_used for the range staged in LDS.
 __global__ __launch_bounds__(256) void cond_attn_bwd_reduce(
     const float* __restrict__ alphas_t,     // [B][Ts]
     const float* __restrict__ dal_buf,      // [Ts][B]
@@ -495,19 +495,22 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_reduce(
     float* __restrict__ gdDwei,             // [A] (atomic)
     float* __restrict__ gdUatt,             // [A] (atomic)
     float* __restrict__ gdcatt,             // [1] (atomic)
-    int B, int Ts, int A, int Apad, int Tpad8) {
+    int B, int Ts, int A, int Apad, int Tpad8, int SCH) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* sm_de = (float*)smem_raw;      // [Ts]
-  float* sm_au = sm_de + Ts;            // [Ts]
+  const int chunk = (Ts + SCH - 1) / SCH;
+  float* sm_de = (float*)smem_raw;      // [chunk]
+  float* sm_au = sm_de + chunk;         // [chunk]
   __shared__ float red[256 / NATS_WAVE];
   const int b = blockIdx.x;
+  const int sbeg = blockIdx.y * chunk;
+  const int send = min(Ts, sbeg + chunk);
   const float dot = dot_buf[b];
   float dc = 0.f;
-  for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
+  for (int s = sbeg + threadIdx.x; s < send; s += blockDim.x) {
     const float de = alphas_t[(long)b * Ts + s] *
                      (dal_buf[(long)s * B + b] - dot);
-    sm_de[s] = de;
-    sm_au[s] = accA_used_t[(long)b * Ts + s];
+    sm_de[s - sbeg] = de;
+    sm_au[s - sbeg] = accA_used_t[(long)b * Ts + s];
     dc += de;
   }
 #pragma unroll
@@ -523,32 +526,53 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_reduce(
     const bf16_t* prow = pc_buf + ((long)b * A + i) * Tpad8;
     const float ua = Uatt[i];
     float sps = 0.f, sdw = 0.f, sua = 0.f;
-    int s = 0;
-    const int T8 = Ts & ~7;
-    for (; s < T8; s += 8) {
+    int s = sbeg;
+    const int a8beg = (sbeg + 7) & ~7;
+    const int a8end = send & ~7;
+    for (; s < min(a8beg, send); ++s) {
+      const float pc = (float)prow[s];
+      const float de = sm_de[s - sbeg];
+      const float dpc = de * (1.f - pc * pc) * ua;
+      sps += dpc;
+      sdw += dpc * sm_au[s - sbeg];
+      sua += de * pc;
+    }
+    for (; s + 8 <= a8end; s += 8) {
       bf16x8 pv = *(const bf16x8*)(prow + s);
 #pragma unroll
       for (int k = 0; k < 8; ++k) {
         const float pc = (float)pv[k];
-        const float de = sm_de[s + k];
+        const float de = sm_de[s + k - sbeg];
         const float dpc = de * (1.f - pc * pc) * ua;
         sps += dpc;
-        sdw += dpc * sm_au[s + k];
+        sdw += dpc * sm_au[s + k - sbeg];
         sua += de * pc;
       }
     }
-    for (; s < Ts; ++s) {
+    for (; s < send; ++s) {
       const float pc = (float)prow[s];
-      const float de = sm_de[s];
+      const float de = sm_de[s - sbeg];
       const float dpc = de * (1.f - pc * pc) * ua;
       sps += dpc;
-      sdw += dpc * sm_au[s];
+      sdw += dpc * sm_au[s - sbeg];
       sua += de * pc;
     }
-    dpstate_t[(long)b * A + i] = sps;
-    dstep_att[(long)b * Apad + i] = (bf16_t)sps;
+    atomicAdd(&dpstate_t[(long)b * A + i], sps);
     atomicAdd(&gdDwei[i], sdw);
     atomicAdd(&gdUatt[i], sua);
+  }
+}
+
+// bf16 copy of the reduced dpstate into the padded GEMM operand
+__global__ void cond_dpstate_cast(const float* __restrict__ dpstate_t,
+                                  bf16_t* __restrict__ dstep_att, int Apad,
+                                  int B, int A) {
+  const long total = (long)B * A;
+  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int b = idx / A;
+    const int i = idx % A;
+    dstep_att[(long)b * Apad + i] = (bf16_t)dpstate_t[idx];
   }
 }
 
@@ -795,12 +819,18 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        dctxs_p ? dctxs_p + (long)t * B * C : nullptr,
                        daccC.data_ptr<float>(), mt,
                        dctx_dir.data_ptr<float>(), B, C);
-    // b3: dctx = dctx_dir + [dpr2|dpu2|dpx2] @ [W_1|Wx_1]^T
-    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpC), dim3(384), 0,
-                       stream, (const bf16_t*)dstepC.data_ptr(),
+    // b2+b3 fused launch: dh1 = ddirect_h1 + dstep1 @ [U_1|Ux_1]^T and
+    // dctx = dctx_dir + dstepC @ [W_1|Wx_1]^T (independent problems)
+    hipLaunchKernelGGL(nats_gru_step_bwd_gemm_dual,
+                       dim3(std::max(ngrpH, ngrpC), 2), dim3(384), 0, stream,
+                       (const bf16_t*)dstep1.data_ptr(),
+                       (const bf16_t*)U1cat.data_ptr(),
+                       ddirect_h1.data_ptr<float>(),
+                       dh1_buf.data_ptr<float>(), H, K3Hpad,
+                       (const bf16_t*)dstepC.data_ptr(),
                        (const bf16_t*)W1cat.data_ptr(),
                        dctx_dir.data_ptr<float>(), dctx_buf.data_ptr<float>(),
-                       B, C, K3Hpad);
+                       C, K3Hpad, B);
     // b4: distraction gate backward
     hipLaunchKernelGGL(cond_gate_bwd, dim3(pwC), dim3(256), 0, stream,
                        dctx_buf.data_ptr<float>(),
@@ -835,8 +865,12 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        Dwei.data_ptr<float>(), Uatt.data_ptr<float>(),
                        daccA.data_ptr<float>(), dpctx_acc.data_ptr<float>(),
                        (bf16_t*)pc_buf.data_ptr(), B, Ts, A, Tpad8);
-    hipLaunchKernelGGL(cond_attn_bwd_reduce, dim3(B), dim3(256),
-                       2 * Ts * sizeof(float), stream,
+    HIP_CHECK(hipMemsetAsync(
+        dpstate_all.data_ptr<float>() + (long)t * B * A, 0,
+        (size_t)B * A * sizeof(float), stream));
+    const int RSCH = std::max(1, std::min(4, Ts / 128));
+    hipLaunchKernelGGL(cond_attn_bwd_reduce, dim3(B, RSCH), dim3(256),
+                       2 * ((Ts + RSCH - 1) / RSCH) * sizeof(float), stream,
                        alphas_all.data_ptr<float>() + (long)t * B * Ts,
                        dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
                        accA_used.data_ptr<float>() + (long)t * B * Ts,
@@ -845,13 +879,12 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        dpstate_all.data_ptr<float>() + (long)t * B * A,
                        (bf16_t*)dstep_att.data_ptr(),
                        gdDwei.data_ptr<float>(), gdUatt.data_ptr<float>(),
-                       gdcatt.data_ptr<float>(), B, Ts, A, Apad32, Tpad8);
-    // b2: dh1 += [dpr2|dpu2|dpxa_lin] @ [U_1|Ux_1]^T + passthrough
-    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH), dim3(384), 0,
-                       stream, (const bf16_t*)dstep1.data_ptr(),
-                       (const bf16_t*)U1cat.data_ptr(),
-                       ddirect_h1.data_ptr<float>(),
-                       dh1_buf.data_ptr<float>(), B, H, K3Hpad);
+                       gdcatt.data_ptr<float>(), B, Ts, A, Apad32, Tpad8,
+                       RSCH);
+    hipLaunchKernelGGL(cond_dpstate_cast, dim3(cdiv_i(B * A, 256)), dim3(256),
+                       0, stream,
+                       dpstate_all.data_ptr<float>() + (long)t * B * A,
+                       (bf16_t*)dstep_att.data_ptr(), Apad32, B, A);
     // b6a: dh1 += dpstate @ W_att^T (in-place add via ddirect aliasing)
     hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH), dim3(384), 0,
                        stream, (const bf16_t*)dstep_att.data_ptr(),
