@@ -425,10 +425,19 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
   const int j0 = wg * JB;
   const long hb = (long)32 * Hpad;
 
+  // register-carry of this thread's own h columns (column-local: the same
+  // thread wrote them last step) — removes an L2 round trip per step
+  float h_keep[2];
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    const int idx = threadIdx.x + it * blockDim.x;
+    const int b = idx / JB;
+    const int j = j0 + idx % JB;
+    h_keep[it] = (idx < B * JB && j < H) ? p.h0[(long)b * H + j] : 0.f;
+  }
+
   for (int t = 0; t < T; ++t) {
     const bf16_t* h_bf_in = p.h_bf + (t % 2) * hb;
-    const float* h_prev =
-        (t == 0) ? p.h0 : (p.h_all + (long)(t - 1) * B * H);
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
     NATS_MFMA_KLOOP_LDSB(acc, h_bf_in, 16 * m, Hpad, upk_lds, g * JB, Hpad,
                          0, Hpad);
@@ -447,12 +456,15 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
     float* h_out = p.h_all + (long)t * B * H;
     bf16_t* h_bf_out = p.h_bf + ((t + 1) % 2) * hb;
     bf16_t* saved_t = p.saved + (long)t * B * 3 * H;
-    for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int idx = threadIdx.x + it * blockDim.x;
+      if (idx >= B * JB) continue;
       const int b = idx / JB;
       const int c = idx % JB;
       const int j = j0 + c;
       if (j >= H) continue;
-      const float hp = h_prev[(long)b * H + j];
+      const float hp = h_keep[it];
       const float pr = pre[0][b][c] + (float)xg_t[(long)b * 2 * H + j];
       const float pu = pre[1][b][c] + (float)xg_t[(long)b * 2 * H + H + j];
       const float px = pre[2][b][c];
@@ -464,6 +476,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
         const float mm = mask_t[b];
         hnew = mm * hnew + (1.f - mm) * hp;
       }
+      h_keep[it] = hnew;
       h_out[(long)b * H + j] = hnew;
       h_bf_out[(long)b * Hpad + j] = (bf16_t)hnew;
       saved_t[(long)b * 3 * H + j] = (bf16_t)r;
@@ -515,6 +528,10 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
   const int kend = min(K3pad, (ks + 1) * kchunk);
   const long ds = (long)32 * K3pad;
 
+  // register-carry of this thread's ddirect columns (column-local; the
+  // global copy is still written — the host's final dh0 GEMM reads it)
+  float dd_keep[2] = {0.f, 0.f};
+
   for (int t = T - 1; t >= 0; --t) {
     const bf16_t* dstep_in = p.dstep + ((t + 1) % 2) * ds;
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
@@ -537,13 +554,16 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
     const float* dh_out_t = p.dh_out + (long)t * B * H;
     bf16_t* dstep_out = p.dstep + (t % 2) * ds;
     bf16_t* dpre_t = p.dpre + (long)t * B * 4 * H;
-    for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int idx = threadIdx.x + it * blockDim.x;
+      if (idx >= B * JB) continue;
       const int b = idx / JB;
       const int c = idx % JB;
       const int j = i0 + c;
       if (j >= H) continue;
       const long bj = (long)b * H + j;
-      float dh = p.ddirect[bj] + part[0][b][c] + part[1][b][c] +
+      float dh = dd_keep[it] + part[0][b][c] + part[1][b][c] +
                  part[2][b][c] + dh_out_t[bj];
       const float r = (float)saved_t[(long)b * 3 * H + j];
       const float u = (float)saved_t[(long)b * 3 * H + H + j];
@@ -558,7 +578,9 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
       const float dr = dpx * px;
       const float dpr = dr * r * (1.f - r);
       const float dpu = du * u * (1.f - u);
-      p.ddirect[bj] = dh * (mm * u + (1.f - mm));
+      const float dd = dh * (mm * u + (1.f - mm));
+      dd_keep[it] = dd;
+      p.ddirect[bj] = dd;
       dstep_out[(long)b * K3pad + j] = (bf16_t)dpr;
       dstep_out[(long)b * K3pad + H + j] = (bf16_t)dpu;
       dstep_out[(long)b * K3pad + 2 * H + j] = (bf16_t)dpxl;
