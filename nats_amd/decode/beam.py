@@ -78,6 +78,13 @@ def distraction_penalties_gpu(hist_a, hist_c, hist_s, cur_a, cur_c, cur_s,
 
     hist_* are (n_hist, live_k, dim) fp32 device tensors; cur_* (live_k,
     dim). Returns a (live_k,) fp32 tensor of summed penalties."""
+    from nats_amd.ops import _hip_ext
+    ext = _hip_ext()
+    if ext is not None and hist_a.is_cuda and hasattr(ext, "rerank_penalties"):
+        return ext.rerank_penalties(
+            hist_a.contiguous(), hist_c.contiguous(), hist_s.contiguous(),
+            cur_a.contiguous(), cur_c.contiguous(), cur_s.contiguous(),
+            kl_factor, ctx_factor, state_factor)
     p = hist_a / hist_a.sum(-1, keepdim=True)
     q = (cur_a / cur_a.sum(-1, keepdim=True)).unsqueeze(0)
     logterm = torch.where(p > 0, p * (torch.log(p) - torch.log(q)),

@@ -18,6 +18,7 @@ SOURCES = [
     os.path.join(HERE, "softmax_ce.hip"),
     os.path.join(HERE, "mfma_test.hip"),
     os.path.join(HERE, "optim.hip"),
+    os.path.join(HERE, "rerank.hip"),
 ]
 
 
