@@ -364,32 +364,40 @@ __device__ __forceinline__ bf16x8 frag_bt_lds_swz(const bf16_t* lds, int row,
   return *(const bf16x8*)((const char*)lds + byte);
 }
 
-// swizzled-LDS MFMA K-loop (B operand from LDS, A from global), 2-deep on A
+// swizzled-LDS MFMA K-loop (B operand from LDS, A from global), 4-deep on
+// the A side: four global fragments stay in flight ahead of each MFMA so
+// the post-acquire (L1-cold) L2 latency amortises over 4 iterations.
 #define NATS_MFMA_KLOOP_LDSB(ACC, APTR, AROW, ALD, LDSB, BROW, BLD, KBEG,    \
                              KEND)                                           \
   do {                                                                       \
-    int _k = (KBEG);                                                         \
     const int _ke = (KEND);                                                  \
-    if (_k + 32 >= _ke) {                                                    \
-      if (_k < _ke) {                                                        \
+    int _k = (KBEG);                                                         \
+    if (_k + 128 <= _ke) {                                                   \
+      bf16x8 _a0 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);               \
+      bf16x8 _a1 = frag_a_rowmajor((APTR), (AROW), (ALD), _k + 32);          \
+      bf16x8 _a2 = frag_a_rowmajor((APTR), (AROW), (ALD), _k + 64);          \
+      bf16x8 _a3 = frag_a_rowmajor((APTR), (AROW), (ALD), _k + 96);          \
+      for (_k += 128; _k + 32 <= _ke; _k += 32) {                            \
+        bf16x8 _an = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
+        bf16x8 _b = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 128);        \
+        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b, ACC, 0, 0, 0);\
+        _a0 = _a1; _a1 = _a2; _a2 = _a3; _a3 = _an;                          \
+      }                                                                      \
+      {                                                                      \
+        bf16x8 _b = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 128);        \
+        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b, ACC, 0, 0, 0);\
+        _b = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 96);                \
+        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a1, _b, ACC, 0, 0, 0);\
+        _b = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 64);                \
+        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a2, _b, ACC, 0, 0, 0);\
+        _b = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 32);                \
+        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a3, _b, ACC, 0, 0, 0);\
+      }                                                                      \
+    } else {                                                                 \
+      for (; _k < _ke; _k += 32) {                                           \
         bf16x8 _a0 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
         bf16x8 _b0 = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k);             \
         ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0);\
-      }                                                                      \
-    } else {                                                                 \
-      bf16x8 _a0 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);               \
-      bf16x8 _a1 = frag_a_rowmajor((APTR), (AROW), (ALD), _k + 32);          \
-      for (_k += 64; _k < _ke; _k += 32) {                                   \
-        bf16x8 _a2 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
-        bf16x8 _b0 = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 64);        \
-        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0);\
-        _a0 = _a1; _a1 = _a2;                                                \
-      }                                                                      \
-      {                                                                      \
-        bf16x8 _b0 = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 64);        \
-        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0);\
-        bf16x8 _b1 = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 32);        \
-        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a1, _b1, ACC, 0, 0, 0);\
       }                                                                      \
     }                                                                        \
   } while (0)
