@@ -203,10 +203,12 @@ __global__ void cond_attn_gate_fwd(
 }
 
 // ---------------- GRU_1 forward ----------------
-// 8 waves: wave w -> (m = w/4, g = w%4) with output groups
+// 16 waves: wave w -> (m = w/8, g = (w%8)/2, ks = w%2) with output groups
 // g0 = r2 (K = h1|ctx), g1 = u2 (K = h1|ctx), g2 = pxa (h1@Ux_1),
 // g3 = pxb (ctx@Wx_1) — K-selectivity comes from zero blocks in W1pk.
-__global__ __launch_bounds__(512) void cond_gru1_step_fwd(
+// K is split across wave pairs (K1 is ~3H — the serial chain at 8 waves
+// measured 34 us/launch, latency-bound).
+__global__ __launch_bounds__(1024) void cond_gru1_step_fwd(
     const bf16_t* __restrict__ hc_bf,  // [32][K1] = [h1 | ctx_t] bf16
     const float* __restrict__ h1_t,    // [B][H] fp32
     const bf16_t* __restrict__ W1pk,   // [ngrp*4*16][K1]
@@ -218,23 +220,27 @@ __global__ __launch_bounds__(512) void cond_gru1_step_fwd(
     int ld_h2bf,
     bf16_t* __restrict__ saved1_t,     // [B][4H] (r2,u2,pxa,hbar)
     int B, int H, int K1) {
-  __shared__ float pre[4][32][JB + 1];
+  __shared__ float pre[4][2][32][JB + 1];
 
   const int wg = blockIdx.x;
   const int wave = threadIdx.x / NATS_WAVE;
-  const int m = wave / 4;
-  const int g = wave % 4;
+  const int m = wave / 8;
+  const int g = (wave % 8) / 2;
+  const int ks = wave % 2;
   const int j0 = wg * JB;
+  const int khalf = ((K1 / 2 + 31) / 32) * 32;
+  const int kbeg = ks * khalf;
+  const int kend = min(K1, (ks + 1) * khalf);
 
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   const bf16_t* brow = W1pk + (long)(wg * 4 + g) * JB * K1;
-  NATS_MFMA_KLOOP(acc, hc_bf, 16 * m, K1, brow, 0, K1, 0, K1);
+  NATS_MFMA_KLOOP(acc, hc_bf, 16 * m, K1, brow, 0, K1, kbeg, kend);
   {
     const int lane = threadIdx.x & (NATS_WAVE - 1);
     const int col = lane & 15;
     const int rbase = 16 * m + (lane >> 4) * 4;
 #pragma unroll
-    for (int i = 0; i < 4; ++i) pre[g][rbase + i][col] = acc[i];
+    for (int i = 0; i < 4; ++i) pre[g][ks][rbase + i][col] = acc[i];
   }
   __syncthreads();
 
@@ -243,10 +249,12 @@ __global__ __launch_bounds__(512) void cond_gru1_step_fwd(
     const int c = idx % JB;
     const int j = j0 + c;
     if (j >= H) continue;
-    const float r2 = nats_sigmoid(pre[0][b][c] + b1[j]);
-    const float u2 = nats_sigmoid(pre[1][b][c] + b1[H + j]);
-    const float pxa = pre[2][b][c];
-    const float pxb = pre[3][b][c];
+    const float r2 =
+        nats_sigmoid(pre[0][0][b][c] + pre[0][1][b][c] + b1[j]);
+    const float u2 =
+        nats_sigmoid(pre[1][0][b][c] + pre[1][1][b][c] + b1[H + j]);
+    const float pxa = pre[2][0][b][c] + pre[2][1][b][c];
+    const float pxb = pre[3][0][b][c] + pre[3][1][b][c];
     const float hbar = tanhf((pxa + bx1[j]) * r2 + pxb);
     const float h1v = h1_t[(long)b * H + j];
     float h2 = u2 * h1v + (1.f - u2) * hbar;
@@ -698,7 +706,7 @@ std::vector<torch::Tensor> cond_gru_fwd(
                        ctxs_all.data_ptr<float>() + (long)t * B * C,
                        (bf16_t*)hc_bf.data_ptr(), Hpad, K1, mt, B, C);
     // 5) GRU_1 -> h2
-    hipLaunchKernelGGL(cond_gru1_step_fwd, dim3(ngrpH), dim3(512), 0, stream,
+    hipLaunchKernelGGL(cond_gru1_step_fwd, dim3(ngrpH), dim3(1024), 0, stream,
                        (const bf16_t*)hc_bf.data_ptr(),
                        h1_all.data_ptr<float>() + (long)t * B * H,
                        (const bf16_t*)W1pk.data_ptr(), b1.data_ptr<float>(),

@@ -1,6 +1,6 @@
 """Manual GPU bisect for cond_gru_fwd: diff every per-step intermediate
 against the eager oracle. Run on a GPU box:
-    python tests/gpu_debug_cond.py
+    python tools/gpu_debug_cond.py
 """
 import os
 import sys
