@@ -89,3 +89,40 @@ def test_prepare_data_truncation():
     assert x.shape[0] == 5  # 4 kept tokens + eos slot
     assert x_mask[:, 0].sum() == 5
     assert y.shape[0] == 4
+
+
+def test_iterator_gzip(tmp_path):
+    import gzip
+    src = tmp_path / "s.txt.gz"
+    tgt = tmp_path / "t.txt.gz"
+    with gzip.open(src, "wt") as f:
+        f.write("a b\nb a\n")
+    with gzip.open(tgt, "wt") as f:
+        f.write("b\na\n")
+    d = dictionary_from_freqs({"a": 2, "b": 2})
+    dic = tmp_path / "d.pkl"
+    with open(dic, "wb") as f:
+        pickle.dump(d, f)
+    it = TextIterator(str(src), str(tgt), str(dic), batch_size=4)
+    s, t = next(it)
+    assert len(s) == 2 and len(t) == 2
+
+
+def test_gen_map_line_char_level():
+    from nats_amd.decode.driver import map_line
+    d = {"a": 2, "b": 3, "ab": 4}
+    # word level: "ab" is one token
+    assert map_line("ab", d, 100) == [4, 0]
+    # char level (-c): per-character ids, unknown -> 1
+    assert map_line("ab", d, 100, chr_level=True) == [2, 3, 0]
+    # n_words cutoff -> UNK
+    assert map_line("b", d, 3) == [1, 0]
+
+
+def test_synthetic_batch_layout():
+    from nats_amd.data.synthetic import synthetic_batch
+    rng = numpy.random.RandomState(0)
+    x, xm, y, ym = synthetic_batch(rng, 3, 10, 4, 50)
+    assert x.shape == (11, 3) and y.shape == (5, 3)
+    assert (x[:10] >= 2).all() and (x[10] == 0).all()
+    assert xm.dtype == numpy.float32 and (xm == 1).all()
