@@ -10,16 +10,21 @@ make_toy_corpus("data")
 EOF
 mkdir -p models
 
-python - <<'EOF'
+FINISH=${FINISH:-5} DIM=${DIM:-32} python - <<'EOF'
+import os
 from nats_amd.engine.trainer import train
-train(dim_word=16, dim=32, dim_att=8, n_words=64, maxlen=50, batch_size=8,
-      valid_batch_size=8, saveto="models/model.npz",
+finish = int(os.environ.get("FINISH", "5"))
+dim = int(os.environ.get("DIM", "32"))
+train(dim_word=dim // 2, dim=dim, dim_att=max(8, dim // 4), n_words=64,
+      maxlen=50, batch_size=8, valid_batch_size=8,
+      saveto="models/model.npz",
       datasets=["data/toy_train_input.txt", "data/toy_train_output.txt"],
       valid_datasets=["data/toy_validation_input.txt",
                       "data/toy_validation_output.txt"],
       dictionary="data/toy_train_input.txt.pkl",
-      validFreq=1000, saveFreq=5, sampleFreq=1000, dispFreq=100,
-      finish_after=5, clip_c=1.0, seed=3)
+      validFreq=max(finish // 3, 5), saveFreq=max(finish // 3, 5),
+      sampleFreq=10 ** 9, dispFreq=max(finish // 10, 1),
+      finish_after=finish, clip_c=1.0, seed=3)
 EOF
 
 python scripts/gen.py -n -p ${NPROC:-2} -k 5 models/model.npz \
