@@ -9,8 +9,6 @@ used CPU workers, test.sh:3). Output format is the reference's
 "word [srcpos]" interleave (gen.py:88-98), consumed by replace_unk.
 """
 
-import os
-
 import numpy
 import torch
 

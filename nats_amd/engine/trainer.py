@@ -28,10 +28,9 @@ import torch
 from ..data.dictionary import load_dictionary, invert_dictionary
 from ..data.iterator import TextIterator
 from ..data.prepare import prepare_data
-from ..models.distraction import NatsModel, default_options
+from ..models.distraction import NatsModel
 from ..parallel.ddp import DataParallelGrads, init_distributed
-from .checkpoint import (load_checkpoint, save_checkpoint, load_options,
-                         save_options)
+from .checkpoint import load_checkpoint, save_checkpoint, load_options
 from .optim import build_optimizer
 from .validate import pred_probs
 

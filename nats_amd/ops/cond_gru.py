@@ -8,7 +8,6 @@ via the returned dpctx).
 """
 
 import torch
-import torch.nn.functional as F
 
 from . import _hip_ext
 from .gru import pack_fwd_weights, _pad_to

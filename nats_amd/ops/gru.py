@@ -7,7 +7,6 @@ split the reference's Theano graph implied (scan grads + batched dots,
 nats.py:1340).
 """
 
-import math
 
 import torch
 import torch.nn.functional as F

@@ -6,7 +6,6 @@ to GPU machines) by ``__graft_entry__.build()`` or
 imports it; it never builds.
 """
 
-import importlib
 import os
 
 
