@@ -435,15 +435,50 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
   // register-carry of this thread's own h columns (column-local: the same
   // thread wrote them last step) — removes an L2 round trip per step
   float h_keep[2];
+  int own_b[2], own_j[2];
+  bool own[2];
 #pragma unroll
   for (int it = 0; it < 2; ++it) {
     const int idx = threadIdx.x + it * blockDim.x;
-    const int b = idx / JB;
-    const int j = j0 + idx % JB;
-    h_keep[it] = (idx < B * JB && j < H) ? p.h0[(long)b * H + j] : 0.f;
+    own_b[it] = idx / JB;
+    own_j[it] = j0 + idx % JB;
+    own[it] = (idx < B * JB) && (own_j[it] < H);
+    h_keep[it] = own[it] ? p.h0[(long)own_b[it] * H + own_j[it]] : 0.f;
   }
+  // deferred saved-writes (not read in-kernel: flushed during the NEXT
+  // step so their store latency never sits inside the barrier drain)
+  float pend_r[2], pend_u[2], pend_px[2];
+  bool have_pend = false;
+  bf16_t* pend_saved = nullptr;
 
   for (int t = 0; t < T; ++t) {
+    // prefetch this step's inputs; latency hides under the MFMA phase
+    const bf16_t* xg_t = p.xg + (long)t * B * 2 * H;
+    const bf16_t* xc_t = p.xc + (long)t * B * H;
+    float pf_xr[2], pf_xu[2], pf_xc[2];
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      if (own[it]) {
+        const long b2 = (long)own_b[it] * 2 * H;
+        pf_xr[it] = (float)xg_t[b2 + own_j[it]];
+        pf_xu[it] = (float)xg_t[b2 + H + own_j[it]];
+        pf_xc[it] = (float)xc_t[(long)own_b[it] * H + own_j[it]];
+      }
+    }
+    // flush the previous step's saved gates (plenty of slack before the
+    // next barrier drain)
+    if (have_pend) {
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        if (own[it]) {
+          const long s3 = (long)own_b[it] * 3 * H + own_j[it];
+          pend_saved[s3] = (bf16_t)pend_r[it];
+          pend_saved[s3 + H] = (bf16_t)pend_u[it];
+          pend_saved[s3 + 2 * H] = (bf16_t)pend_px[it];
+        }
+      }
+    }
+
     const bf16_t* h_bf_in = p.h_bf + (t % 2) * hb;
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
     NATS_MFMA_KLOOP_LDSB(acc, h_bf_in, 16 * m, Hpad, upk_lds, g * JB, Hpad,
@@ -457,27 +492,22 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
     }
     __syncthreads();
 
-    const bf16_t* xg_t = p.xg + (long)t * B * 2 * H;
-    const bf16_t* xc_t = p.xc + (long)t * B * H;
     const float* mask_t = p.mask ? p.mask + (long)t * B : nullptr;
     float* h_out = p.h_all + (long)t * B * H;
     bf16_t* h_bf_out = p.h_bf + ((t + 1) % 2) * hb;
-    bf16_t* saved_t = p.saved + (long)t * B * 3 * H;
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
-      const int idx = threadIdx.x + it * blockDim.x;
-      if (idx >= B * JB) continue;
-      const int b = idx / JB;
-      const int c = idx % JB;
-      const int j = j0 + c;
-      if (j >= H) continue;
+      if (!own[it]) continue;
+      const int b = own_b[it];
+      const int c = (threadIdx.x + it * blockDim.x) % JB;
+      const int j = own_j[it];
       const float hp = h_keep[it];
-      const float pr = pre[0][b][c] + (float)xg_t[(long)b * 2 * H + j];
-      const float pu = pre[1][b][c] + (float)xg_t[(long)b * 2 * H + H + j];
+      const float pr = pre[0][b][c] + pf_xr[it];
+      const float pu = pre[1][b][c] + pf_xu[it];
       const float px = pre[2][b][c];
       const float r = nats_sigmoid(pr);
       const float u = nats_sigmoid(pu);
-      const float hbar = tanhf(px * r + (float)xc_t[(long)b * H + j]);
+      const float hbar = tanhf(px * r + pf_xc[it]);
       float hnew = u * hp + (1.f - u) * hbar;
       if (mask_t != nullptr) {
         const float mm = mask_t[b];
@@ -486,16 +516,30 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
       h_keep[it] = hnew;
       h_out[(long)b * H + j] = hnew;
       h_bf_out[(long)b * Hpad + j] = (bf16_t)hnew;
-      saved_t[(long)b * 3 * H + j] = (bf16_t)r;
-      saved_t[(long)b * 3 * H + H + j] = (bf16_t)u;
-      saved_t[(long)b * 3 * H + 2 * H + j] = (bf16_t)px;
+      pend_r[it] = r;
+      pend_u[it] = u;
+      pend_px[it] = px;
     }
+    pend_saved = p.saved + (long)t * B * 3 * H;
+    have_pend = true;
     if (unsafe_nobarrier) {  // TIMING EXPERIMENTS ONLY (racy!)
       __syncthreads();
     } else if (!nats_grid_barrier(sync, (unsigned)(t + 1), nwg)) {
       // poison output so a barrier give-up surfaces as NaN, never a hang
       if (threadIdx.x == 0) p.h_all[0] = __builtin_nanf("");
       return;
+    }
+  }
+  // flush the last step's saved gates
+  if (have_pend) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      if (own[it]) {
+        const long s3 = (long)own_b[it] * 3 * H + own_j[it];
+        pend_saved[s3] = (bf16_t)pend_r[it];
+        pend_saved[s3 + H] = (bf16_t)pend_u[it];
+        pend_saved[s3 + 2 * H] = (bf16_t)pend_px[it];
+      }
     }
   }
 }
