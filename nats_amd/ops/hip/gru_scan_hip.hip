@@ -583,8 +583,54 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
   // register-carry of this thread's ddirect columns (column-local; the
   // global copy is still written — the host's final dh0 GEMM reads it)
   float dd_keep[2] = {0.f, 0.f};
+  int own_b[2], own_j[2];
+  bool own[2];
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    const int idx = threadIdx.x + it * blockDim.x;
+    own_b[it] = idx / JB;
+    own_j[it] = i0 + idx % JB;
+    own[it] = (idx < B * JB) && (own_j[it] < H);
+  }
+  // deferred dpre writes (only consumed by the host's time-batched GEMMs)
+  float pend[2][4];
+  bool have_pend = false;
+  bf16_t* pend_dpre = nullptr;
 
   for (int t = T - 1; t >= 0; --t) {
+    // prefetch this step's pointwise inputs under the MFMA phase
+    const bf16_t* saved_t = p.saved + (long)t * B * 3 * H;
+    const bf16_t* xc_t = p.xc + (long)t * B * H;
+    const float* h_prev =
+        (t == 0) ? p.h0 : (p.h_all + (long)(t - 1) * B * H);
+    const float* dh_out_t = p.dh_out + (long)t * B * H;
+    float pf_r[2], pf_u[2], pf_px[2], pf_xc[2], pf_hp[2], pf_dho[2];
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      if (own[it]) {
+        const long bj = (long)own_b[it] * H + own_j[it];
+        const long s3 = (long)own_b[it] * 3 * H + own_j[it];
+        pf_r[it] = (float)saved_t[s3];
+        pf_u[it] = (float)saved_t[s3 + H];
+        pf_px[it] = (float)saved_t[s3 + 2 * H];
+        pf_xc[it] = (float)xc_t[bj];
+        pf_hp[it] = h_prev[bj];
+        pf_dho[it] = dh_out_t[bj];
+      }
+    }
+    if (have_pend) {
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        if (own[it]) {
+          const long d4 = (long)own_b[it] * 4 * H + own_j[it];
+          pend_dpre[d4] = (bf16_t)pend[it][0];
+          pend_dpre[d4 + H] = (bf16_t)pend[it][1];
+          pend_dpre[d4 + 2 * H] = (bf16_t)pend[it][2];
+          pend_dpre[d4 + 3 * H] = (bf16_t)pend[it][3];
+        }
+      }
+    }
+
     const bf16_t* dstep_in = p.dstep + ((t + 1) % 2) * ds;
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
     NATS_MFMA_KLOOP_LDSB(acc, dstep_in, 16 * m, K3pad, ub_lds, 0, K3pad,
@@ -598,30 +644,22 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
     }
     __syncthreads();
 
-    const bf16_t* saved_t = p.saved + (long)t * B * 3 * H;
-    const bf16_t* xc_t = p.xc + (long)t * B * H;
-    const float* h_prev =
-        (t == 0) ? p.h0 : (p.h_all + (long)(t - 1) * B * H);
     const float* mask_t = p.mask ? p.mask + (long)t * B : nullptr;
-    const float* dh_out_t = p.dh_out + (long)t * B * H;
     bf16_t* dstep_out = p.dstep + (t % 2) * ds;
-    bf16_t* dpre_t = p.dpre + (long)t * B * 4 * H;
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
-      const int idx = threadIdx.x + it * blockDim.x;
-      if (idx >= B * JB) continue;
-      const int b = idx / JB;
-      const int c = idx % JB;
-      const int j = i0 + c;
-      if (j >= H) continue;
+      if (!own[it]) continue;
+      const int b = own_b[it];
+      const int c = (threadIdx.x + it * blockDim.x) % JB;
+      const int j = own_j[it];
       const long bj = (long)b * H + j;
       float dh = dd_keep[it] + part[0][b][c] + part[1][b][c] +
-                 part[2][b][c] + dh_out_t[bj];
-      const float r = (float)saved_t[(long)b * 3 * H + j];
-      const float u = (float)saved_t[(long)b * 3 * H + H + j];
-      const float px = (float)saved_t[(long)b * 3 * H + 2 * H + j];
-      const float hbar = tanhf(px * r + (float)xc_t[bj]);
-      const float hp = h_prev[bj];
+                 part[2][b][c] + pf_dho[it];
+      const float r = pf_r[it];
+      const float u = pf_u[it];
+      const float px = pf_px[it];
+      const float hbar = tanhf(px * r + pf_xc[it]);
+      const float hp = pf_hp[it];
       const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
       const float du = dh * mm * (hp - hbar);
       const float dhbar = dh * mm * (1.f - u);
@@ -636,14 +674,28 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
       dstep_out[(long)b * K3pad + j] = (bf16_t)dpr;
       dstep_out[(long)b * K3pad + H + j] = (bf16_t)dpu;
       dstep_out[(long)b * K3pad + 2 * H + j] = (bf16_t)dpxl;
-      dpre_t[(long)b * 4 * H + j] = (bf16_t)dpr;
-      dpre_t[(long)b * 4 * H + H + j] = (bf16_t)dpu;
-      dpre_t[(long)b * 4 * H + 2 * H + j] = (bf16_t)dpx;
-      dpre_t[(long)b * 4 * H + 3 * H + j] = (bf16_t)dpxl;
+      pend[it][0] = dpr;
+      pend[it][1] = dpu;
+      pend[it][2] = dpx;
+      pend[it][3] = dpxl;
     }
+    pend_dpre = p.dpre + (long)t * B * 4 * H;
+    have_pend = true;
     if (!nats_grid_barrier(sync, (unsigned)(T - t), nwg)) {
       if (threadIdx.x == 0) p.ddirect[0] = __builtin_nanf("");
       return;
+    }
+  }
+  if (have_pend) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      if (own[it]) {
+        const long d4 = (long)own_b[it] * 4 * H + own_j[it];
+        pend_dpre[d4] = (bf16_t)pend[it][0];
+        pend_dpre[d4 + H] = (bf16_t)pend[it][1];
+        pend_dpre[d4 + 2 * H] = (bf16_t)pend[it][2];
+        pend_dpre[d4 + 3 * H] = (bf16_t)pend[it][3];
+      }
     }
   }
 }
