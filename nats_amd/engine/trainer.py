@@ -161,7 +161,7 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
                 logger.debug("Epoch {0} Update {1} Cost {2} UD {3}".format(
                     eidx, uidx, cost_val, ud))
                 if verbose and clip_c > 0:
-                    logger.debug("Grad {0}".format(norm_g))
+                    logger.debug("Grad {0}".format(float(norm_g)))
 
             if rank == 0 and numpy.mod(uidx, saveFreq) == 0:
                 print("Saving...", end=" ")
