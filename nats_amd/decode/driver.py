@@ -108,12 +108,29 @@ def generate_file(model_path, dictionary, source_file, saveto, k=5,
 
     if n_process <= 1:
         model = _make_model(model_path, options, devices[0])
-        for idx, x in jobs:
-            trans[idx], pos[idx] = _translate_one(
-                model, x, devices[0], k, normalize, kl_factor, ctx_factor,
-                state_factor)
-            if verbose and idx % 10 == 0:
-                print("Sample %d / %d Done" % (idx + 1, n_samples))
+        # batch several sentences' beams through each decode step (the
+        # fused decoder kernels take up to 32 rows)
+        sent_batch = max(1, 32 // max(k, 1))
+        from .batched import gen_sample_batched
+        for base in range(0, n_samples, sent_batch):
+            chunk = jobs[base:base + sent_batch]
+            xt = [torch.tensor(x, dtype=torch.int64,
+                               device=devices[0]).reshape(-1, 1)
+                  for _, x in chunk]
+            outs = gen_sample_batched(model, xt, k=k, maxlen=100,
+                                      use_unk=True, kl_factor=kl_factor,
+                                      ctx_factor=ctx_factor,
+                                      state_factor=state_factor)
+            for (idx, _), (sample, score, alphas) in zip(chunk, outs):
+                score = numpy.array(score)
+                if normalize:
+                    lengths = numpy.array([len(s2) for s2 in sample])
+                    score = score / lengths
+                sidx = int(numpy.argmin(score))
+                trans[idx] = sample[sidx]
+                pos[idx] = [int(numpy.argmax(a)) for a in alphas[sidx]]
+            if verbose and (base // sent_batch) % 4 == 0:
+                print("Sample %d / %d Done" % (base + 1, n_samples))
     else:
         import torch.multiprocessing as mp
         ctx = mp.get_context("spawn")

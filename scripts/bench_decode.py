@@ -20,15 +20,26 @@ from nats_amd.decode.beam import gen_sample
 from nats_amd.models.distraction import NatsModel, default_options
 
 
-def run(model, xs, k, maxlen, use_graph, lam):
+def run(model, xs, k, maxlen, use_graph, lam, batched=False):
     n_tokens = 0
     t0 = time.perf_counter()
-    for x in xs:
-        sample, score, alphas = gen_sample(
-            model, x, k=k, maxlen=maxlen, stochastic=False, use_unk=True,
-            kl_factor=lam, ctx_factor=lam, state_factor=lam,
-            use_graph=use_graph)
-        n_tokens += sum(len(s) for s in sample)
+    if batched:
+        from nats_amd.decode.batched import gen_sample_batched
+        sb = max(1, 32 // k)
+        for base in range(0, len(xs), sb):
+            outs = gen_sample_batched(model, xs[base:base + sb], k=k,
+                                      maxlen=maxlen, use_unk=True,
+                                      kl_factor=lam, ctx_factor=lam,
+                                      state_factor=lam)
+            for sample, _, _ in outs:
+                n_tokens += sum(len(s) for s in sample)
+    else:
+        for x in xs:
+            sample, score, alphas = gen_sample(
+                model, x, k=k, maxlen=maxlen, stochastic=False, use_unk=True,
+                kl_factor=lam, ctx_factor=lam, state_factor=lam,
+                use_graph=use_graph)
+            n_tokens += sum(len(s) for s in sample)
     if torch.cuda.is_available():
         torch.cuda.synchronize()
     dt = time.perf_counter() - t0
@@ -55,13 +66,15 @@ def main():
         dtype=torch.int64, device=device).reshape(-1, 1)
         for _ in range(args.n)]
 
-    variants = [("beam", False, 0.0), ("beam+distraction", False, 0.5)]
+    variants = [("beam", False, 0.0, False),
+                ("beam+distraction", False, 0.5, False),
+                ("beam+batched", False, 0.0, True),
+                ("beam+distraction+batched", False, 0.5, True)]
     if device == "cuda":
-        variants.append(("beam+hipgraph", True, 0.0))
-        variants.append(("beam+distraction+hipgraph", True, 0.5))
-    for name, graph, lam in variants:
-        run(model, xs[:2], args.k, args.maxlen, graph, lam)  # warmup
-        sps, ntok = run(model, xs, args.k, args.maxlen, graph, lam)
+        variants.append(("beam+hipgraph", True, 0.0, False))
+    for name, graph, lam, batched in variants:
+        run(model, xs[:2], args.k, args.maxlen, graph, lam, batched)
+        sps, ntok = run(model, xs, args.k, args.maxlen, graph, lam, batched)
         print(json.dumps({
             "metric": "summaries_per_sec", "variant": name, "value": sps,
             "beam": args.k, "n": args.n, "src_len": args.src,
