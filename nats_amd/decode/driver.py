@@ -43,16 +43,27 @@ def _translate_one(model, seq, device, k, normalize, kl_factor, ctx_factor,
 
 def translate_worker(queue, rqueue, pid, model_path, options, k, normalize,
                      kl_factor, ctx_factor, state_factor, device):
-    """Worker loop (translate_model, gen.py:15-58)."""
+    """Worker loop (translate_model, gen.py:15-58). Jobs arrive as CHUNKS
+    of (idx, ids) pairs and are decoded jointly (batched beams)."""
+    from .batched import gen_sample_batched
     model = _make_model(model_path, options, device)
     while True:
         req = queue.get()
         if req is None:
             break
-        idx, x = req
-        seq, pos = _translate_one(model, x, device, k, normalize, kl_factor,
-                                  ctx_factor, state_factor)
-        rqueue.put((idx, seq, pos))
+        xt = [torch.tensor(x, dtype=torch.int64, device=device).reshape(-1, 1)
+              for _, x in req]
+        outs = gen_sample_batched(model, xt, k=k, maxlen=100, use_unk=True,
+                                  kl_factor=kl_factor, ctx_factor=ctx_factor,
+                                  state_factor=state_factor)
+        for (idx, _), (sample, score, alphas) in zip(req, outs):
+            score = numpy.array(score)
+            if normalize:
+                lengths = numpy.array([len(s2) for s2 in sample])
+                score = score / lengths
+            sidx = int(numpy.argmin(score))
+            rqueue.put((idx, sample[sidx],
+                        [int(numpy.argmax(a)) for a in alphas[sidx]]))
 
 
 def seqs2words(caps, pos, word_idict):
@@ -144,8 +155,9 @@ def generate_file(model_path, dictionary, source_file, saveto, k=5,
                                   state_factor, dev))
             p.start()
             procs.append(p)
-        for job in jobs:
-            queue.put(job)
+        sent_batch = max(1, 32 // max(k, 1))
+        for base in range(0, n_samples, sent_batch):
+            queue.put(jobs[base:base + sent_batch])
         for _ in range(n_process):
             queue.put(None)
         for i in range(n_samples):

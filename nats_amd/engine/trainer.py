@@ -58,11 +58,12 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
           batch_size=16, valid_batch_size=16, saveto="model.npz",
           validFreq=1000, saveFreq=1000, sampleFreq=100, datasets=[],
           valid_datasets=[], dictionary="", use_dropout=False, reload_=False,
-          verbose=False, device=None, seed=None):
+          verbose=False, device=None, seed=None, enc_depth=1):
     """Train the distraction model; returns final validation error.
 
-    Signature (and defaults) mirror nats.py:1230-1257; `device`/`seed` are
-    framework additions (device None = cuda if available else cpu).
+    Signature (and defaults) mirror nats.py:1230-1257; `device`/`seed`/
+    `enc_depth` are framework additions (device None = cuda if available
+    else cpu; enc_depth>1 stacks bi-GRU encoder layers).
     """
     logging.basicConfig(
         level=logging.DEBUG,
