@@ -63,17 +63,45 @@ def gru_scan(x_gates, x_cand, mask, U, Ux, h0=None):
     """
     if _use_hip(x_gates, U):
         from .gru import gru_scan_hip
-        return gru_scan_hip(x_gates, x_cand, mask, U, Ux, h0)
+        B = x_gates.shape[1]
+        if B <= MAX_KERNEL_BATCH:
+            return gru_scan_hip(x_gates, x_cand, mask, U, Ux, h0)
+        return torch.cat([gru_scan_hip(
+            x_gates[:, a:b], x_cand[:, a:b],
+            mask[:, a:b] if mask is not None else None, U, Ux,
+            h0[a:b] if h0 is not None else None)
+            for a, b in _batch_chunks(B)], dim=1)
     return eager.gru_scan(x_gates, x_cand, mask, U, Ux, h0)
+
+
+MAX_KERNEL_BATCH = 32  # MFMA M-tiling of the step kernels (2x 16-row tiles)
+
+
+def _batch_chunks(B):
+    return [(b0, min(b0 + MAX_KERNEL_BATCH, B))
+            for b0 in range(0, B, MAX_KERNEL_BATCH)]
 
 
 def gru_scan_bidir(xg0, xc0, mask0, U0, Ux0, xg1, xc1, mask1, U1, Ux1):
     """Both encoder directions (independent scans) in one fused launch
-    sequence on GPU; two eager scans on CPU. Returns (h_fwd, h_bwd)."""
+    sequence on GPU; two eager scans on CPU. Returns (h_fwd, h_bwd).
+
+    Batches beyond the kernels' 32-row MFMA tiling are processed in exact
+    per-chunk passes (batch rows are independent)."""
     if _use_hip(xg0, U0):
         from .gru import gru_scan_bidir_hip
-        return gru_scan_bidir_hip(xg0, xc0, mask0, U0, Ux0, xg1, xc1, mask1,
-                                  U1, Ux1)
+        B = xg0.shape[1]
+        if B <= MAX_KERNEL_BATCH:
+            return gru_scan_bidir_hip(xg0, xc0, mask0, U0, Ux0, xg1, xc1,
+                                      mask1, U1, Ux1)
+        outs = [gru_scan_bidir_hip(
+            xg0[:, a:b], xc0[:, a:b],
+            mask0[:, a:b] if mask0 is not None else None, U0, Ux0,
+            xg1[:, a:b], xc1[:, a:b],
+            mask1[:, a:b] if mask1 is not None else None, U1, Ux1)
+            for a, b in _batch_chunks(B)]
+        return (torch.cat([o[0] for o in outs], dim=1),
+                torch.cat([o[1] for o in outs], dim=1))
     return (eager.gru_scan(xg0, xc0, mask0, U0, Ux0),
             eager.gru_scan(xg1, xc1, mask1, U1, Ux1))
 
@@ -86,8 +114,18 @@ def cond_gru_scan(y_gates, y_cand, mask, init_state, ctx, ctx_mask, pctx, P):
     """
     if _use_hip(y_gates, ctx):
         from .cond_gru import cond_gru_scan_hip
-        return cond_gru_scan_hip(y_gates, y_cand, mask, init_state, ctx,
-                                 ctx_mask, pctx, P)
+        B = y_gates.shape[1]
+        if B <= MAX_KERNEL_BATCH:
+            return cond_gru_scan_hip(y_gates, y_cand, mask, init_state, ctx,
+                                     ctx_mask, pctx, P)
+        outs = [cond_gru_scan_hip(
+            y_gates[:, a:b], y_cand[:, a:b],
+            mask[:, a:b] if mask is not None else None,
+            init_state[a:b], ctx[:, a:b],
+            ctx_mask[:, a:b] if ctx_mask is not None else None,
+            pctx[:, a:b], P) for a, b in _batch_chunks(B)]
+        return tuple(torch.cat([o[i] for o in outs],
+                               dim=1 if i < 3 else 0) for i in range(5))
     return eager.cond_gru_scan(y_gates, y_cand, mask, init_state, ctx,
                                ctx_mask, pctx, P)
 
