@@ -57,33 +57,41 @@ __device__ __forceinline__ void store_cd_rowmajor(float* base, const f32x4& d,
   for (int i = 0; i < 4; ++i) base[(long)(rbase + i) * ld + col] = d[i];
 }
 
-// 2-deep software-pipelined MFMA K-loop: two iterations' fragment loads
-// stay in flight ahead of each MFMA so L2 latency (~300cy for the 16
-// scattered 16B lines per fragment) hides under the matrix ops.
-// (The plain loop measured ~13.9us/step on gfx950 — latency-bound.)
+// 4-deep software-pipelined MFMA K-loop: four iterations' fragment loads
+// (both operands) stay in flight ahead of each MFMA so the L2 latency of
+// the 16 scattered 16B lines per fragment amortises over 4 iterations.
+// (The plain loop measured ~13.9us/step on gfx950 — latency-bound; the
+// 2-deep variant was still ~30us on the K=3k decoder GEMMs.)
 #define NATS_MFMA_KLOOP(ACC, APTR, AROW, ALD, BPTR, BROW, BLD, KBEG, KEND)   \
   do {                                                                       \
-    int _k = (KBEG);                                                         \
     const int _ke = (KEND);                                                  \
-    if (_k + 32 >= _ke) {                                                    \
-      if (_k < _ke) {                                                        \
-        bf16x8 _a0 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
-        bf16x8 _b0 = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k);            \
-        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0);\
-      }                                                                      \
-    } else {                                                                 \
+    int _k = (KBEG);                                                         \
+    if (_k + 128 <= _ke) {                                                   \
       bf16x8 _a0 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);               \
       bf16x8 _b0 = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k);              \
       bf16x8 _a1 = frag_a_rowmajor((APTR), (AROW), (ALD), _k + 32);          \
       bf16x8 _b1 = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k + 32);         \
-      for (_k += 64; _k < _ke; _k += 32) {                                   \
-        bf16x8 _a2 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
-        bf16x8 _b2 = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k);            \
+      bf16x8 _a2 = frag_a_rowmajor((APTR), (AROW), (ALD), _k + 64);          \
+      bf16x8 _b2 = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k + 64);         \
+      bf16x8 _a3 = frag_a_rowmajor((APTR), (AROW), (ALD), _k + 96);          \
+      bf16x8 _b3 = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k + 96);         \
+      for (_k += 128; _k + 32 <= _ke; _k += 32) {                            \
+        bf16x8 _an = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
+        bf16x8 _bn = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k);            \
         ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0);\
         _a0 = _a1; _b0 = _b1; _a1 = _a2; _b1 = _b2;                          \
+        _a2 = _a3; _b2 = _b3; _a3 = _an; _b3 = _bn;                          \
       }                                                                      \
       ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0); \
       ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a1, _b1, ACC, 0, 0, 0); \
+      ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a2, _b2, ACC, 0, 0, 0); \
+      ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a3, _b3, ACC, 0, 0, 0); \
+    } else {                                                                 \
+      for (; _k < _ke; _k += 32) {                                           \
+        bf16x8 _a0 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
+        bf16x8 _b0 = frag_bt_rowmajor((BPTR), (BROW), (BLD), _k);            \
+        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0);\
+      }                                                                      \
     }                                                                        \
   } while (0)
 
