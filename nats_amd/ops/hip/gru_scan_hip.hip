@@ -1136,3 +1136,37 @@ std::vector<torch::Tensor> gru_scan_bwd(torch::Tensor dh_out,
   HIP_CHECK(hipGetLastError());
   return {dpre_all, dh_buf};
 }
+
+
+// ---- grid-barrier micro-benchmark: isolates the per-step sync cost of
+// the persistent scans (timing evidence in profiles/README.md). ----
+__global__ __launch_bounds__(384) void nats_barrier_bench_kernel(
+    unsigned* sync, unsigned nwg, int iters, float* out) {
+  for (int t = 0; t < iters; ++t) {
+    if (!nats_grid_barrier(sync, (unsigned)(t + 1), nwg)) return;
+  }
+  if (threadIdx.x == 0 && blockIdx.x == 0 && blockIdx.y == 0) out[0] = 1.f;
+}
+
+double barrier_bench(int nwg_x, int nwg_y, int iters) {
+  auto opts = torch::TensorOptions().dtype(torch::kInt32).device(torch::kCUDA);
+  auto sync = torch::zeros({18}, opts);
+  auto out = torch::zeros(
+      {1}, torch::TensorOptions().dtype(torch::kFloat32).device(torch::kCUDA));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  hipEvent_t e0, e1;
+  HIP_CHECK(hipEventCreate(&e0));
+  HIP_CHECK(hipEventCreate(&e1));
+  HIP_CHECK(hipEventRecord(e0, stream));
+  hipLaunchKernelGGL(nats_barrier_bench_kernel, dim3(nwg_x, nwg_y), dim3(384),
+                     0, stream, (unsigned*)sync.data_ptr<int>(),
+                     (unsigned)(nwg_x * nwg_y), iters, out.data_ptr<float>());
+  HIP_CHECK(hipEventRecord(e1, stream));
+  HIP_CHECK(hipEventSynchronize(e1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+  (void)hipEventDestroy(e0);
+  (void)hipEventDestroy(e1);
+  TORCH_CHECK(out.item<float>() == 1.f, "barrier bench gave up");
+  return (double)ms * 1000.0 / iters;  // us per barrier
+}

@@ -40,3 +40,4 @@ torch::Tensor mfma_gemm_bt(torch::Tensor A, torch::Tensor Bt) {
   HIP_CHECK(hipGetLastError());
   return C;
 }
+
