@@ -42,7 +42,8 @@ def _translate_one(model, seq, device, k, normalize, kl_factor, ctx_factor,
 
 
 def translate_worker(queue, rqueue, pid, model_path, options, k, normalize,
-                     kl_factor, ctx_factor, state_factor, device):
+                     kl_factor, ctx_factor, state_factor, device,
+                     maxlen=100):
     """Worker loop (translate_model, gen.py:15-58). Jobs arrive as CHUNKS
     of (idx, ids) pairs and are decoded jointly (batched beams)."""
     from .batched import gen_sample_batched
@@ -53,8 +54,9 @@ def translate_worker(queue, rqueue, pid, model_path, options, k, normalize,
             break
         xt = [torch.tensor(x, dtype=torch.int64, device=device).reshape(-1, 1)
               for _, x in req]
-        outs = gen_sample_batched(model, xt, k=k, maxlen=100, use_unk=True,
-                                  kl_factor=kl_factor, ctx_factor=ctx_factor,
+        outs = gen_sample_batched(model, xt, k=k, maxlen=maxlen,
+                                  use_unk=True, kl_factor=kl_factor,
+                                  ctx_factor=ctx_factor,
                                   state_factor=state_factor)
         for (idx, _), (sample, score, alphas) in zip(req, outs):
             score = numpy.array(score)
@@ -95,7 +97,7 @@ def map_line(line, word_dict, n_words, chr_level=False):
 def generate_file(model_path, dictionary, source_file, saveto, k=5,
                   normalize=False, n_process=5, chr_level=False,
                   kl_factor=0.0, ctx_factor=0.0, state_factor=0.0,
-                  devices=None, verbose=True):
+                  devices=None, verbose=True, maxlen=100):
     """gen.py main() equivalent."""
     options = load_options(model_path)
     word_dict = load_dictionary(dictionary)
@@ -128,7 +130,7 @@ def generate_file(model_path, dictionary, source_file, saveto, k=5,
             xt = [torch.tensor(x, dtype=torch.int64,
                                device=devices[0]).reshape(-1, 1)
                   for _, x in chunk]
-            outs = gen_sample_batched(model, xt, k=k, maxlen=100,
+            outs = gen_sample_batched(model, xt, k=k, maxlen=maxlen,
                                       use_unk=True, kl_factor=kl_factor,
                                       ctx_factor=ctx_factor,
                                       state_factor=state_factor)
@@ -152,7 +154,7 @@ def generate_file(model_path, dictionary, source_file, saveto, k=5,
             p = ctx.Process(target=translate_worker,
                             args=(queue, rqueue, midx, model_path, options, k,
                                   normalize, kl_factor, ctx_factor,
-                                  state_factor, dev))
+                                  state_factor, dev, maxlen))
             p.start()
             procs.append(p)
         sent_batch = max(1, 32 // max(k, 1))

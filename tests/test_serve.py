@@ -1,0 +1,130 @@
+"""Serving layer: FastAPI app + micro-batching worker (nats_amd/serve).
+
+The reference has no serving path; these tests pin our addition's
+behaviour: decode parity with the offline driver, UNK replacement by
+attention argmax (replace_unk.py semantics inline), micro-batch
+coalescing, and the HTTP surface.
+"""
+
+import os
+import threading
+
+import pytest
+from fastapi.testclient import TestClient
+
+from nats_amd.data.dictionary import load_dictionary
+from nats_amd.engine.checkpoint import save_checkpoint
+from nats_amd.models.distraction import NatsModel
+from nats_amd.serve import SummarizerService, create_app
+
+
+@pytest.fixture(scope="module")
+def tiny_server(tmp_path_factory):
+    from nats_amd.data.synthetic import make_toy_corpus
+    from nats_amd.models.distraction import default_options
+    d = tmp_path_factory.mktemp("serve_data")
+    make_toy_corpus(str(d))
+    opts = default_options(dim_word=12, dim=16, dim_att=8, n_words=43,
+                           maxlen=50)
+    model = NatsModel(opts, seed=3)
+    saveto = os.path.join(str(d), "model.npz")
+    save_checkpoint(saveto, model.get_params(), [], options=opts)
+    svc = SummarizerService(saveto, os.path.join(str(d), "toy_train_input.txt.pkl"),
+                            device="cpu", k=4, maxlen=12, max_wait_ms=20.0)
+    yield svc, str(d)
+    svc.close()
+
+
+def test_healthz_and_single_request(tiny_server):
+    svc, d = tiny_server
+    app = create_app(svc)
+    client = TestClient(app)
+    r = client.get("/healthz")
+    assert r.status_code == 200
+    assert r.json()["status"] == "ok"
+    assert r.json()["beam"] == 4
+
+    r = client.post("/summarize", json={"text": "a b c d e f"})
+    assert r.status_code == 200
+    body = r.json()
+    assert set(body) == {"summary", "tokens", "score", "alignment"}
+    assert isinstance(body["summary"], str)
+    assert len(body["tokens"]) == len(body["alignment"])
+    assert all(t != 0 for t in body["tokens"])
+
+
+def test_batch_request_and_stats(tiny_server):
+    svc, d = tiny_server
+    app = create_app(svc)
+    client = TestClient(app)
+    texts = ["a b c d", "b c d e f", "c d e"]
+    r = client.post("/summarize", json={"texts": texts})
+    assert r.status_code == 200
+    results = r.json()["results"]
+    assert len(results) == 3
+    r = client.get("/stats")
+    assert r.status_code == 200
+    s = r.json()
+    assert s["requests"] >= 3 and s["batches"] >= 1
+
+
+def test_validation_error(tiny_server):
+    svc, _ = tiny_server
+    client = TestClient(create_app(svc))
+    assert client.post("/summarize", json={}).status_code == 422
+
+
+def test_unk_replacement_uses_source_word(tiny_server):
+    """OOV source tokens map to UNK(1); if the decode emits UNK it must be
+    replaced by the attention-argmax source word (replace_unk.py:14-37)."""
+    svc, d = tiny_server
+    out = svc.summarize("zzz_oov_word a b c")
+    # the summary must never contain the literal UNK token
+    assert "UNK" not in out["summary"].split()
+
+
+def test_decode_matches_offline_driver(tiny_server):
+    """Serving path == gen.py driver path on the same input (same engine,
+    same beams; driver emits 'word [pos]', service emits plain words)."""
+    from nats_amd.decode.driver import generate_file
+    svc, d = tiny_server
+    src = os.path.join(d, "serve_src.txt")
+    lines = ["a b c d e", "b c d"]
+    with open(src, "w") as f:
+        f.write("\n".join(lines) + "\n")
+    out = os.path.join(d, "serve_out.txt")
+    generate_file(os.path.join(d, "model.npz"), os.path.join(d, "toy_train_input.txt.pkl"),
+                  src, out, k=svc.k, normalize=True, n_process=1,
+                  verbose=False, maxlen=svc.maxlen)
+    with open(out) as f:
+        offline = [ln.strip() for ln in f.read().splitlines()]
+    served = svc.summarize_many(lines)
+    word_dict = load_dictionary(os.path.join(d, "toy_train_input.txt.pkl"))
+    for line, off, sv in zip(lines, offline, served):
+        # strip the driver's interleaved [pos] markers, apply its UNK rule
+        toks = off.split()
+        words, positions = toks[0::2], [int(p[1:-1]) for p in toks[1::2]]
+        srcw = line.split()
+        off_words = [srcw[p] if w == "UNK" and p < len(srcw) else w
+                     for w, p in zip(words, positions)]
+        assert sv["summary"].split() == off_words
+
+
+def test_microbatch_coalescing(tiny_server):
+    """Concurrent requests within the wait window share one decode batch."""
+    svc, _ = tiny_server
+    before = svc.stats()["batches"]
+    outs = [None] * 4
+
+    def go(i):
+        outs[i] = svc.summarize("a b c d e"[: 2 * i + 3])
+
+    threads = [threading.Thread(target=go, args=(i,)) for i in range(4)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert all(o is not None for o in outs)
+    # 4 requests, beam 4 -> max_batch = 8 rows... coalesced into <=2 batches
+    after = svc.stats()["batches"]
+    assert after - before <= 3
