@@ -128,3 +128,33 @@ def test_microbatch_coalescing(tiny_server):
     # 4 requests, beam 4 -> max_batch = 8 rows... coalesced into <=2 batches
     after = svc.stats()["batches"]
     assert after - before <= 3
+
+
+@pytest.mark.gpu
+def test_serve_on_gpu(tmp_path):
+    """Service end-to-end on the HIP kernel path (real deployment shape)."""
+    import torch
+    from nats_amd.data.synthetic import make_toy_corpus
+    from nats_amd.models.distraction import default_options
+    d = str(tmp_path)
+    make_toy_corpus(d)
+    opts = default_options(dim_word=32, dim=64, dim_att=32, n_words=43,
+                           maxlen=50)
+    model = NatsModel(opts, seed=5)
+    saveto = os.path.join(d, "model.npz")
+    save_checkpoint(saveto, model.get_params(), [], options=opts)
+    svc = SummarizerService(saveto, os.path.join(d, "toy_train_input.txt.pkl"),
+                            device="cuda:0", k=8, maxlen=15,
+                            kl_factor=0.3, ctx_factor=0.3, state_factor=0.3)
+    try:
+        client = TestClient(create_app(svc))
+        r = client.post("/summarize",
+                        json={"texts": ["a b c d e", "b c d", "e f g h"]})
+        assert r.status_code == 200
+        results = r.json()["results"]
+        assert len(results) == 3
+        for res in results:
+            assert "UNK" not in res["summary"].split()
+        assert torch.cuda.is_available()
+    finally:
+        svc.close()
