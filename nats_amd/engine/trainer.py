@@ -124,17 +124,31 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
     valid_err = None
     for eidx in range(max_epochs):
         n_samples = 0
-        for bidx, (xs, ys) in enumerate(train_it):
-            # data-parallel sharding: each rank takes every world-th batch
-            if world > 1 and (bidx % world) != rank:
-                continue
+        # Data-parallel sharding: batches are consumed in complete groups of
+        # `world` (rank r takes the r-th batch of each group) and a ragged
+        # final group is DROPPED — otherwise ranks would run different step
+        # counts per epoch and the gradient/validation collectives would
+        # desynchronise across ranks.
+        def _sharded(it):
+            group = []
+            for batch in it:
+                group.append(batch)
+                if len(group) == world:
+                    yield group[rank]
+                    group = []
+
+        for bidx, (xs, ys) in enumerate(
+                _sharded(train_it) if world > 1 else train_it):
             n_samples += len(xs)
             uidx += 1
 
             x, x_mask, y, y_mask = prepare_data(xs, ys, maxlen=maxlen,
                                                 n_words=n_words)
-            if x is None:
-                print("Minibatch with zero sample under length ", maxlen)
+            # the skip must be agreed across ranks, else collective counts
+            # diverge (one rank skipping while others all-reduce -> hang)
+            if not dp.all_agree(x is not None):
+                if x is None:
+                    print("Minibatch with zero sample under length ", maxlen)
                 uidx -= 1
                 continue
 

@@ -140,6 +140,18 @@ class DataParallelGrads:
         self._pending = []
         self._ready_count = [0] * len(self.buckets)
 
+    def all_agree(self, ok):
+        """True iff EVERY rank passes ok=True (MIN all-reduce). Used to
+        make data-dependent step skips (empty minibatch after maxlen
+        truncation) collective so per-rank collective counts stay equal."""
+        if not self.enabled:
+            return bool(ok)
+        t = torch.tensor([1.0 if ok else 0.0], dtype=torch.float32,
+                         device="cuda" if dist.get_backend(self.group) == "nccl"
+                         else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MIN, group=self.group)
+        return bool(t.item() >= 0.5)
+
     def all_reduce_scalar(self, value, average=True):
         """All-reduce a python float (validation loss, token counts)."""
         if not self.enabled:

@@ -65,3 +65,53 @@ def test_ddp_grad_allreduce_gloo(tmp_path):
     port = 29531
     mp.spawn(_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
     assert os.path.exists(tmp_path / "ok")
+
+
+def _ragged_worker(rank, world, port, tmpdir):
+    """Full train() with a batch count per epoch NOT divisible by world and
+    one maxlen-unsatisfiable batch: ranks must stay collective-aligned
+    (ragged tail dropped, empty-batch skip agreed via all_agree)."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    try:
+        from nats_amd.engine.trainer import train
+
+        src = os.path.join(tmpdir, "src.txt")
+        tgt = os.path.join(tmpdir, "tgt.txt")
+        words = ["w%d" % i for i in range(2, 28)]
+        rng = numpy.random.RandomState(7)
+        with open(src, "w") as fs, open(tgt, "w") as ft:
+            for i in range(14):  # 7 batches of 2 -> odd, not % 2
+                # one over-length source line makes batch 3 empty at maxlen=8
+                n = 40 if i in (6, 7) else 5
+                fs.write(" ".join(rng.choice(words, size=n)) + "\n")
+                ft.write(" ".join(rng.choice(words, size=3)) + "\n")
+        from nats_amd.data.dictionary import build_dictionary
+        dpath = os.path.join(tmpdir, "d_r%d.pkl" % rank)
+        build_dictionary(src, dpath)  # deterministic -> identical per rank
+        err = train(dim_word=8, dim=10, dim_att=6, n_words=30, maxlen=8,
+                    batch_size=2, valid_batch_size=2, max_epochs=3,
+                    dispFreq=100, validFreq=100, saveFreq=100, sampleFreq=100,
+                    datasets=[src, tgt], valid_datasets=[src, tgt],
+                    dictionary=dpath,
+                    saveto=os.path.join(tmpdir, "m_r%d.npz" % rank),
+                    device="cpu", seed=1)
+        assert numpy.isfinite(err)
+        if rank == 0:
+            open(os.path.join(tmpdir, "ok_ragged"), "w").write("ok")
+    finally:
+        for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE"):
+            os.environ.pop(k, None)
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ddp_ragged_batches_no_deadlock(tmp_path):
+    port = 29541
+    mp.spawn(_ragged_worker, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
+    assert os.path.exists(tmp_path / "ok_ragged")
