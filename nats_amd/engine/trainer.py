@@ -58,12 +58,24 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
           batch_size=16, valid_batch_size=16, saveto="model.npz",
           validFreq=1000, saveFreq=1000, sampleFreq=100, datasets=[],
           valid_datasets=[], dictionary="", use_dropout=False, reload_=False,
-          verbose=False, device=None, seed=None, enc_depth=1):
+          verbose=False, device=None, seed=None, enc_depth=1,
+          profile=False, resume_optimizer=False):
     """Train the distraction model; returns final validation error.
 
     Signature (and defaults) mirror nats.py:1230-1257; `device`/`seed`/
     `enc_depth` are framework additions (device None = cuda if available
     else cpu; enc_depth>1 stacks bi-GRU encoder layers).
+
+    ``use_dropout`` is accepted but has no effect — exact parity: the
+    reference defines dropout_layer (nats.py:50) and the use_noise toggle
+    but never applies either in build_model (nats.py:683-772).
+
+    ``profile=True`` is the rebuild of the reference's module-level
+    ``profile`` flag (nats.py:26): per-section (forward / backward /
+    allreduce / optimizer) cuda-synchronised wall times, reported at every
+    dispFreq. ``resume_optimizer=True`` also saves/loads the optimizer
+    sidecar ``<saveto>.opt.npz`` with each checkpoint (default off: the
+    reference restarts adadelta accumulators from zero on resume).
     """
     logging.basicConfig(
         level=logging.DEBUG,
@@ -111,6 +123,15 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
     opt = build_optimizer(model_options["optimizer"],
                           list(model.P.items()),
                           lrate=lrate, clip_c=clip_c)
+    if resume_optimizer and reload_:
+        from .checkpoint import load_optimizer_state
+        if load_optimizer_state(saveto, opt):
+            print("Reload optimizer state")
+
+    step_timer = None
+    if profile:
+        from ..utils import StepTimer
+        step_timer = StepTimer()
 
     history_errs = []
     if reload_ and os.path.exists(saveto):
@@ -156,14 +177,27 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
             x, x_mask, y, y_mask = _to_device((x, x_mask, y, y_mask), device)
 
             opt.zero_grad()
+            if step_timer is not None:
+                step_timer.start("forward")
             cost_vec = model(x, x_mask, y, y_mask)
             cost = cost_vec.mean()
             if decay_c > 0.0:
                 weight_decay = sum((p ** 2).sum() for p in model.parameters())
                 cost = cost + decay_c * weight_decay
+            if step_timer is not None:
+                step_timer.stop("forward")
+                step_timer.start("backward")
             cost.backward()
+            if step_timer is not None:
+                step_timer.stop("backward")
+                step_timer.start("allreduce")
             dp.finish()
+            if step_timer is not None:
+                step_timer.stop("allreduce")
+                step_timer.start("optimizer")
             norm_g = opt.step()
+            if step_timer is not None:
+                step_timer.stop("optimizer")
             cost_val = float(cost.detach())
             ud = time.time() - ud_start
 
@@ -177,12 +211,17 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
                     eidx, uidx, cost_val, ud))
                 if verbose and clip_c > 0:
                     logger.debug("Grad {0}".format(float(norm_g)))
+                if step_timer is not None:
+                    logger.debug("Step breakdown:\n%s", step_timer.report())
 
             if rank == 0 and numpy.mod(uidx, saveFreq) == 0:
                 print("Saving...", end=" ")
                 params = best_p if best_p is not None else model.get_params()
                 save_checkpoint(saveto, params, history_errs,
                                 options=model_options)
+                if resume_optimizer:
+                    from .checkpoint import save_optimizer_state
+                    save_optimizer_state(saveto, opt)
                 print("Done")
 
             if rank == 0 and numpy.mod(uidx, sampleFreq) == 0:

@@ -191,10 +191,12 @@ class SummarizerService:
             for w, p in zip(ids, align):
                 if w == 0:  # eos
                     break
-                if w == 1 and p < len(src_words):  # UNK -> aligned src word
+                # UNK (and any id outside the dictionary, e.g. a checkpoint
+                # trained with n_words > dict size) -> aligned source word
+                if (w == 1 or w not in self.word_idict) and p < len(src_words):
                     words.append(src_words[p])
                 else:
-                    words.append(self.word_idict[w])
+                    words.append(self.word_idict.get(w, "UNK"))
             results.append({
                 "summary": " ".join(words),
                 "tokens": [int(w) for w in ids if w != 0],

@@ -39,6 +39,10 @@ def main():
     vocab_words = ["w%d" % i for i in range(args.vocab - 2)]
     corpus = os.path.join(d, "corpus.txt")
     with open(corpus, "w") as f:
+        # every vocab word appears (freq-desc ids then cover the model's
+        # full n_words range), plus random lines for realistic freq skew
+        for i in range(0, len(vocab_words), 20):
+            f.write(" ".join(vocab_words[i:i + 20]) + "\n")
         for _ in range(200):
             f.write(" ".join(rng.choice(vocab_words, size=10)) + "\n")
     from nats_amd.data.dictionary import build_dictionary
