@@ -30,16 +30,22 @@ def gen_sample_batched(model, xs, k=1, maxlen=30, use_unk=False,
     device = xs[0].device
     any_lambda = (kl_factor > 0.0 or ctx_factor > 0.0 or state_factor > 0.0)
 
-    # encode each sentence (its own length), then pad to the longest
-    inits, ctxs0, lens = [], [], []
-    for x in xs:
-        init, c = model.f_init(x)
-        inits.append(init)
-        ctxs0.append(c)
-        lens.append(c.shape[0])
+    # ONE batched masked encode for all sentences (identical per-column to
+    # each sentence's own unpadded encode — see f_init): pads to the
+    # longest source so the encoder scans run S columns per launch instead
+    # of S separate T-step launches.
+    lens = [int(x.shape[0]) for x in xs]
     Ts = max(lens)
-    C = ctxs0[0].shape[2]
-    H = inits[0].shape[1]
+    x_pad = torch.zeros(Ts, S, dtype=torch.int64, device=device)
+    x_mask = torch.zeros(Ts, S, device=device)
+    for i, x in enumerate(xs):
+        x_pad[:lens[i], i] = x[:, 0]
+        x_mask[:lens[i], i] = 1.0
+    init_all, ctx_all = model.f_init(x_pad, x_mask)
+    C = ctx_all.shape[2]
+    H = init_all.shape[1]
+    inits = [init_all[i:i + 1] for i in range(S)]
+    ctxs0 = [ctx_all[:lens[i], i:i + 1] for i in range(S)]
 
     # per-sentence beam state (numpy bookkeeping identical to gen_sample)
     st = [dict(live=1, dead=0, samples=[[]],
