@@ -7,7 +7,17 @@ decode step as one (S*k)-row batch through the fused decoder kernels
 are identical to decode.beam.gen_sample — sources are padded to the
 longest and masked via ctx_mask, so attention, the distraction terms and
 the histories are unaffected by padding.
+
+Execution layout (MI355X): all recurrent state (h2, acc_ctx, acc_alpha,
+rerank histories) stays device-resident across steps and hypothesis
+reshuffles are device index_selects; the only per-step host transfers
+are the top-k candidate ids/costs and the selected attention rows for
+the alignment output (two small synchronisations per step — the naive
+numpy bookkeeping round-trip was ~25% of decode time in the kernel
+trace, profiles/decode_kernel_stats.csv).
 """
+
+import math
 
 import numpy
 import torch
@@ -37,54 +47,44 @@ def gen_sample_batched(model, xs, k=1, maxlen=30, use_unk=False,
     lens = [int(x.shape[0]) for x in xs]
     Ts = max(lens)
     x_pad = torch.zeros(Ts, S, dtype=torch.int64, device=device)
-    x_mask = torch.zeros(Ts, S, device=device)
+    src_mask = torch.zeros(Ts, S, device=device)
     for i, x in enumerate(xs):
         x_pad[:lens[i], i] = x[:, 0]
-        x_mask[:lens[i], i] = 1.0
-    init_all, ctx_all = model.f_init(x_pad, x_mask)
+        src_mask[:lens[i], i] = 1.0
+    init_all, ctx_all = model.f_init(x_pad, src_mask)
     C = ctx_all.shape[2]
-    H = init_all.shape[1]
-    inits = [init_all[i:i + 1] for i in range(S)]
-    ctxs0 = [ctx_all[:lens[i], i:i + 1] for i in range(S)]
 
-    # per-sentence beam state (numpy bookkeeping identical to gen_sample)
+    # per-sentence host bookkeeping (sample token lists + scores; identical
+    # math to gen_sample) — everything tensor-valued lives on device
     st = [dict(live=1, dead=0, samples=[[]],
                scores=numpy.zeros(1, dtype="float32"),
-               alphas=[[]], ctx_hist=[[]], st_hist=[[]],
-               out_samples=[], out_scores=[], out_alphas=[])
+               alphas=[[]], out_samples=[], out_scores=[], out_alphas=[])
           for _ in range(S)]
-    # device state per sentence
-    dstate = [dict(state=inits[i], w=torch.full((1,), -1, dtype=torch.int64,
-                                                device=device),
+    dstate = [dict(state=init_all[i:i + 1],
+                   w=torch.full((1,), -1, dtype=torch.int64, device=device),
                    acc_c=torch.zeros(1, C, device=device),
                    acc_a=torch.zeros(1, Ts, device=device),
                    ha=None, hc=None, hs=None)
               for i in range(S)]
 
-    ctx_pad = torch.zeros(Ts, S, C, device=device, dtype=ctxs0[0].dtype)
-    ctx_mask_s = torch.zeros(Ts, S, device=device)
-    for i, c in enumerate(ctxs0):
-        ctx_pad[:lens[i], i] = c[:, 0]
-        ctx_mask_s[:lens[i], i] = 1.0
+    ctx_pad = torch.zeros(Ts, S, C, device=device, dtype=ctx_all.dtype)
+    for i in range(S):
+        ctx_pad[:lens[i], i] = ctx_all[:lens[i], i]
+    ctx_mask_s = src_mask
     pctx_pad = model.project_ctx(ctx_pad)
 
+    NEG_UNK = math.log(1e-20)
+
     for ii in range(maxlen):
-        rows = []   # (sentence, local row)
-        for i in range(S):
-            for r in range(st[i]["live"]):
-                rows.append((i, r))
-        if not rows:
+        alive = [i for i in range(S) if st[i]["live"] > 0]
+        if not alive:
             break
-        B = len(rows)
-        sent_idx = torch.tensor([i for i, _ in rows], device=device)
-        y = torch.cat([dstate[i]["w"] for i in range(S)
-                       if st[i]["live"] > 0])
-        state = torch.cat([dstate[i]["state"] for i in range(S)
-                           if st[i]["live"] > 0])
-        acc_c = torch.cat([dstate[i]["acc_c"] for i in range(S)
-                           if st[i]["live"] > 0])
-        acc_a = torch.cat([dstate[i]["acc_a"] for i in range(S)
-                           if st[i]["live"] > 0])
+        sent_idx = torch.tensor(
+            sum(([i] * st[i]["live"] for i in alive), []), device=device)
+        y = torch.cat([dstate[i]["w"] for i in alive])
+        state = torch.cat([dstate[i]["state"] for i in alive])
+        acc_c = torch.cat([dstate[i]["acc_c"] for i in alive])
+        acc_a = torch.cat([dstate[i]["acc_a"] for i in alive])
         ctx_b = ctx_pad[:, sent_idx]
         cmask_b = ctx_mask_s[:, sent_idx]
         pctx_b = pctx_pad[:, sent_idx]
@@ -92,75 +92,83 @@ def gen_sample_batched(model, xs, k=1, maxlen=30, use_unk=False,
         probs, _, h2, alpha, ctx_t, acc_c, acc_a = model.f_next(
             y, ctx_b, cmask_b, pctx_b, state, acc_c, acc_a,
             sample_draw=False)
+        V = probs.shape[1]
 
-        probs_np = probs.float().cpu().numpy()
+        # ---- phase 1: per-sentence top-k ON DEVICE, one host transfer ----
+        logp = probs.float().log()
         if not use_unk:
-            probs_np[:, 1] = 1e-20
-        alpha_np = alpha.float().cpu().numpy()
-        ctxt_np = ctx_t.float().cpu().numpy()
-        h2_np = h2.float().cpu().numpy()
-        accc_np = acc_c.float().cpu().numpy()
-        acca_np = acc_a.float().cpu().numpy()
-
-        # per-sentence selection (same math as gen_sample)
+            logp[:, 1] = NEG_UNK
+        sel_parts = []   # per alive sentence: (ranks (want,), costs (want,))
         row0 = 0
-        for i in range(S):
+        for i in alive:
             live = st[i]["live"]
-            if live == 0:
-                continue
             sl = slice(row0, row0 + live)
             row0 += live
-            p_i = probs_np[sl]
-            cand = st[i]["scores"][:, None] - numpy.log(p_i)
+            cand = torch.as_tensor(st[i]["scores"],
+                                   device=device)[:, None] - logp[sl]
             cand_flat = cand.flatten()
             want = k - st[i]["dead"]
-            ranks = cand_flat.argsort()[:want]
+            sel_flat = cand_flat
             if ii > 0 and any_lambda and dstate[i]["ha"] is not None:
                 pen = distraction_penalties_gpu(
                     dstate[i]["ha"], dstate[i]["hc"], dstate[i]["hs"],
                     alpha[sl].float(), ctx_t[sl].float(), h2[sl].float(),
-                    kl_factor, ctx_factor, state_factor).cpu().numpy()
-                ranks = (cand + pen[:, None]).flatten().argsort()[:want]
-            V = p_i.shape[1]
-            tis = (ranks // V).astype(int)
-            wis = (ranks % V).astype(int)
-            costs = cand_flat[ranks]
+                    kl_factor, ctx_factor, state_factor)
+                sel_flat = (cand + pen[:, None]).flatten()
+            ranks = sel_flat.topk(want, largest=False).indices
+            sel_parts.append((ranks, cand_flat[ranks]))  # UN-reranked costs
+        flat_host = torch.cat([torch.cat([r.double(), c.double()])
+                               for r, c in sel_parts]).cpu().numpy()
 
-            new_samples, new_scores, new_states = [], [], []
-            new_alphas, new_ctxh, new_sth = [], [], []
-            new_accc, new_acca, new_words = [], [], []
-            for rank_i, (ti, wi) in enumerate(zip(tis, wis)):
+        # ---- phase 2: bookkeeping + device-side state reshuffle ----
+        alpha_gather = []   # device rows to fetch for alignment histories
+        pend = []           # (i, sl, tis, wis, costs, sel_dev)
+        row0 = 0
+        off = 0
+        for i in alive:
+            live = st[i]["live"]
+            sl = slice(row0, row0 + live)
+            row0 += live
+            want = k - st[i]["dead"]
+            ranks = flat_host[off:off + want].astype(numpy.int64)
+            costs = flat_host[off + want:off + 2 * want].astype("float32")
+            off += 2 * want
+            tis = ranks // V
+            wis = ranks % V
+            sel_dev = torch.as_tensor(tis, device=device) + sl.start
+            alpha_gather.append(alpha.float()[sel_dev])
+            pend.append((i, sl, tis, wis, costs, sel_dev))
+        alpha_sel = torch.cat(alpha_gather).cpu().numpy()
+
+        arow = 0
+        for i, sl, tis, wis, costs, sel_dev in pend:
+            want = len(tis)
+            a_np = alpha_sel[arow:arow + want]
+            arow += want
+
+            new_samples, new_scores, new_alphas, new_words = [], [], [], []
+            for r, (ti, wi) in enumerate(zip(tis, wis)):
                 new_samples.append(st[i]["samples"][ti] + [int(wi)])
-                new_scores.append(float(costs[rank_i]))
-                new_states.append(h2_np[sl][ti])
+                new_scores.append(float(costs[r]))
                 new_alphas.append(st[i]["alphas"][ti] +
-                                  [alpha_np[sl][ti, :lens[i]].copy()])
-                new_ctxh.append(st[i]["ctx_hist"][ti] +
-                                [ctxt_np[sl][ti].copy()])
-                new_sth.append(st[i]["st_hist"][ti] + [h2_np[sl][ti].copy()])
-                new_accc.append(accc_np[sl][ti])
-                new_acca.append(acca_np[sl][ti])
+                                  [a_np[r, :lens[i]].copy()])
                 new_words.append(int(wi))
 
             # device histories for the rerank (pre-filter selection order)
             if any_lambda:
-                sel = torch.tensor(tis, dtype=torch.int64, device=device)
-                base = sl.start
-                cur_a = alpha[base:base + live][sel].unsqueeze(0).float()
-                cur_c = ctx_t[base:base + live][sel].unsqueeze(0).float()
-                cur_s = h2[base:base + live][sel].unsqueeze(0).float()
+                cur_a = alpha.float()[sel_dev].unsqueeze(0)
+                cur_c = ctx_t.float()[sel_dev].unsqueeze(0)
+                cur_s = h2.float()[sel_dev].unsqueeze(0)
                 if dstate[i]["ha"] is None:
                     ha, hc, hs = cur_a, cur_c, cur_s
                 else:
-                    ha = torch.cat([dstate[i]["ha"][:, sel], cur_a], 0)
-                    hc = torch.cat([dstate[i]["hc"][:, sel], cur_c], 0)
-                    hs = torch.cat([dstate[i]["hs"][:, sel], cur_s], 0)
+                    ti_dev = sel_dev - sl.start
+                    ha = torch.cat([dstate[i]["ha"][:, ti_dev], cur_a], 0)
+                    hc = torch.cat([dstate[i]["hc"][:, ti_dev], cur_c], 0)
+                    hs = torch.cat([dstate[i]["hs"][:, ti_dev], cur_s], 0)
 
-            keep = []
-            samples, scores, states_l = [], [], []
-            alphas_l, ctxh_l, sth_l, accc_l, acca_l, words_l = \
-                [], [], [], [], [], []
-            for idx2 in range(len(new_samples)):
+            keep, samples, scores, alphas_l, words_l = [], [], [], [], []
+            for idx2 in range(want):
                 if new_samples[idx2][-1] == 0:
                     st[i]["out_samples"].append(new_samples[idx2])
                     st[i]["out_scores"].append(new_scores[idx2])
@@ -170,31 +178,23 @@ def gen_sample_batched(model, xs, k=1, maxlen=30, use_unk=False,
                     keep.append(idx2)
                     samples.append(new_samples[idx2])
                     scores.append(new_scores[idx2])
-                    states_l.append(new_states[idx2])
                     alphas_l.append(new_alphas[idx2])
-                    ctxh_l.append(new_ctxh[idx2])
-                    sth_l.append(new_sth[idx2])
-                    accc_l.append(new_accc[idx2])
-                    acca_l.append(new_acca[idx2])
                     words_l.append(new_words[idx2])
 
-            st[i].update(samples=samples, alphas=alphas_l, ctx_hist=ctxh_l,
-                         st_hist=sth_l,
+            st[i].update(samples=samples, alphas=alphas_l,
                          scores=numpy.array(scores, dtype="float32"))
             st[i]["live"] = len(samples)
             if st[i]["dead"] >= k:
                 st[i]["live"] = 0
             if st[i]["live"] > 0:
+                kt = torch.tensor(keep, dtype=torch.int64, device=device)
+                sel_keep = sel_dev[kt]
                 dstate[i]["w"] = torch.tensor(words_l, dtype=torch.int64,
                                               device=device)
-                dstate[i]["state"] = torch.from_numpy(
-                    numpy.array(states_l, dtype="float32")).to(device)
-                dstate[i]["acc_c"] = torch.from_numpy(
-                    numpy.array(accc_l, dtype="float32")).to(device)
-                dstate[i]["acc_a"] = torch.from_numpy(
-                    numpy.array(acca_l, dtype="float32")).to(device)
+                dstate[i]["state"] = h2[sel_keep].float()
+                dstate[i]["acc_c"] = acc_c[sel_keep].float()
+                dstate[i]["acc_a"] = acc_a[sel_keep].float()
                 if any_lambda:
-                    kt = torch.tensor(keep, dtype=torch.int64, device=device)
                     dstate[i]["ha"] = ha[:, kt]
                     dstate[i]["hc"] = hc[:, kt]
                     dstate[i]["hs"] = hs[:, kt]
