@@ -271,41 +271,103 @@ __global__ __launch_bounds__(384) void nats_gru_step_bwd_fused_bidir(
 // placement-independent, bounded spin with a give-up flag the host
 // checks — a barrier bug aborts instead of hanging the box).
 
-// XCD-hierarchical grid barrier (microarch 'barrier-xcd': ~4us at 256 WGs
-// vs ~7.4us for a single flat counter). Blocks are bucketed by linear id
-// mod 8 — a SPEED heuristic matching the observed b%8 dispatcher placement
-// (correctness never depends on it: all traffic is agent-scope).
-// Sync layout (zeroed per launch): [0..7] per-bucket arrive counters,
-// [8] top counter, [9..16] per-bucket generation words, [17] give-up.
+// XCD-hierarchical grid barrier (microarch 'barrier-xcd' scheme). Blocks
+// are bucketed by their REAL XCC id (s_getreg XCC_ID, gfx942/950 —
+// amd_device_functions.h:754-793), established once by a census at kernel
+// start (nats_barrier_init). Per barrier, only the LAST ARRIVER of each
+// XCD executes the agent-release fence (ONE buffer_wbl2 L2 writeback per
+// XCD instead of one per block — the per-block variant measured 7.8us at
+// 126 WGs / 12.1us at 256; see profiles/barrier_bench.json); it then
+// arrives at the top counter, acquire-fences, and publishes the XCD's
+// generation word. Non-leaders poll their XCD's generation word and
+// acquire-fence (buffer_inv) before returning. Correctness relies only on
+// bucket == physical XCD (true by construction from XCC_ID): the leader's
+// wbl2 flushes exactly the L2 holding its co-located blocks' drained
+// (vmcnt(0)) stores.
+//
+// Sync layout (zeroed per launch, NATS_SYNC_WORDS ints):
+//   [0..7]  per-XCD arrive counters (monotonic: nper*epoch)
+//   [8]     top counter (monotonic: nxcd*epoch)
+//   [9..16] per-XCD generation words
+//   [17]    give-up flag (checked by the host via NaN poisoning)
+//   [18..25] census: resident blocks per XCD
+//   [26]    census arrive counter
+
+struct NatsBarrierCtx {  // valid on thread 0 only
+  unsigned bucket;  // this block's physical XCD (0..7)
+  unsigned nper;    // blocks resident on this XCD
+  unsigned ntop;    // number of XCDs with >=1 block
+};
+
+__device__ __forceinline__ unsigned nats_xcc_id() {
+  // s_getreg_b32 XCC_ID[3:0] (hwreg 20), gfx942/950
+  return __builtin_amdgcn_s_getreg((3u << 11) | (0u << 6) | 20u) & 7u;
+}
+
+__device__ __forceinline__ bool nats_barrier_init(unsigned* sync,
+                                                  unsigned nwg,
+                                                  NatsBarrierCtx& ctx) {
+  __shared__ unsigned sh[4];
+  if (threadIdx.x == 0) {
+    const unsigned xcc = nats_xcc_id();
+    __hip_atomic_fetch_add(sync + 18 + xcc, 1u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+    __hip_atomic_fetch_add(sync + 26, 1u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+    unsigned ok = 1u, spins = 0u;
+    while (__hip_atomic_load(sync + 26, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT) < nwg) {
+      if (__hip_atomic_load(sync + 17, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_AGENT) != 0u ||
+          ++spins > 200000000u) {
+        __hip_atomic_store(sync + 17, 1u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        ok = 0u;
+        break;
+      }
+      __builtin_amdgcn_s_sleep(2);
+    }
+    unsigned ntop = 0u;
+    for (int i = 0; i < 8; ++i)
+      ntop += (__hip_atomic_load(sync + 18 + i, __ATOMIC_RELAXED,
+                                 __HIP_MEMORY_SCOPE_AGENT) != 0u);
+    sh[0] = xcc;
+    sh[1] = __hip_atomic_load(sync + 18 + xcc, __ATOMIC_RELAXED,
+                              __HIP_MEMORY_SCOPE_AGENT);
+    sh[2] = ntop;
+    sh[3] = ok;
+  }
+  __syncthreads();
+  ctx.bucket = sh[0];
+  ctx.nper = sh[1];
+  ctx.ntop = sh[2];
+  return sh[3] != 0u;
+}
+
 __device__ __forceinline__ bool nats_grid_barrier(unsigned* sync,
                                                   unsigned epoch,
-                                                  unsigned nwg) {
+                                                  const NatsBarrierCtx& ctx) {
   __shared__ unsigned ok_sh;
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // every wave drains
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // stores drained to L2
   __syncthreads();
   if (threadIdx.x == 0) {
-    const unsigned nb = nwg < 8u ? nwg : 8u;
-    const unsigned lid = blockIdx.y * gridDim.x + blockIdx.x;
-    const unsigned bucket = lid % nb;
-    const unsigned nper = nwg / nb + (bucket < nwg % nb ? 1u : 0u);
-    unsigned* arrive = sync + bucket;
+    unsigned* arrive = sync + ctx.bucket;
     unsigned* top = sync + 8;
-    unsigned* gen = sync + 9 + bucket;
+    unsigned* gen = sync + 9 + ctx.bucket;
     unsigned* give_up = sync + 17;
+    unsigned ok = 1u, spins = 0u;
 
-    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     const unsigned prev = __hip_atomic_fetch_add(
         arrive, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    unsigned ok = 1u, spins = 0u;
-    if (prev + 1u == nper * epoch) {
-      // bucket leader: arrive at the top counter, wait for all buckets,
-      // then publish this bucket's generation word (bucket-local polling)
+    if (prev + 1u == ctx.nper * epoch) {
+      // last arriver on this XCD: one L2 writeback covers every
+      // co-located block's (already vmcnt-drained) stores
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
       __hip_atomic_fetch_add(top, 1u, __ATOMIC_RELAXED,
                              __HIP_MEMORY_SCOPE_AGENT);
       for (;;) {
         if (__hip_atomic_load(top, __ATOMIC_RELAXED,
-                              __HIP_MEMORY_SCOPE_AGENT) >= nb * epoch)
+                              __HIP_MEMORY_SCOPE_AGENT) >= ctx.ntop * epoch)
           break;
         if (__hip_atomic_load(give_up, __ATOMIC_RELAXED,
                               __HIP_MEMORY_SCOPE_AGENT) != 0u ||
@@ -317,6 +379,7 @@ __device__ __forceinline__ bool nats_grid_barrier(unsigned* sync,
         }
         __builtin_amdgcn_s_sleep(2);
       }
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
       __hip_atomic_store(gen, epoch, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_AGENT);
     } else {
@@ -334,13 +397,15 @@ __device__ __forceinline__ bool nats_grid_barrier(unsigned* sync,
         }
         __builtin_amdgcn_s_sleep(2);
       }
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     }
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     ok_sh = ok;
   }
   __syncthreads();
   return ok_sh != 0u;
 }
+
+#define NATS_SYNC_WORDS 27
 
 // stage a [rows][Kpad] bf16 slice into LDS with the (row&15)<<4 byte-XOR
 // swizzle; read back with the same XOR (write+read swizzled together).
@@ -426,6 +491,11 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
   const int wg = blockIdx.x;
   stage_weights_lds(upk_lds, p.Upk + (long)wg * 3 * JB * Hpad, 3 * JB, Hpad);
   __syncthreads();
+  NatsBarrierCtx bctx;
+  if (!nats_barrier_init(sync, nwg, bctx)) {
+    if (threadIdx.x == 0) p.h_all[0] = __builtin_nanf("");
+    return;
+  }
 
   const int wave = threadIdx.x / NATS_WAVE;
   const int m = wave / 3;
@@ -525,7 +595,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
     have_pend = true;
     if (unsafe_nobarrier) {  // TIMING EXPERIMENTS ONLY (racy!)
       __syncthreads();
-    } else if (!nats_grid_barrier(sync, (unsigned)(t + 1), nwg)) {
+    } else if (!nats_grid_barrier(sync, (unsigned)(t + 1), bctx)) {
       // poison output so a barrier give-up surfaces as NaN, never a hang
       if (threadIdx.x == 0) p.h_all[0] = __builtin_nanf("");
       return;
@@ -570,6 +640,11 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
   const int wg = blockIdx.x;
   stage_weights_lds(ub_lds, p.Ubwd + (long)wg * JB * K3pad, JB, K3pad);
   __syncthreads();
+  NatsBarrierCtx bctx;
+  if (!nats_barrier_init(sync, nwg, bctx)) {
+    if (threadIdx.x == 0) p.ddirect[0] = __builtin_nanf("");
+    return;
+  }
 
   const int wave = threadIdx.x / NATS_WAVE;
   const int m = wave / 3;
@@ -681,7 +756,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
     }
     pend_dpre = p.dpre + (long)t * B * 4 * H;
     have_pend = true;
-    if (!nats_grid_barrier(sync, (unsigned)(T - t), nwg)) {
+    if (!nats_grid_barrier(sync, (unsigned)(T - t), bctx)) {
       if (threadIdx.x == 0) p.ddirect[0] = __builtin_nanf("");
       return;
     }
@@ -896,7 +971,7 @@ std::vector<torch::Tensor> gru_scan_fwd_bidir(
   const bool persistent = (2 * ngrp <= 192) && (smem_fwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
-    auto sync = torch::zeros({18}, xg0.options().dtype(torch::kInt32));
+    auto sync = torch::zeros({NATS_SYNC_WORDS}, xg0.options().dtype(torch::kInt32));
     unsigned* sync_p = (unsigned*)sync.data_ptr<int>();
     GruPersistFwd p0{(const bf16_t*)xg0.data_ptr(),
                      (const bf16_t*)xc0.data_ptr(),
@@ -998,7 +1073,7 @@ std::vector<torch::Tensor> gru_scan_bwd_bidir(
   const bool persistent = (2 * ngrp <= 192) && (smem_bwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
-    auto sync = torch::zeros({18}, dh_out0.options().dtype(torch::kInt32));
+    auto sync = torch::zeros({NATS_SYNC_WORDS}, dh_out0.options().dtype(torch::kInt32));
     unsigned* sync_p = (unsigned*)sync.data_ptr<int>();
     GruPersistBwd p0{dh0c.data_ptr<float>(),
                      h_all0.data_ptr<float>(),
@@ -1142,15 +1217,17 @@ std::vector<torch::Tensor> gru_scan_bwd(torch::Tensor dh_out,
 // the persistent scans (timing evidence in profiles/README.md). ----
 __global__ __launch_bounds__(384) void nats_barrier_bench_kernel(
     unsigned* sync, unsigned nwg, int iters, float* out) {
+  NatsBarrierCtx bctx;
+  if (!nats_barrier_init(sync, nwg, bctx)) return;
   for (int t = 0; t < iters; ++t) {
-    if (!nats_grid_barrier(sync, (unsigned)(t + 1), nwg)) return;
+    if (!nats_grid_barrier(sync, (unsigned)(t + 1), bctx)) return;
   }
   if (threadIdx.x == 0 && blockIdx.x == 0 && blockIdx.y == 0) out[0] = 1.f;
 }
 
 double barrier_bench(int nwg_x, int nwg_y, int iters) {
   auto opts = torch::TensorOptions().dtype(torch::kInt32).device(torch::kCUDA);
-  auto sync = torch::zeros({18}, opts);
+  auto sync = torch::zeros({NATS_SYNC_WORDS}, opts);
   auto out = torch::zeros(
       {1}, torch::TensorOptions().dtype(torch::kFloat32).device(torch::kCUDA));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
