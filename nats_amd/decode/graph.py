@@ -21,11 +21,23 @@ import weakref
 _STEPPER_CACHE = weakref.WeakKeyDictionary()
 
 
+def _weights_version(model):
+    """Monotone tag over the decoder weights: in-place optimizer updates
+    bump tensor _version, which must invalidate captured graphs (packed
+    weights are baked into the graph at capture — ops/cond_gru.py
+    _step_packed)."""
+    return tuple(p._version for p in model.P.values())
+
+
 def get_stepper(model, ctx0, pctx0, k):
     """Cached stepper per (model, Ts, k): the captured graph is reused
     across sentences of the same source length — only the static context
-    buffers are refreshed."""
+    buffers are refreshed. Any weight update forces a fresh capture."""
+    ver = _weights_version(model)
     per_model = _STEPPER_CACHE.setdefault(model, {})
+    if per_model.get("__ver__") != ver:
+        per_model.clear()
+        per_model["__ver__"] = ver
     key = (int(ctx0.shape[0]), int(k))
     st = per_model.get(key)
     if st is None:

@@ -182,26 +182,33 @@ def cond_gru_scan_hip(y_gates, y_cand, mask, init_state, ctx, ctx_mask, pctx,
                                None)
 
 
-@torch.no_grad()
-def cond_gru_step_hip(h_prev, x_g, x_c, ctx, ctx_mask, pctx, acc_ctx,
-                      acc_alpha, P):
-    """One decode step = T=1 fused scan with carried accumulators."""
-    ext = _hip_ext()
+_STEP_PACK_KEYS = ("decoder_U", "decoder_Ux", "decoder_U_1", "decoder_W_1",
+                   "decoder_Ux_1", "decoder_Wx_1", "decoder_W_att",
+                   "decoder_b_1", "decoder_bx_1", "decoder_U_att",
+                   "decoder_c_att", "decoder_D_wei", "decoder_W_con",
+                   "decoder_U_con")
+
+
+def _step_packed(P):
+    """Packed decoder weights for the one-step path, cached on the
+    ParameterDict and invalidated by in-place updates (tensor _version):
+    decode runs hundreds of steps against frozen weights, and re-packing
+    (zero-fill + scatter + transpose + bf16 cast of every decoder matrix)
+    per step dominated the non-graph beam loop."""
+    vers = tuple(P[k]._version for k in _STEP_PACK_KEYS)
+    cached = getattr(P, "_nats_step_pack", None)
+    if cached is not None and cached[0] == vers:
+        return cached[1]
     H = P["decoder_Ux"].shape[1]
-    C = ctx.shape[2]
     A = P["decoder_U_att"].shape[0]
+    C = P["decoder_W_1"].shape[0]
     Hpad, Cpad = _ceil(H, 32), _ceil(C, 32)
-    Upk2 = pack_fwd_weights(P["decoder_U"], P["decoder_Ux"])
-    W1pk = pack_gru1_weights(P["decoder_U_1"].float(), P["decoder_W_1"].float(),
-                             P["decoder_Ux_1"].float(),
-                             P["decoder_Wx_1"].float(), Hpad, Cpad)
-    WattPk = _pack_rows(P["decoder_W_att"].float().t(), _ceil(A, 16), Hpad)
-    outs = ext.cond_gru_fwd(
-        x_g.unsqueeze(0).to(torch.bfloat16).contiguous(),
-        x_c.unsqueeze(0).to(torch.bfloat16).contiguous(),
-        None, h_prev, ctx.to(torch.bfloat16).contiguous(),
-        ctx_mask.float().contiguous() if ctx_mask is not None else None,
-        pctx.float().contiguous(), Upk2, W1pk, WattPk,
+    packed = (
+        pack_fwd_weights(P["decoder_U"], P["decoder_Ux"]),
+        pack_gru1_weights(P["decoder_U_1"].float(), P["decoder_W_1"].float(),
+                          P["decoder_Ux_1"].float(),
+                          P["decoder_Wx_1"].float(), Hpad, Cpad),
+        _pack_rows(P["decoder_W_att"].float().t(), _ceil(A, 16), Hpad),
         P["decoder_b_1"].float().contiguous(),
         P["decoder_bx_1"].float().contiguous(),
         P["decoder_U_att"].float().reshape(-1).contiguous(),
@@ -209,6 +216,28 @@ def cond_gru_step_hip(h_prev, x_g, x_c, ctx, ctx_mask, pctx, acc_ctx,
         P["decoder_D_wei"].float().reshape(-1).contiguous(),
         P["decoder_W_con"].float().reshape(-1).contiguous(),
         P["decoder_U_con"].float().reshape(-1).contiguous(),
+    )
+    try:
+        P._nats_step_pack = (vers, packed)
+    except Exception:
+        pass
+    return packed
+
+
+@torch.no_grad()
+def cond_gru_step_hip(h_prev, x_g, x_c, ctx, ctx_mask, pctx, acc_ctx,
+                      acc_alpha, P):
+    """One decode step = T=1 fused scan with carried accumulators."""
+    ext = _hip_ext()
+    (Upk2, W1pk, WattPk, b1, bx1, Uatt, catt, Dwei, Wcon,
+     Ucon) = _step_packed(P)
+    outs = ext.cond_gru_fwd(
+        x_g.unsqueeze(0).to(torch.bfloat16).contiguous(),
+        x_c.unsqueeze(0).to(torch.bfloat16).contiguous(),
+        None, h_prev, ctx.to(torch.bfloat16).contiguous(),
+        ctx_mask.float().contiguous() if ctx_mask is not None else None,
+        pctx.float().contiguous(), Upk2, W1pk, WattPk, b1, bx1, Uatt, catt,
+        Dwei, Wcon, Ucon,
         acc_ctx.float().contiguous(), acc_alpha.float().contiguous())
     h2_all, ctxs_all, alphas_all, accC, accA = outs[:5]
     return h2_all[0], ctxs_all[0], alphas_all[0], accC, accA
