@@ -201,35 +201,33 @@ __global__ void cond_attn_gate_fwd(
   accC[(long)b * C + c] = accCv + mm * g;
 }
 
-// ---------------- GRU_1 forward ----------------
-// 16 waves: wave w -> (m = w/8, g = (w%8)/2, ks = w%2) with output groups
-// g0 = r2 (K = h1|ctx), g1 = u2 (K = h1|ctx), g2 = pxa (h1@Ux_1),
-// g3 = pxb (ctx@Wx_1) — K-selectivity comes from zero blocks in W1pk.
-// K is split across wave pairs (K1 is ~3H — the serial chain at 8 waves
-// measured 34 us/launch, latency-bound).
-__global__ __launch_bounds__(1024) void cond_gru1_step_fwd(
+// ---------------- GRU_1 forward (split-K) ----------------
+// The single-kernel variant ran ngrpH (=63 at dim 1000) workgroups — a
+// quarter of the chip — at 31.4us/step (profiles/
+// cnn_kernel_stats_barrierv2.csv). Split-K: grid (ngrpH, KS) blocks each
+// compute a K-quarter of all four gate GEMMs (16 waves: m x group x
+// half-split as before) and store fp32 partials; a pointwise kernel sums
+// the KS partials and applies the gate nonlinearities (nats.py:529-553).
+__global__ __launch_bounds__(1024) void cond_gru1_gemm_splitk(
     const bf16_t* __restrict__ hc_bf,  // [32][K1] = [h1 | ctx_t] bf16
-    const float* __restrict__ h1_t,    // [B][H] fp32
     const bf16_t* __restrict__ W1pk,   // [ngrp*4*16][K1]
-    const float* __restrict__ b1,      // [2H]
-    const float* __restrict__ bx1,     // [H]
-    const float* __restrict__ mask_t,  // [B] or null
-    float* __restrict__ h2_t,          // [B][H] out
-    bf16_t* __restrict__ h2bf_out,     // [32][Hpad] out
-    int ld_h2bf,
-    bf16_t* __restrict__ saved1_t,     // [B][4H] (r2,u2,pxa,hbar)
-    int B, int H, int K1) {
+    float* __restrict__ part,          // [KS][4][32][Hpad]
+    int H, int K1, int Hpad) {
   __shared__ float pre[4][2][32][JB + 1];
 
   const int wg = blockIdx.x;
+  const int ksb = blockIdx.y;
+  const int KS = gridDim.y;
   const int wave = threadIdx.x / NATS_WAVE;
   const int m = wave / 8;
   const int g = (wave % 8) / 2;
   const int ks = wave % 2;
   const int j0 = wg * JB;
-  const int khalf = ((K1 / 2 + 31) / 32) * 32;
-  const int kbeg = ks * khalf;
-  const int kend = min(K1, (ks + 1) * khalf);
+  const int nchunk = 2 * KS;
+  const int kchunk = ((K1 / nchunk + 31) / 32) * 32;
+  const int kidx = 2 * ksb + ks;
+  const int kbeg = min(K1, kidx * kchunk);
+  const int kend = min(K1, (kidx + 1) * kchunk);
 
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   const bf16_t* brow = W1pk + (long)(wg * 4 + g) * JB * K1;
@@ -243,25 +241,53 @@ __global__ __launch_bounds__(1024) void cond_gru1_step_fwd(
   }
   __syncthreads();
 
-  for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
-    const int b = idx / JB;
+  float* dst = part + ((long)ksb * 4 + 0) * 32 * Hpad;
+  for (int idx = threadIdx.x; idx < 4 * 32 * JB; idx += blockDim.x) {
+    const int gg = idx / (32 * JB);
+    const int b = (idx / JB) % 32;
     const int c = idx % JB;
-    const int j = j0 + c;
-    if (j >= H) continue;
-    const float r2 =
-        nats_sigmoid(pre[0][0][b][c] + pre[0][1][b][c] + b1[j]);
-    const float u2 =
-        nats_sigmoid(pre[1][0][b][c] + pre[1][1][b][c] + b1[H + j]);
-    const float pxa = pre[2][0][b][c] + pre[2][1][b][c];
-    const float pxb = pre[3][0][b][c] + pre[3][1][b][c];
-    const float hbar = tanhf((pxa + bx1[j]) * r2 + pxb);
-    const float h1v = h1_t[(long)b * H + j];
+    dst[((long)gg * 32 + b) * Hpad + j0 + c] =
+        pre[gg][0][b][c] + pre[gg][1][b][c];
+  }
+}
+
+__global__ void cond_gru1_step_pointwise(
+    const float* __restrict__ part,    // [KS][4][32][Hpad]
+    int KS,
+    const float* __restrict__ h1_t,    // [B][H] fp32
+    const float* __restrict__ b1,      // [2H]
+    const float* __restrict__ bx1,     // [H]
+    const float* __restrict__ mask_t,  // [B] or null
+    float* __restrict__ h2_t,          // [B][H] out
+    bf16_t* __restrict__ h2bf_out,     // [32][ld_h2bf] out
+    int ld_h2bf,
+    bf16_t* __restrict__ saved1_t,     // [B][4H] (r2,u2,pxa,hbar)
+    int B, int H, int Hpad) {
+  const long total = (long)B * H;
+  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int b = idx / H;
+    const int j = idx % H;
+    float p0 = 0.f, p1 = 0.f, p2 = 0.f, p3 = 0.f;
+    const long bj = (long)b * Hpad + j;
+    for (int k = 0; k < KS; ++k) {
+      const float* pk = part + (long)k * 4 * 32 * Hpad;
+      p0 += pk[bj];
+      p1 += pk[(long)32 * Hpad + bj];
+      p2 += pk[(long)2 * 32 * Hpad + bj];
+      p3 += pk[(long)3 * 32 * Hpad + bj];
+    }
+    const float r2 = nats_sigmoid(p0 + b1[j]);
+    const float u2 = nats_sigmoid(p1 + b1[H + j]);
+    const float pxa = p2;
+    const float hbar = tanhf((pxa + bx1[j]) * r2 + p3);
+    const float h1v = h1_t[idx];
     float h2 = u2 * h1v + (1.f - u2) * hbar;
     if (mask_t != nullptr) {
       const float mm = mask_t[b];
       h2 = mm * h2 + (1.f - mm) * h1v;
     }
-    h2_t[(long)b * H + j] = h2;
+    h2_t[idx] = h2;
     h2bf_out[(long)b * ld_h2bf + j] = (bf16_t)h2;
     saved1_t[(long)b * 4 * H + j] = (bf16_t)r2;
     saved1_t[(long)b * 4 * H + H + j] = (bf16_t)u2;
@@ -629,6 +655,8 @@ std::vector<torch::Tensor> cond_gru_fwd(
   auto accC = accC0.has_value() ? accC0->contiguous().to(torch::kFloat32)
                                 : torch::zeros({B, C}, optsF);
   auto h2bf = torch::zeros({2, 32, Hpad}, optsB);
+  const int GRU1_KS = 4;
+  auto gru1_part = torch::empty({GRU1_KS, 4, 32, Hpad}, optsF);
   auto hc_bf = torch::zeros({32, K1}, optsB);
   auto e_buf = torch::empty({Ts, B}, optsF);
   auto ctxpre_f32 = torch::empty({B, C}, optsF);
@@ -704,16 +732,19 @@ std::vector<torch::Tensor> cond_gru_fwd(
                        (bf16_t*)ctxpre_all.data_ptr() + (long)t * B * C,
                        ctxs_all.data_ptr<float>() + (long)t * B * C,
                        (bf16_t*)hc_bf.data_ptr(), Hpad, K1, mt, B, C);
-    // 5) GRU_1 -> h2
-    hipLaunchKernelGGL(cond_gru1_step_fwd, dim3(ngrpH), dim3(1024), 0, stream,
-                       (const bf16_t*)hc_bf.data_ptr(),
-                       h1_all.data_ptr<float>() + (long)t * B * H,
-                       (const bf16_t*)W1pk.data_ptr(), b1.data_ptr<float>(),
-                       bx1.data_ptr<float>(), mt,
+    // 5) GRU_1 -> h2 (split-K GEMM + pointwise combine)
+    hipLaunchKernelGGL(cond_gru1_gemm_splitk, dim3(ngrpH, GRU1_KS),
+                       dim3(1024), 0, stream, (const bf16_t*)hc_bf.data_ptr(),
+                       (const bf16_t*)W1pk.data_ptr(),
+                       gru1_part.data_ptr<float>(), H, K1, Hpad);
+    hipLaunchKernelGGL(cond_gru1_step_pointwise, dim3(cdiv_i(B * H, 256)),
+                       dim3(256), 0, stream, gru1_part.data_ptr<float>(),
+                       GRU1_KS, h1_all.data_ptr<float>() + (long)t * B * H,
+                       b1.data_ptr<float>(), bx1.data_ptr<float>(), mt,
                        h2_all.data_ptr<float>() + (long)t * B * H,
                        h2bf_p + ((t + 1) % 2) * hbstride, Hpad,
                        (bf16_t*)saved1.data_ptr() + (long)t * B * 4 * H, B, H,
-                       K1);
+                       Hpad);
   }
   HIP_CHECK(hipGetLastError());
   return {h2_all, ctxs_all, alphas_all, accC, accA, h1_all, saved2, saved1,
