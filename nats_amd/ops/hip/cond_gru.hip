@@ -463,13 +463,21 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_scatter(
     const float* __restrict__ pstate_t,     // [B][A]
     const float* __restrict__ accA_used_t,  // [B][Ts]
     const float* __restrict__ Dwei, const float* __restrict__ Uatt,
-    float* __restrict__ daccA,              // [B][Ts] in/out
+    float* __restrict__ daccA,              // [B][Ts] (+=, atomic)
     float* __restrict__ dpctx_acc,          // [Ts][B][A] (+=)
     bf16_t* __restrict__ pc_buf,            // [B][A][Tpad8]
     int B, int Ts, int A, int Tpad8) {
+  // grid (B, s-chunks, A-chunks): the un-chunked variant ran only
+  // B * Ts/256 blocks (2 waves/CU at the CNN/DM shape) and was
+  // latency-bound on its serial per-A float4 RMW chain.
   const int b = blockIdx.x;
   const int s = blockIdx.y * blockDim.x + threadIdx.x;
   if (s >= Ts) return;
+  const int ACH = gridDim.z;
+  const int chunkA = ((A + ACH - 1) / ACH + 3) & ~3;  // 16B-aligned splits
+  const int ibeg = blockIdx.z * chunkA;
+  const int iend = min(A, ibeg + chunkA);
+  if (ibeg >= iend) return;
   const float al = alphas_t[(long)b * Ts + s];
   const float de = al * (dal_buf[(long)s * B + b] - dot_buf[b]);
   const float accAu = accA_used_t[(long)b * Ts + s];
@@ -477,8 +485,8 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_scatter(
   const float* srow = pstate_t + (long)b * A;
   float* dprow = dpctx_acc + ((long)s * B + b) * A;
   float daccA_add = 0.f;
-  const int A4 = (A % 4 == 0) ? A : 0;  // f32x4 rows need A % 4 == 0
-  int i = 0;
+  const int A4 = (A % 4 == 0) ? iend : ibeg;  // f32x4 needs A % 4 == 0
+  int i = ibeg;
   for (; i < A4; i += 4) {
     const float4 p = *(const float4*)(prow + i);
     const float4 st = *(const float4*)(srow + i);
@@ -504,14 +512,18 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_scatter(
     pc_buf[((long)b * A + i + 2) * Tpad8 + s] = (bf16_t)pc2;
     pc_buf[((long)b * A + i + 3) * Tpad8 + s] = (bf16_t)pc3;
   }
-  for (; i < A; ++i) {
+  for (; i < iend; ++i) {
     const float pc = tanhf(prow[i] + srow[i] + accAu * Dwei[i]);
     const float dpc = de * (1.f - pc * pc) * Uatt[i];
     dprow[i] += dpc;
     daccA_add += dpc * Dwei[i];
     pc_buf[((long)b * A + i) * Tpad8 + s] = (bf16_t)pc;
   }
-  daccA[(long)b * Ts + s] += daccA_add;
+  if (ACH == 1) {
+    daccA[(long)b * Ts + s] += daccA_add;
+  } else {
+    atomicAdd(daccA + (long)b * Ts + s, daccA_add);
+  }
 }
 
 // attention backward, stage 3 (grid (b, s-chunk)): reduce over an
@@ -894,7 +906,8 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        alphas_all.data_ptr<float>() + (long)t * B * Ts,
                        dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
                        B, Ts);
-    hipLaunchKernelGGL(cond_attn_bwd_scatter, dim3(B, cdiv_i(Ts, 256)),
+    hipLaunchKernelGGL(cond_attn_bwd_scatter,
+                       dim3(B, cdiv_i(Ts, 256), A >= 16 ? 4 : 1),
                        dim3(256), 0, stream,
                        alphas_all.data_ptr<float>() + (long)t * B * Ts,
                        dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
