@@ -669,6 +669,8 @@ std::vector<torch::Tensor> cond_gru_fwd(
   auto h2bf = torch::zeros({2, 32, Hpad}, optsB);
   const int GRU1_KS = 4;
   auto gru1_part = torch::empty({GRU1_KS, 4, 32, Hpad}, optsF);
+  const int GRU2_KS = 4;
+  auto gru2_part = torch::empty({GRU2_KS, 3, 32, Hpad}, optsF);
   auto hc_bf = torch::zeros({32, K1}, optsB);
   auto e_buf = torch::empty({Ts, B}, optsF);
   auto ctxpre_f32 = torch::empty({B, C}, optsF);
@@ -699,10 +701,14 @@ std::vector<torch::Tensor> cond_gru_fwd(
                               ? init_f.data_ptr<float>()
                               : h2_all.data_ptr<float>() + (long)(t - 1) * B * H;
     const float* mt = mask_p ? mask_p + (long)t * B : nullptr;
-    // 1) GRU_2 -> h1 (bf16 into hc_bf cols [0,H))
-    hipLaunchKernelGGL(nats_gru_step_fwd, dim3(ngrpH), dim3(384), 0, stream,
-                       h2bf_p + (t % 2) * hbstride, h2prev,
+    // 1) GRU_2 -> h1 (bf16 into hc_bf cols [0,H)); split-K + pointwise
+    hipLaunchKernelGGL(nats_gru2_gemm_splitk, dim3(ngrpH, GRU2_KS), dim3(384),
+                       0, stream, h2bf_p + (t % 2) * hbstride,
                        (const bf16_t*)Upk2.data_ptr(),
+                       gru2_part.data_ptr<float>(), Hpad);
+    hipLaunchKernelGGL(nats_gru2_step_pointwise, dim3(cdiv_i(B * H, 256)),
+                       dim3(256), 0, stream, gru2_part.data_ptr<float>(),
+                       GRU2_KS, h2prev,
                        (const bf16_t*)yg.data_ptr() + (long)t * B * 2 * H,
                        (const bf16_t*)yc.data_ptr() + (long)t * B * H, mt,
                        h1_all.data_ptr<float>() + (long)t * B * H,

@@ -15,6 +15,14 @@ __global__ __launch_bounds__(384) void nats_gru_step_fwd(
     float* __restrict__ h_out, bf16_t* __restrict__ h_bf_out, int ld_bfout,
     bf16_t* __restrict__ saved_t, int B, int H, int Hpad);
 
+__global__ __launch_bounds__(384) void nats_gru2_gemm_splitk(const bf16_t* h_bf, const bf16_t* Upk,
+                                      float* part, int Hpad);
+
+__global__ void nats_gru2_step_pointwise(
+    const float* part, int KS, const float* h_prev, const bf16_t* xg_t,
+    const bf16_t* xc_t, const float* mask_t, float* h_out, bf16_t* h_bf_out,
+    int ld_bfout, bf16_t* saved_t, int B, int H, int Hpad);
+
 // backward pointwise: dh -> gate preactivation grads. dh_out_t may be null.
 __global__ void nats_gru_step_bwd_pointwise(
     const float* __restrict__ dh_buf, const float* __restrict__ dh_out_t,

@@ -97,6 +97,78 @@ __global__ __launch_bounds__(384) void nats_gru_step_fwd(
   }
 }
 
+// ---------------- forward step, split-K (decoder GRU_2) ----------------
+// Same math as nats_gru_step_fwd but grid (ngrp, KS): each block computes
+// a K-chunk of the three gate GEMMs and stores fp32 partials (no LDS
+// exchange needed — one chunk per block); nats_gru2_step_pointwise sums
+// the chunks and applies gates. Raises block count 4x for the decoder's
+// per-step call (the fused variant ran ngrp=63 blocks, 25% of the chip).
+__global__ __launch_bounds__(384) void nats_gru2_gemm_splitk(
+    const bf16_t* __restrict__ h_bf,  // [32][Hpad]
+    const bf16_t* __restrict__ Upk,   // [ngrp*3*16][Hpad]
+    float* __restrict__ part,         // [KS][3][32][Hpad]
+    int Hpad) {
+  const int wg = blockIdx.x;
+  const int ksb = blockIdx.y;
+  const int KS = gridDim.y;
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int m = wave / 3;
+  const int g = wave % 3;
+  const int kchunk = ((Hpad / KS + 31) / 32) * 32;
+  const int kbeg = min(Hpad, ksb * kchunk);
+  const int kend = min(Hpad, kbeg + kchunk);
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const bf16_t* brow = Upk + (long)(wg * 3 + g) * JB * Hpad;
+  NATS_MFMA_KLOOP(acc, h_bf, 16 * m, Hpad, brow, 0, Hpad, kbeg, kend);
+  store_cd_rowmajor(part + ((long)ksb * 3 + g) * 32 * Hpad, acc, 16 * m,
+                    Hpad, wg * JB);
+}
+
+__global__ void nats_gru2_step_pointwise(
+    const float* __restrict__ part,    // [KS][3][32][Hpad]
+    int KS,
+    const float* __restrict__ h_prev,  // [B][H]
+    const bf16_t* __restrict__ xg_t,   // [B][2H]
+    const bf16_t* __restrict__ xc_t,   // [B][H]
+    const float* __restrict__ mask_t,  // [B] or null
+    float* __restrict__ h_out,         // [B][H]
+    bf16_t* __restrict__ h_bf_out,     // row stride ld_bfout
+    int ld_bfout,
+    bf16_t* __restrict__ saved_t,      // [B][3H] (r,u,pxl)
+    int B, int H, int Hpad) {
+  const long total = (long)B * H;
+  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int b = idx / H;
+    const int j = idx % H;
+    float p0 = 0.f, p1 = 0.f, p2 = 0.f;
+    const long bj = (long)b * Hpad + j;
+    for (int k = 0; k < KS; ++k) {
+      const float* pk = part + (long)k * 3 * 32 * Hpad;
+      p0 += pk[bj];
+      p1 += pk[(long)32 * Hpad + bj];
+      p2 += pk[(long)2 * 32 * Hpad + bj];
+    }
+    const float hp = h_prev[idx];
+    const float pr = p0 + (float)xg_t[(long)b * 2 * H + j];
+    const float pu = p1 + (float)xg_t[(long)b * 2 * H + H + j];
+    const float px = p2;
+    const float r = nats_sigmoid(pr);
+    const float u = nats_sigmoid(pu);
+    const float hbar = tanhf(px * r + (float)xc_t[idx]);
+    float hnew = u * hp + (1.f - u) * hbar;
+    if (mask_t != nullptr) {
+      const float mm = mask_t[b];
+      hnew = mm * hnew + (1.f - mm) * hp;
+    }
+    h_out[idx] = hnew;
+    h_bf_out[(long)b * ld_bfout + j] = (bf16_t)hnew;
+    saved_t[(long)b * 3 * H + j] = (bf16_t)r;
+    saved_t[(long)b * 3 * H + H + j] = (bf16_t)u;
+    saved_t[(long)b * 3 * H + 2 * H + j] = (bf16_t)px;
+  }
+}
+
 // ---------------- backward pointwise ----------------
 __global__ void nats_gru_step_bwd_pointwise(
     const float* __restrict__ dh_buf,   // [B][H] recurrent dh (from t+1)
