@@ -202,6 +202,75 @@ __global__ void cond_attn_gate_fwd(
   accC[(long)b * C + c] = accCv + mm * g;
 }
 
+// ---------------- GRU_1 forward (fused, small-K path) ----------------
+// 16 waves: wave w -> (m = w/8, g = (w%8)/2, ks = w%2) with output groups
+// g0 = r2 (K = h1|ctx), g1 = u2 (K = h1|ctx), g2 = pxa (h1@Ux_1),
+// g3 = pxb (ctx@Wx_1) — K-selectivity comes from zero blocks in W1pk.
+// K is split across wave pairs (K1 is ~3H — the serial chain at 8 waves
+// measured 34 us/launch, latency-bound).
+__global__ __launch_bounds__(1024) void cond_gru1_step_fwd(
+    const bf16_t* __restrict__ hc_bf,  // [32][K1] = [h1 | ctx_t] bf16
+    const float* __restrict__ h1_t,    // [B][H] fp32
+    const bf16_t* __restrict__ W1pk,   // [ngrp*4*16][K1]
+    const float* __restrict__ b1,      // [2H]
+    const float* __restrict__ bx1,     // [H]
+    const float* __restrict__ mask_t,  // [B] or null
+    float* __restrict__ h2_t,          // [B][H] out
+    bf16_t* __restrict__ h2bf_out,     // [32][Hpad] out
+    int ld_h2bf,
+    bf16_t* __restrict__ saved1_t,     // [B][4H] (r2,u2,pxa,hbar)
+    int B, int H, int K1) {
+  __shared__ float pre[4][2][32][JB + 1];
+
+  const int wg = blockIdx.x;
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int m = wave / 8;
+  const int g = (wave % 8) / 2;
+  const int ks = wave % 2;
+  const int j0 = wg * JB;
+  const int khalf = ((K1 / 2 + 31) / 32) * 32;
+  const int kbeg = ks * khalf;
+  const int kend = min(K1, (ks + 1) * khalf);
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const bf16_t* brow = W1pk + (long)(wg * 4 + g) * JB * K1;
+  NATS_MFMA_KLOOP(acc, hc_bf, 16 * m, K1, brow, 0, K1, kbeg, kend);
+  {
+    const int lane = threadIdx.x & (NATS_WAVE - 1);
+    const int col = lane & 15;
+    const int rbase = 16 * m + (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) pre[g][ks][rbase + i][col] = acc[i];
+  }
+  __syncthreads();
+
+  for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+    const int b = idx / JB;
+    const int c = idx % JB;
+    const int j = j0 + c;
+    if (j >= H) continue;
+    const float r2 =
+        nats_sigmoid(pre[0][0][b][c] + pre[0][1][b][c] + b1[j]);
+    const float u2 =
+        nats_sigmoid(pre[1][0][b][c] + pre[1][1][b][c] + b1[H + j]);
+    const float pxa = pre[2][0][b][c] + pre[2][1][b][c];
+    const float pxb = pre[3][0][b][c] + pre[3][1][b][c];
+    const float hbar = tanhf((pxa + bx1[j]) * r2 + pxb);
+    const float h1v = h1_t[(long)b * H + j];
+    float h2 = u2 * h1v + (1.f - u2) * hbar;
+    if (mask_t != nullptr) {
+      const float mm = mask_t[b];
+      h2 = mm * h2 + (1.f - mm) * h1v;
+    }
+    h2_t[(long)b * H + j] = h2;
+    h2bf_out[(long)b * ld_h2bf + j] = (bf16_t)h2;
+    saved1_t[(long)b * 4 * H + j] = (bf16_t)r2;
+    saved1_t[(long)b * 4 * H + H + j] = (bf16_t)u2;
+    saved1_t[(long)b * 4 * H + 2 * H + j] = (bf16_t)pxa;
+    saved1_t[(long)b * 4 * H + 3 * H + j] = (bf16_t)hbar;
+  }
+}
+
 // ---------------- GRU_1 forward (split-K) ----------------
 // The single-kernel variant ran ngrpH (=63 at dim 1000) workgroups — a
 // quarter of the chip — at 31.4us/step (profiles/
@@ -702,20 +771,32 @@ std::vector<torch::Tensor> cond_gru_fwd(
                               ? init_f.data_ptr<float>()
                               : h2_all.data_ptr<float>() + (long)(t - 1) * B * H;
     const float* mt = mask_p ? mask_p + (long)t * B : nullptr;
-    // 1) GRU_2 -> h1 (bf16 into hc_bf cols [0,H)); split-K + pointwise
-    hipLaunchKernelGGL(nats_gru2_gemm_splitk, dim3(ngrpH, GRU2_KS), dim3(384),
-                       0, stream, h2bf_p + (t % 2) * hbstride,
-                       (const bf16_t*)Upk2.data_ptr(),
-                       gru2_part.data_ptr<float>(), Hpad);
-    hipLaunchKernelGGL(nats_gru2_step_pointwise, dim3(cdiv_i(B * H, 256)),
-                       dim3(256), 0, stream, gru2_part.data_ptr<float>(),
-                       GRU2_KS, h2prev,
-                       (const bf16_t*)yg.data_ptr() + (long)t * B * 2 * H,
-                       (const bf16_t*)yc.data_ptr() + (long)t * B * H, mt,
-                       h1_all.data_ptr<float>() + (long)t * B * H,
-                       (bf16_t*)hc_bf.data_ptr(), K1,
-                       (bf16_t*)saved2.data_ptr() + (long)t * B * 3 * H, B, H,
-                       Hpad);
+    // 1) GRU_2 -> h1 (bf16 into hc_bf cols [0,H)); split-K when K large
+    if (Hpad >= 1024) {
+      hipLaunchKernelGGL(nats_gru2_gemm_splitk, dim3(ngrpH, GRU2_KS),
+                         dim3(384), 0, stream, h2bf_p + (t % 2) * hbstride,
+                         (const bf16_t*)Upk2.data_ptr(),
+                         gru2_part.data_ptr<float>(), Hpad);
+      hipLaunchKernelGGL(nats_gru2_step_pointwise, dim3(cdiv_i(B * H, 256)),
+                         dim3(256), 0, stream, gru2_part.data_ptr<float>(),
+                         GRU2_KS, h2prev,
+                         (const bf16_t*)yg.data_ptr() + (long)t * B * 2 * H,
+                         (const bf16_t*)yc.data_ptr() + (long)t * B * H, mt,
+                         h1_all.data_ptr<float>() + (long)t * B * H,
+                         (bf16_t*)hc_bf.data_ptr(), K1,
+                         (bf16_t*)saved2.data_ptr() + (long)t * B * 3 * H, B,
+                         H, Hpad);
+    } else {
+      hipLaunchKernelGGL(nats_gru_step_fwd, dim3(ngrpH), dim3(384), 0, stream,
+                         h2bf_p + (t % 2) * hbstride, h2prev,
+                         (const bf16_t*)Upk2.data_ptr(),
+                         (const bf16_t*)yg.data_ptr() + (long)t * B * 2 * H,
+                         (const bf16_t*)yc.data_ptr() + (long)t * B * H, mt,
+                         h1_all.data_ptr<float>() + (long)t * B * H,
+                         (bf16_t*)hc_bf.data_ptr(), K1,
+                         (bf16_t*)saved2.data_ptr() + (long)t * B * 3 * H, B,
+                         H, Hpad);
+    }
     // 2) pstate = h1 @ W_att
     hipLaunchKernelGGL(cond_small_gemm_bt, dim3(2, Apad16 / 16), dim3(64), 0,
                        stream, (const bf16_t*)hc_bf.data_ptr(),
@@ -751,19 +832,33 @@ std::vector<torch::Tensor> cond_gru_fwd(
                        (bf16_t*)ctxpre_all.data_ptr() + (long)t * B * C,
                        ctxs_all.data_ptr<float>() + (long)t * B * C,
                        (bf16_t*)hc_bf.data_ptr(), Hpad, K1, mt, B, C);
-    // 5) GRU_1 -> h2 (split-K GEMM + pointwise combine)
-    hipLaunchKernelGGL(cond_gru1_gemm_splitk, dim3(ngrpH, GRU1_KS),
-                       dim3(1024), 0, stream, (const bf16_t*)hc_bf.data_ptr(),
-                       (const bf16_t*)W1pk.data_ptr(),
-                       gru1_part.data_ptr<float>(), H, K1, Hpad);
-    hipLaunchKernelGGL(cond_gru1_step_pointwise, dim3(cdiv_i(B * H, 256)),
-                       dim3(256), 0, stream, gru1_part.data_ptr<float>(),
-                       GRU1_KS, h1_all.data_ptr<float>() + (long)t * B * H,
-                       b1.data_ptr<float>(), bx1.data_ptr<float>(), mt,
-                       h2_all.data_ptr<float>() + (long)t * B * H,
-                       h2bf_p + ((t + 1) % 2) * hbstride, Hpad,
-                       (bf16_t*)saved1.data_ptr() + (long)t * B * 4 * H, B, H,
-                       Hpad);
+    // 5) GRU_1 -> h2: split-K for large K (raises block parallelism 4x),
+    // single fused kernel when K is small (split overhead dominates)
+    if (K1 >= 2048) {
+      hipLaunchKernelGGL(cond_gru1_gemm_splitk, dim3(ngrpH, GRU1_KS),
+                         dim3(1024), 0, stream,
+                         (const bf16_t*)hc_bf.data_ptr(),
+                         (const bf16_t*)W1pk.data_ptr(),
+                         gru1_part.data_ptr<float>(), H, K1, Hpad);
+      hipLaunchKernelGGL(cond_gru1_step_pointwise, dim3(cdiv_i(B * H, 256)),
+                         dim3(256), 0, stream, gru1_part.data_ptr<float>(),
+                         GRU1_KS, h1_all.data_ptr<float>() + (long)t * B * H,
+                         b1.data_ptr<float>(), bx1.data_ptr<float>(), mt,
+                         h2_all.data_ptr<float>() + (long)t * B * H,
+                         h2bf_p + ((t + 1) % 2) * hbstride, Hpad,
+                         (bf16_t*)saved1.data_ptr() + (long)t * B * 4 * H, B,
+                         H, Hpad);
+    } else {
+      hipLaunchKernelGGL(cond_gru1_step_fwd, dim3(ngrpH), dim3(1024), 0,
+                         stream, (const bf16_t*)hc_bf.data_ptr(),
+                         h1_all.data_ptr<float>() + (long)t * B * H,
+                         (const bf16_t*)W1pk.data_ptr(), b1.data_ptr<float>(),
+                         bx1.data_ptr<float>(), mt,
+                         h2_all.data_ptr<float>() + (long)t * B * H,
+                         h2bf_p + ((t + 1) % 2) * hbstride, Hpad,
+                         (bf16_t*)saved1.data_ptr() + (long)t * B * 4 * H, B,
+                         H, K1);
+    }
   }
   HIP_CHECK(hipGetLastError());
   return {h2_all, ctxs_all, alphas_all, accC, accA, h1_all, saved2, saved1,

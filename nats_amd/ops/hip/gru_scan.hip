@@ -560,6 +560,10 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
 
   const GruPersistFwd& p = (blockIdx.y == 0) ? p0 : p1;
   const int wg = blockIdx.x;
+  // per-direction barrier: the two directions are data-independent, so
+  // each grid.y half syncs only its own gridDim.x blocks (own sync slab)
+  sync += (long)blockIdx.y * NATS_SYNC_WORDS;
+  nwg = gridDim.x;
   stage_weights_lds(upk_lds, p.Upk + (long)wg * 3 * JB * Hpad, 3 * JB, Hpad);
   __syncthreads();
   NatsBarrierCtx bctx;
@@ -709,6 +713,8 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
 
   const GruPersistBwd& p = (blockIdx.y == 0) ? p0 : p1;
   const int wg = blockIdx.x;
+  sync += (long)blockIdx.y * NATS_SYNC_WORDS;
+  nwg = gridDim.x;
   stage_weights_lds(ub_lds, p.Ubwd + (long)wg * JB * K3pad, JB, K3pad);
   __syncthreads();
   NatsBarrierCtx bctx;
@@ -1042,7 +1048,7 @@ std::vector<torch::Tensor> gru_scan_fwd_bidir(
   const bool persistent = (2 * ngrp <= 192) && (smem_fwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
-    auto sync = torch::zeros({NATS_SYNC_WORDS}, xg0.options().dtype(torch::kInt32));
+    auto sync = torch::zeros({2 * NATS_SYNC_WORDS}, xg0.options().dtype(torch::kInt32));
     unsigned* sync_p = (unsigned*)sync.data_ptr<int>();
     GruPersistFwd p0{(const bf16_t*)xg0.data_ptr(),
                      (const bf16_t*)xc0.data_ptr(),
@@ -1144,7 +1150,7 @@ std::vector<torch::Tensor> gru_scan_bwd_bidir(
   const bool persistent = (2 * ngrp <= 192) && (smem_bwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
-    auto sync = torch::zeros({NATS_SYNC_WORDS}, dh_out0.options().dtype(torch::kInt32));
+    auto sync = torch::zeros({2 * NATS_SYNC_WORDS}, dh_out0.options().dtype(torch::kInt32));
     unsigned* sync_p = (unsigned*)sync.data_ptr<int>();
     GruPersistBwd p0{dh0c.data_ptr<float>(),
                      h_all0.data_ptr<float>(),
