@@ -31,20 +31,29 @@ namespace {
 constexpr int JB = 16;
 
 // ---------------- small GEMM: pstate = h1 @ W_att ----------------
-// A = [32][ldK] bf16 (zero-padded), Bt = [Npad][ldK] bf16; C f32 [B][N].
+// A = [32][ldK] bf16 (zero-padded), Bt = [Npad][ldK] bf16. Split-K over
+// grid.z into Cpart [KS][32][N] f32 partials (the unsplit version was 14
+// waves on the whole chip, 11.3us of pure load latency): consumers
+// (cond_attn_escore inline, cond_attn_softmax for the saved pstate_all)
+// sum the KS partials.
 __global__ void cond_small_gemm_bt(const bf16_t* __restrict__ A,
                                    const bf16_t* __restrict__ Bt,
-                                   float* __restrict__ C, int B, int N,
+                                   float* __restrict__ Cpart, int B, int N,
                                    int ldK, int Kpad) {
   const int m0 = blockIdx.x * 16;
   const int n0 = blockIdx.y * 16;
+  const int KS = gridDim.z;
+  const int kchunk = ((Kpad / KS + 31) / 32) * 32;
+  const int kbeg = min(Kpad, (int)blockIdx.z * kchunk);
+  const int kend = min(Kpad, kbeg + kchunk);
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   // NB: Bt's row stride is Kpad (the packed weight width), NOT ldK.
-  NATS_MFMA_KLOOP(acc, A, m0, ldK, Bt, n0, Kpad, 0, Kpad);
+  NATS_MFMA_KLOOP(acc, A, m0, ldK, Bt, n0, Kpad, kbeg, kend);
   const int lane = threadIdx.x & (NATS_WAVE - 1);
   const int col = n0 + (lane & 15);
   const int rbase = m0 + (lane >> 4) * 4;
   if (col >= N) return;
+  float* C = Cpart + (long)blockIdx.z * 32 * N;
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     const int row = rbase + i;
@@ -53,10 +62,14 @@ __global__ void cond_small_gemm_bt(const bf16_t* __restrict__ A,
 }
 
 // ---------------- attention e-scores (s-parallel) ----------------
-// grid (B, ceil(Ts/256)); thread = one source position s.
+// grid (B, ceil(Ts/256), ACH): thread = one source position s, grid.z
+// chunks the A (attention-dim) loop; chunks atomicAdd into e_buf, which
+// arrives ZEROED (cond_attn_softmax re-zeroes it after its last read, so
+// no per-step memset). pstate arrives as [KS][32][A] split-K partials.
 __global__ __launch_bounds__(256) void cond_attn_escore(
     const float* __restrict__ pctx,      // [Ts][B][A]
-    const float* __restrict__ pstate_t,  // [B][A]
+    const float* __restrict__ ps_part,   // [KS][32][A] pstate partials
+    int PS_KS,
     const float* __restrict__ accA,      // [B][Ts] (pre-update)
     const float* __restrict__ Dwei,      // [A]
     const float* __restrict__ Uatt,      // [A]
@@ -66,22 +79,24 @@ __global__ __launch_bounds__(256) void cond_attn_escore(
   const int b = blockIdx.x;
   const int s = blockIdx.y * blockDim.x + threadIdx.x;
   if (s >= Ts) return;
+  const int ACH = gridDim.z;
+  const int chunkA = ((A + ACH - 1) / ACH + 3) & ~3;
+  const int ibeg = blockIdx.z * chunkA;
+  const int iend = min(A, ibeg + chunkA);
+  if (ibeg >= iend) return;
   const float accAu = accA[(long)b * Ts + s];
   const float* prow = pctx + ((long)s * B + b) * A;
-  const float* srow = pstate_t + (long)b * A;
-  float e = catt_p[0];
-  int i = 0;
-  const int A4 = A & ~3;
-  for (; i < A4; i += 4) {
-    const float4 p = *(const float4*)(prow + i);
-    e += tanhf(p.x + srow[i] + accAu * Dwei[i]) * Uatt[i];
-    e += tanhf(p.y + srow[i + 1] + accAu * Dwei[i + 1]) * Uatt[i + 1];
-    e += tanhf(p.z + srow[i + 2] + accAu * Dwei[i + 2]) * Uatt[i + 2];
-    e += tanhf(p.w + srow[i + 3] + accAu * Dwei[i + 3]) * Uatt[i + 3];
+  float e = (blockIdx.z == 0) ? catt_p[0] : 0.f;
+  for (int i = ibeg; i < iend; ++i) {
+    float ps = 0.f;
+    for (int k = 0; k < PS_KS; ++k) ps += ps_part[((long)k * 32 + b) * A + i];
+    e += tanhf(prow[i] + ps + accAu * Dwei[i]) * Uatt[i];
   }
-  for (; i < A; ++i)
-    e += tanhf(prow[i] + srow[i] + accAu * Dwei[i]) * Uatt[i];
-  e_buf[(long)s * B + b] = e;
+  if (ACH == 1 && PS_KS >= 0) {
+    e_buf[(long)s * B + b] += e;  // single chunk: still additive (zeroed)
+  } else {
+    atomicAdd(e_buf + (long)s * B + b, e);
+  }
 }
 
 // ---------------- softmax finish + acc_alpha update (one WG per b) ----
@@ -90,9 +105,22 @@ __global__ __launch_bounds__(256) void cond_attn_softmax(
     float* __restrict__ accA_used_t,     // [B][Ts] out (pre-update copy)
     const float* __restrict__ ctx_mask,  // [Ts][B] or null
     const float* __restrict__ mask_t,    // [B] or null
-    const float* __restrict__ e_buf,     // [Ts][B]
+    float* __restrict__ e_buf,           // [Ts][B]; re-zeroed for t+1
     float* __restrict__ alphas_t,        // [B][Ts] out
+    const float* __restrict__ ps_part,   // [KS][32][A] pstate partials
+    int PS_KS,
+    float* __restrict__ pstate_t,        // [B][A] combined (for backward)
+    int A,
     int B, int Ts) {
+  { // combine the split-K pstate partials once (backward reads pstate_all)
+    const int b0 = blockIdx.x;
+    for (int i = threadIdx.x; i < A; i += blockDim.x) {
+      float ps = 0.f;
+      for (int k = 0; k < PS_KS; ++k)
+        ps += ps_part[((long)k * 32 + b0) * A + i];
+      pstate_t[(long)b0 * A + i] = ps;
+    }
+  }
   const int b = blockIdx.x;
   __shared__ float red[256 / NATS_WAVE];
   __shared__ float bcast;
@@ -118,6 +146,7 @@ __global__ __launch_bounds__(256) void cond_attn_softmax(
   float lsum = 0.f;
   for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
     float a = __expf(e_buf[(long)s * B + b] - M);
+    e_buf[(long)s * B + b] = 0.f;  // ready for the next step's atomics
     if (ctx_mask != nullptr) a *= ctx_mask[(long)s * B + b];
     alphas_t[(long)b * Ts + s] = a;  // unnormalised, fixed below
     lsum += a;
@@ -742,7 +771,9 @@ std::vector<torch::Tensor> cond_gru_fwd(
   const int GRU2_KS = 4;
   auto gru2_part = torch::empty({GRU2_KS, 3, 32, Hpad}, optsF);
   auto hc_bf = torch::zeros({32, K1}, optsB);
-  auto e_buf = torch::empty({Ts, B}, optsF);
+  auto e_buf = torch::zeros({Ts, B}, optsF);  // escore accumulates into it
+  const int PS_KS = 4;
+  auto ps_part = torch::empty({PS_KS, 32, A}, optsF);
   auto ctxpre_f32 = torch::empty({B, C}, optsF);
   const int SCH = std::max(1, std::min(8, Ts / 64));
   auto init_f = init_state.contiguous().to(torch::kFloat32);
@@ -797,16 +828,17 @@ std::vector<torch::Tensor> cond_gru_fwd(
                          (bf16_t*)saved2.data_ptr() + (long)t * B * 3 * H, B,
                          H, Hpad);
     }
-    // 2) pstate = h1 @ W_att
-    hipLaunchKernelGGL(cond_small_gemm_bt, dim3(2, Apad16 / 16), dim3(64), 0,
-                       stream, (const bf16_t*)hc_bf.data_ptr(),
+    // 2) pstate = h1 @ W_att (split-K partials; combined in softmax)
+    hipLaunchKernelGGL(cond_small_gemm_bt, dim3(2, Apad16 / 16, PS_KS),
+                       dim3(64), 0, stream, (const bf16_t*)hc_bf.data_ptr(),
                        (const bf16_t*)WattPk.data_ptr(),
-                       pstate_all.data_ptr<float>() + (long)t * B * A, B, A,
-                       K1, (int)WattPk.size(1));
-    // 3) attention scores (s-parallel) + softmax finish + acc_alpha
-    hipLaunchKernelGGL(cond_attn_escore, dim3(B, cdiv_i(Ts, 256)), dim3(256),
+                       ps_part.data_ptr<float>(), B, A, K1,
+                       (int)WattPk.size(1));
+    // 3) attention scores (s- and A-parallel) + softmax + acc_alpha
+    hipLaunchKernelGGL(cond_attn_escore,
+                       dim3(B, cdiv_i(Ts, 256), A >= 16 ? 4 : 1), dim3(256),
                        0, stream, pctx.data_ptr<float>(),
-                       pstate_all.data_ptr<float>() + (long)t * B * A,
+                       ps_part.data_ptr<float>(), PS_KS,
                        accA.data_ptr<float>(), Dwei.data_ptr<float>(),
                        Uatt.data_ptr<float>(), catt.data_ptr<float>(),
                        e_buf.data_ptr<float>(), B, Ts, A);
@@ -814,7 +846,9 @@ std::vector<torch::Tensor> cond_gru_fwd(
                        accA.data_ptr<float>(),
                        accA_used.data_ptr<float>() + (long)t * B * Ts,
                        cmask_p, mt, e_buf.data_ptr<float>(),
-                       alphas_all.data_ptr<float>() + (long)t * B * Ts, B,
+                       alphas_all.data_ptr<float>() + (long)t * B * Ts,
+                       ps_part.data_ptr<float>(), PS_KS,
+                       pstate_all.data_ptr<float>() + (long)t * B * A, A, B,
                        Ts);
     // 4) weighted context (s-chunked partials) + gate + acc_ctx
     HIP_CHECK(hipMemsetAsync(ctxpre_f32.data_ptr<float>(), 0,
