@@ -66,7 +66,7 @@ __global__ void cond_small_gemm_bt(const bf16_t* __restrict__ A,
 // arrives ZEROED (cond_attn_softmax re-zeroes it after its last read, so
 // no per-step memset). pstate arrives as [KS][32][A] split-K partials.
 __global__ __launch_bounds__(256) void cond_attn_escore(
-    const float* __restrict__ pctxT,     // [B][A][Ts] (s-coalesced)
+    const float* __restrict__ pctx,      // [Ts][B][A]
     const float* __restrict__ ps_part,   // [KS][32][A] pstate partials
     int PS_KS,
     const float* __restrict__ accA,      // [B][Ts] (pre-update)
@@ -84,12 +84,12 @@ __global__ __launch_bounds__(256) void cond_attn_escore(
   const int iend = min(A, ibeg + chunkA);
   if (ibeg >= iend) return;
   const float accAu = accA[(long)b * Ts + s];
-  const float* pcol = pctxT + (long)b * A * Ts + s;
+  const float* prow = pctx + ((long)s * B + b) * A;
   float e = (blockIdx.z == 0) ? catt_p[0] : 0.f;
   for (int i = ibeg; i < iend; ++i) {
     float ps = 0.f;
     for (int k = 0; k < PS_KS; ++k) ps += ps_part[((long)k * 32 + b) * A + i];
-    e += tanhf(pcol[(long)i * Ts] + ps + accAu * Dwei[i]) * Uatt[i];
+    e += tanhf(prow[i] + ps + accAu * Dwei[i]) * Uatt[i];
   }
   if (ACH == 1 && PS_KS >= 0) {
     e_buf[(long)s * B + b] += e;  // single chunk: still additive (zeroed)
@@ -557,12 +557,12 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_scatter(
     const float* __restrict__ alphas_t,     // [B][Ts]
     const float* __restrict__ dal_buf,      // [Ts][B]
     const float* __restrict__ dot_buf,      // [B]
-    const float* __restrict__ pctxT,        // [B][A][Ts] (s-coalesced)
+    const float* __restrict__ pctx,         // [Ts][B][A]
     const float* __restrict__ pstate_t,     // [B][A]
     const float* __restrict__ accA_used_t,  // [B][Ts]
     const float* __restrict__ Dwei, const float* __restrict__ Uatt,
     float* __restrict__ daccA,              // [B][Ts] (+=, atomic)
-    float* __restrict__ dpctxT_acc,         // [B][A][Ts] (+=)
+    float* __restrict__ dpctx_acc,          // [Ts][B][A] (+=)
     bf16_t* __restrict__ pc_buf,            // [B][A][Tpad8]
     int B, int Ts, int A, int Tpad8) {
   // grid (B, s-chunks, A-chunks): the un-chunked variant ran only
@@ -579,14 +579,41 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_scatter(
   const float al = alphas_t[(long)b * Ts + s];
   const float de = al * (dal_buf[(long)s * B + b] - dot_buf[b]);
   const float accAu = accA_used_t[(long)b * Ts + s];
-  const float* pcol = pctxT + (long)b * A * Ts + s;
+  const float* prow = pctx + ((long)s * B + b) * A;
   const float* srow = pstate_t + (long)b * A;
-  float* dpcol = dpctxT_acc + (long)b * A * Ts + s;
+  float* dprow = dpctx_acc + ((long)s * B + b) * A;
   float daccA_add = 0.f;
-  for (int i = ibeg; i < iend; ++i) {
-    const float pc = tanhf(pcol[(long)i * Ts] + srow[i] + accAu * Dwei[i]);
+  const int A4 = (A % 4 == 0) ? iend : ibeg;  // f32x4 needs A % 4 == 0
+  int i = ibeg;
+  for (; i < A4; i += 4) {
+    const float4 p = *(const float4*)(prow + i);
+    const float4 st = *(const float4*)(srow + i);
+    const float4 dw = *(const float4*)(Dwei + i);
+    const float4 ua = *(const float4*)(Uatt + i);
+    float4 dpv = *(const float4*)(dprow + i);
+    const float pc0 = tanhf(p.x + st.x + accAu * dw.x);
+    const float pc1 = tanhf(p.y + st.y + accAu * dw.y);
+    const float pc2 = tanhf(p.z + st.z + accAu * dw.z);
+    const float pc3 = tanhf(p.w + st.w + accAu * dw.w);
+    const float d0 = de * (1.f - pc0 * pc0) * ua.x;
+    const float d1 = de * (1.f - pc1 * pc1) * ua.y;
+    const float d2 = de * (1.f - pc2 * pc2) * ua.z;
+    const float d3 = de * (1.f - pc3 * pc3) * ua.w;
+    dpv.x += d0;
+    dpv.y += d1;
+    dpv.z += d2;
+    dpv.w += d3;
+    *(float4*)(dprow + i) = dpv;
+    daccA_add += d0 * dw.x + d1 * dw.y + d2 * dw.z + d3 * dw.w;
+    pc_buf[((long)b * A + i) * Tpad8 + s] = (bf16_t)pc0;
+    pc_buf[((long)b * A + i + 1) * Tpad8 + s] = (bf16_t)pc1;
+    pc_buf[((long)b * A + i + 2) * Tpad8 + s] = (bf16_t)pc2;
+    pc_buf[((long)b * A + i + 3) * Tpad8 + s] = (bf16_t)pc3;
+  }
+  for (; i < iend; ++i) {
+    const float pc = tanhf(prow[i] + srow[i] + accAu * Dwei[i]);
     const float dpc = de * (1.f - pc * pc) * Uatt[i];
-    dpcol[(long)i * Ts] += dpc;
+    dprow[i] += dpc;
     daccA_add += dpc * Dwei[i];
     pc_buf[((long)b * A + i) * Tpad8 + s] = (bf16_t)pc;
   }
@@ -746,10 +773,6 @@ std::vector<torch::Tensor> cond_gru_fwd(
   auto e_buf = torch::zeros({Ts, B}, optsF);  // escore accumulates into it
   const int PS_KS = 4;
   auto ps_part = torch::empty({PS_KS, 32, A}, optsF);
-  // s-coalesced attention keys: pctx is constant for the whole call, so
-  // one [Ts][B][A] -> [B][A][Ts] transpose turns every per-step escore
-  // read into a coalesced column access
-  auto pctxT = pctx.permute({1, 2, 0}).contiguous();
   auto ctxpre_f32 = torch::empty({B, C}, optsF);
   const int SCH = std::max(1, std::min(8, Ts / 64));
   auto init_f = init_state.contiguous().to(torch::kFloat32);
@@ -813,7 +836,7 @@ std::vector<torch::Tensor> cond_gru_fwd(
     // 3) attention scores (s- and A-parallel) + softmax + acc_alpha
     hipLaunchKernelGGL(cond_attn_escore,
                        dim3(B, cdiv_i(Ts, 256), A >= 16 ? 4 : 1), dim3(256),
-                       0, stream, pctxT.data_ptr<float>(),
+                       0, stream, pctx.data_ptr<float>(),
                        ps_part.data_ptr<float>(), PS_KS,
                        accA.data_ptr<float>(), Dwei.data_ptr<float>(),
                        Uatt.data_ptr<float>(), catt.data_ptr<float>(),
@@ -909,8 +932,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
   auto dctxpre_all = torch::empty({T, B, C}, optsB);
   auto dgate_all = torch::empty({T, B, C}, optsB);
   auto dpstate_all = torch::empty({T, B, A}, optsF);
-  auto pctxT = pctx.permute({1, 2, 0}).contiguous();
-  auto dpctxT_acc = torch::zeros({B, A, Ts}, optsF);
+  auto dpctx_acc = torch::zeros({Ts, B, A}, optsF);
   auto gdDwei = torch::zeros({A}, optsF);
   auto gdUatt = torch::zeros({A}, optsF);
   auto gdcatt = torch::zeros({1}, optsF);
@@ -1024,11 +1046,11 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        dim3(256), 0, stream,
                        alphas_all.data_ptr<float>() + (long)t * B * Ts,
                        dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
-                       pctxT.data_ptr<float>(),
+                       pctx.data_ptr<float>(),
                        pstate_all.data_ptr<float>() + (long)t * B * A,
                        accA_used.data_ptr<float>() + (long)t * B * Ts,
                        Dwei.data_ptr<float>(), Uatt.data_ptr<float>(),
-                       daccA.data_ptr<float>(), dpctxT_acc.data_ptr<float>(),
+                       daccA.data_ptr<float>(), dpctx_acc.data_ptr<float>(),
                        (bf16_t*)pc_buf.data_ptr(), B, Ts, A, Tpad8);
     HIP_CHECK(hipMemsetAsync(
         dpstate_all.data_ptr<float>() + (long)t * B * A, 0,
@@ -1074,6 +1096,5 @@ std::vector<torch::Tensor> cond_gru_bwd(
   }
   HIP_CHECK(hipGetLastError());
   return {dpre1_all, dpre2_all, dctxpre_all, dgate_all, dpstate_all,
-          dpctxT_acc.permute({2, 0, 1}).contiguous(), gdDwei, gdUatt,
-          gdcatt, dh_carry, daccC, daccA};
+          dpctx_acc, gdDwei, gdUatt, gdcatt, dh_carry, daccC, daccA};
 }
