@@ -1043,7 +1043,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
                        B, Ts);
     hipLaunchKernelGGL(cond_attn_bwd_scatter,
-                       dim3(B, cdiv_i(Ts, 256), A >= 16 ? 4 : 1),
+                       dim3(B, cdiv_i(Ts, 256), A >= 32 ? 8 : (A >= 16 ? 4 : 1)),
                        dim3(256), 0, stream,
                        alphas_all.data_ptr<float>() + (long)t * B * Ts,
                        dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
@@ -1056,7 +1056,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
     HIP_CHECK(hipMemsetAsync(
         dpstate_all.data_ptr<float>() + (long)t * B * A, 0,
         (size_t)B * A * sizeof(float), stream));
-    const int RSCH = std::max(1, std::min(4, Ts / 128));
+    const int RSCH = std::max(1, std::min(12, Ts / 64));
     hipLaunchKernelGGL(cond_attn_bwd_reduce, dim3(B, RSCH), dim3(256),
                        2 * ((Ts + RSCH - 1) / RSCH) * sizeof(float), stream,
                        alphas_all.data_ptr<float>() + (long)t * B * Ts,

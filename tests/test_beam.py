@@ -1,5 +1,4 @@
 import numpy
-import pytest
 import torch
 
 from nats_amd.decode.beam import (_cosine_dist, _kl_div,

@@ -2,7 +2,6 @@ import os
 import pickle
 
 import numpy
-import pytest
 
 from nats_amd.data.dictionary import (build_dictionary, dictionary_from_freqs,
                                       invert_dictionary, load_dictionary)

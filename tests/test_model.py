@@ -1,7 +1,6 @@
 import warnings
 
 import numpy
-import pytest
 import torch
 
 from nats_amd.data.prepare import prepare_data

@@ -6,7 +6,6 @@ over generated corpora, beyond the hand-picked cases in test_data.py.
 """
 
 import numpy
-import pytest
 from hypothesis import given, settings, strategies as st
 
 from nats_amd.data.dictionary import dictionary_from_freqs
