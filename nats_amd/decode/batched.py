@@ -31,12 +31,19 @@ def gen_sample_batched(model, xs, k=1, maxlen=30, use_unk=False,
     """Beam-decode a list of sources jointly.
 
     xs: list of (T_i, 1) int64 tensors (same device as model).
+    maxlen: an int, or a per-sentence list (serving: each request brings
+    its own cap — a sentence whose cap is reached flushes its live
+    hypotheses and drops out of the shared batch, gen_sample
+    semantics nats.py:1068-1074 per sentence).
     Returns a list of (sample, sample_score, sample_dec_alphas) tuples,
     one per sentence, each shaped exactly like gen_sample's beam output
     (alignment alphas are trimmed to the sentence's own length).
     """
     S = len(xs)
     assert S >= 1
+    maxlens = ([int(maxlen)] * S if isinstance(maxlen, int)
+               else [int(m) for m in maxlen])
+    assert len(maxlens) == S
     device = xs[0].device
     any_lambda = (kl_factor > 0.0 or ctx_factor > 0.0 or state_factor > 0.0)
 
@@ -75,7 +82,17 @@ def gen_sample_batched(model, xs, k=1, maxlen=30, use_unk=False,
 
     NEG_UNK = math.log(1e-20)
 
-    for ii in range(maxlen):
+    for ii in range(max(maxlens)):
+        # sentences at their own maxlen flush live hypotheses and leave
+        for i in range(S):
+            if st[i]["live"] > 0 and ii >= maxlens[i]:
+                for idx2 in range(st[i]["live"]):
+                    st[i]["out_samples"].append(st[i]["samples"][idx2])
+                    st[i]["out_scores"].append(float(st[i]["scores"][idx2]))
+                    st[i]["out_alphas"].append(st[i]["alphas"][idx2])
+                st[i].update(samples=[], alphas=[],
+                             scores=numpy.zeros(0, dtype="float32"))
+                st[i]["live"] = 0
         alive = [i for i in range(S) if st[i]["live"] > 0]
         if not alive:
             break

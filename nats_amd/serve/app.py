@@ -77,13 +77,16 @@ class SummarizerService:
 
     # ---- request side ----
 
-    def summarize(self, text, timeout=120.0):
+    def summarize(self, text, timeout=120.0, maxlen=None):
         """Decode one source text; blocks until its micro-batch is done.
 
-        Returns dict(summary, tokens, score, alignment).
+        maxlen overrides the service default for this request (requests
+        with different caps still share a batch — the batched decoder
+        takes per-sentence maxlens). Returns dict(summary, tokens,
+        score, alignment).
         """
         item = {"text": text, "event": threading.Event(), "result": None,
-                "error": None}
+                "error": None, "maxlen": int(maxlen or self.maxlen)}
         with self._cv:
             if self._closed:
                 raise RuntimeError("service is shut down")
@@ -95,10 +98,11 @@ class SummarizerService:
             raise item["error"]
         return item["result"]
 
-    def summarize_many(self, texts, timeout=300.0):
+    def summarize_many(self, texts, timeout=300.0, maxlen=None):
         """Enqueue several texts at once (they co-batch immediately)."""
+        ml = int(maxlen or self.maxlen)
         items = [{"text": t, "event": threading.Event(), "result": None,
-                  "error": None} for t in texts]
+                  "error": None, "maxlen": ml} for t in texts]
         with self._cv:
             if self._closed:
                 raise RuntimeError("service is shut down")
@@ -151,7 +155,8 @@ class SummarizerService:
                 return
             try:
                 t0 = time.perf_counter()
-                results = self._decode([it["text"] for it in batch])
+                results = self._decode([it["text"] for it in batch],
+                                       [it["maxlen"] for it in batch])
                 dt = time.perf_counter() - t0
                 with self._cv:
                     self._stats["requests"] += len(batch)
@@ -166,7 +171,7 @@ class SummarizerService:
                     it["error"] = e
                     it["event"].set()
 
-    def _decode(self, texts):
+    def _decode(self, texts, maxlens=None):
         srcs = [(t.strip().split() if not self.chr_level
                  else list(t.strip())) for t in texts]
         seqs = [map_line(t, self.word_dict, self.options["n_words"],
@@ -175,7 +180,8 @@ class SummarizerService:
                            device=self.device).reshape(-1, 1) for s in seqs]
         with torch.no_grad():
             outs = gen_sample_batched(
-                self.model, xs, k=self.k, maxlen=self.maxlen, use_unk=True,
+                self.model, xs, k=self.k,
+                maxlen=maxlens if maxlens else self.maxlen, use_unk=True,
                 kl_factor=self.kl_factor, ctx_factor=self.ctx_factor,
                 state_factor=self.state_factor)
         results = []
@@ -214,6 +220,7 @@ def create_app(service):
     class SummarizeRequest(BaseModel):
         text: str = None
         texts: list = None
+        maxlen: int = None
 
     app = FastAPI(title="nats_amd summarizer",
                   description="Distraction-based neural summarization "
@@ -237,8 +244,9 @@ def create_app(service):
                                 detail="provide 'text' or 'texts'")
         try:
             if req.text is not None:
-                return service.summarize(req.text)
-            return {"results": service.summarize_many(req.texts)}
+                return service.summarize(req.text, maxlen=req.maxlen)
+            return {"results": service.summarize_many(req.texts,
+                                                      maxlen=req.maxlen)}
         except TimeoutError as e:
             raise HTTPException(status_code=504, detail=str(e))
 

@@ -158,3 +158,25 @@ def test_serve_on_gpu(tmp_path):
         assert torch.cuda.is_available()
     finally:
         svc.close()
+
+
+def test_per_request_maxlen(tiny_server):
+    svc, _ = tiny_server
+    client = TestClient(create_app(svc))
+    r = client.post("/summarize", json={"text": "a b c d e f g", "maxlen": 3})
+    assert r.status_code == 200
+    assert len(r.json()["tokens"]) <= 3
+
+
+def test_batched_per_sentence_maxlen(tiny_server):
+    """Mixed maxlens in one shared batch: each sentence respects its own."""
+    from nats_amd.decode.batched import gen_sample_batched
+    import torch
+    svc, _ = tiny_server
+    xs = [torch.tensor([5, 6, 7, 0]).reshape(-1, 1),
+          torch.tensor([8, 9, 10, 11, 0]).reshape(-1, 1)]
+    outs = gen_sample_batched(svc.model, xs, k=3, maxlen=[2, 9], use_unk=True)
+    for (samples, scores, alphas), cap in zip(outs, [2, 9]):
+        assert len(samples) >= 1
+        assert all(len(s) <= cap for s in samples)
+        assert len(scores) == len(samples) == len(alphas)
