@@ -207,7 +207,7 @@ __global__ void cond_attn_ctx_partial(
 
 // ---------------- distraction gate + acc_ctx update ----------------
 __global__ void cond_attn_gate_fwd(
-    const float* __restrict__ ctxpre_f32,  // [B][C]
+    float* __restrict__ ctxpre_f32,  // [B][C]; re-zeroed for t+1's atomics
     const float* __restrict__ Ucon, const float* __restrict__ Wcon,
     float* __restrict__ accC,            // [B][C] in/out
     bf16_t* __restrict__ accC_used_t,    // [B][C]
@@ -219,6 +219,7 @@ __global__ void cond_attn_gate_fwd(
   const int c = blockIdx.y * blockDim.x + threadIdx.x;
   if (c >= C) return;
   const float sum = ctxpre_f32[(long)b * C + c];
+  ctxpre_f32[(long)b * C + c] = 0.f;  // next step's ctx_partial atomics
   ctxpre_t[(long)b * C + c] = (bf16_t)sum;
   const float accCv = accC[(long)b * C + c];
   accC_used_t[(long)b * C + c] = (bf16_t)accCv;
@@ -773,7 +774,7 @@ std::vector<torch::Tensor> cond_gru_fwd(
   auto e_buf = torch::zeros({Ts, B}, optsF);  // escore accumulates into it
   const int PS_KS = 4;
   auto ps_part = torch::empty({PS_KS, 32, A}, optsF);
-  auto ctxpre_f32 = torch::empty({B, C}, optsF);
+  auto ctxpre_f32 = torch::zeros({B, C}, optsF);  // gate_fwd re-zeroes
   const int SCH = std::max(1, std::min(8, Ts / 64));
   auto init_f = init_state.contiguous().to(torch::kFloat32);
   h2bf[0].slice(0, 0, B).slice(1, 0, H).copy_(init_f.to(torch::kBFloat16));
@@ -850,8 +851,7 @@ std::vector<torch::Tensor> cond_gru_fwd(
                        pstate_all.data_ptr<float>() + (long)t * B * A, A, B,
                        Ts);
     // 4) weighted context (s-chunked partials) + gate + acc_ctx
-    HIP_CHECK(hipMemsetAsync(ctxpre_f32.data_ptr<float>(), 0,
-                             (size_t)B * C * sizeof(float), stream));
+    // (ctxpre_f32 arrives zeroed: gate_fwd re-zeroes it after reading)
     hipLaunchKernelGGL(cond_attn_ctx_partial,
                        dim3(B, cdiv_i(C, 256), SCH), dim3(256), 0, stream,
                        (const bf16_t*)ctx_bf.data_ptr(),
@@ -931,7 +931,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
   auto dpre2_all = torch::empty({T, B, 4 * H}, optsB);
   auto dctxpre_all = torch::empty({T, B, C}, optsB);
   auto dgate_all = torch::empty({T, B, C}, optsB);
-  auto dpstate_all = torch::empty({T, B, A}, optsF);
+  auto dpstate_all = torch::zeros({T, B, A}, optsF);  // reduce atomic-adds
   auto dpctx_acc = torch::zeros({Ts, B, A}, optsF);
   auto gdDwei = torch::zeros({A}, optsF);
   auto gdUatt = torch::zeros({A}, optsF);
@@ -1052,9 +1052,6 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        Dwei.data_ptr<float>(), Uatt.data_ptr<float>(),
                        daccA.data_ptr<float>(), dpctx_acc.data_ptr<float>(),
                        (bf16_t*)pc_buf.data_ptr(), B, Ts, A, Tpad8);
-    HIP_CHECK(hipMemsetAsync(
-        dpstate_all.data_ptr<float>() + (long)t * B * A, 0,
-        (size_t)B * A * sizeof(float), stream));
     const int RSCH = std::max(1, std::min(12, Ts / 64));
     hipLaunchKernelGGL(cond_attn_bwd_reduce, dim3(B, RSCH), dim3(256),
                        2 * ((Ts + RSCH - 1) / RSCH) * sizeof(float), stream,
