@@ -409,7 +409,7 @@ __global__ void cond_gru1_step_pointwise(
 // ---------------- backward kernels ----------------
 
 __global__ void cond_gru1_bwd_pointwise(
-    float* __restrict__ dh_carry,         // [B][H]; re-zeroed (split-K b9)
+    const float* __restrict__ dh_carry,   // [B][H]
     const float* __restrict__ dh2_all_t,  // [B][H] or null
     const bf16_t* __restrict__ saved1_t,  // [B][4H]
     const float* __restrict__ h1_all_t,   // [B][H]
@@ -427,7 +427,6 @@ __global__ void cond_gru1_bwd_pointwise(
     const int b = idx / H;
     const int j = idx % H;
     float dh2 = dh_carry[idx];
-    dh_carry[idx] = 0.f;  // next step's split-K gemm atomics
     if (dh2_all_t != nullptr) dh2 += dh2_all_t[idx];
     const float r2 = (float)saved1_t[(long)b * 4 * H + j];
     const float u2 = (float)saved1_t[(long)b * 4 * H + H + j];
@@ -471,7 +470,7 @@ __global__ void cond_dctx_dir(const float* __restrict__ dctxs_t,
   }
 }
 
-__global__ void cond_gate_bwd(float* __restrict__ dctx_buf,  // re-zeroed
+__global__ void cond_gate_bwd(const float* __restrict__ dctx_buf,
                               const float* __restrict__ ctxs_t,  // gated val
                               const float* __restrict__ Ucon,
                               const float* __restrict__ Wcon,
@@ -486,7 +485,6 @@ __global__ void cond_gate_bwd(float* __restrict__ dctx_buf,  // re-zeroed
     const int c = idx % C;
     const float g = ctxs_t[idx];
     const float dg = dctx_buf[idx] * (1.f - g * g);
-    dctx_buf[idx] = 0.f;  // next step's split-K dual-gemm atomics
     const float dpre = dg * Ucon[c];
     dctxpre_f32[idx] = dpre;
     dctxpre_all_t[idx] = (bf16_t)dpre;
@@ -955,7 +953,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
   auto ddirect_h1 = torch::empty({B, H}, optsF);
   auto ddirect2 = torch::empty({B, H}, optsF);
   auto dctx_dir = torch::empty({B, C}, optsF);
-  auto dctx_buf = torch::zeros({B, C}, optsF);  // split-K atomics
+  auto dctx_buf = torch::empty({B, C}, optsF);
   auto dctxpre_f32 = torch::empty({B, C}, optsF);
   auto daccA = daccA_f.has_value() ? daccA_f->contiguous().to(torch::kFloat32)
                                    : torch::zeros({B, Ts}, optsF);
@@ -1021,8 +1019,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
     // b2+b3 fused launch: dh1 = ddirect_h1 + dstep1 @ [U_1|Ux_1]^T and
     // dctx = dctx_dir + dstepC @ [W_1|Wx_1]^T (independent problems)
     hipLaunchKernelGGL(nats_gru_step_bwd_gemm_dual,
-                       dim3(std::max(ngrpH, ngrpC), 2, 2), dim3(384), 0,
-                       stream,
+                       dim3(std::max(ngrpH, ngrpC), 2), dim3(384), 0, stream,
                        (const bf16_t*)dstep1.data_ptr(),
                        (const bf16_t*)U1cat.data_ptr(),
                        ddirect_h1.data_ptr<float>(),
@@ -1099,8 +1096,8 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        (bf16_t*)dpre2_all.data_ptr() + (long)t * B * 4 * H, B,
                        H);
     // b9: dh_{t-1} = [dpr1|dpu1|dpxl1] @ [U|Ux]^T + passthrough
-    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH, 1, 2), dim3(384),
-                       0, stream, (const bf16_t*)dstep2.data_ptr(),
+    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH), dim3(384), 0,
+                       stream, (const bf16_t*)dstep2.data_ptr(),
                        (const bf16_t*)U2cat.data_ptr(),
                        ddirect2.data_ptr<float>(), dh_carry.data_ptr<float>(),
                        B, H, K3Hpad);

@@ -17,17 +17,6 @@
 // (SURVEY §7 "hard parts" (ii)); weight gradients that factor over time
 // (dU_1, dW_1, dW_att, dWc_att, dU_con, ...) are computed as single
 // time-batched GEMMs in python from the per-step buffers saved here.
-//
-// Occupancy design: at dim 1000 an output-tile grid is only ngrpH=63
-// workgroups (a quarter of the 256 CUs), so every per-step GEMM splits
-// its K dimension across extra grid dimensions and the elementwise
-// reductions chunk their serial axis, with fp32 partials summed by the
-// NEXT kernel in the chain rather than an extra pass (gru1/gru2 split-K
-// -> pointwise combine; pstate split-K -> escore/softmax inline sums;
-// escore/scatter A-chunks -> e_buf/daccA atomics). Scratch that is
-// atomically accumulated each step (e_buf, ctxpre_f32) is re-zeroed by
-// its consumer in the same pass, so the steady-state loop launches no
-// memsets. Measured ladder in profiles/README.md.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -410,7 +399,7 @@ __global__ void cond_gru1_step_pointwise(
 // ---------------- backward kernels ----------------
 
 __global__ void cond_gru1_bwd_pointwise(
-    float* __restrict__ dh_carry,         // [B][H]; re-zeroed (split-K b9)
+    const float* __restrict__ dh_carry,   // [B][H]
     const float* __restrict__ dh2_all_t,  // [B][H] or null
     const bf16_t* __restrict__ saved1_t,  // [B][4H]
     const float* __restrict__ h1_all_t,   // [B][H]
@@ -428,7 +417,6 @@ __global__ void cond_gru1_bwd_pointwise(
     const int b = idx / H;
     const int j = idx % H;
     float dh2 = dh_carry[idx];
-    dh_carry[idx] = 0.f;  // next step's split-K gemm atomics
     if (dh2_all_t != nullptr) dh2 += dh2_all_t[idx];
     const float r2 = (float)saved1_t[(long)b * 4 * H + j];
     const float u2 = (float)saved1_t[(long)b * 4 * H + H + j];
@@ -472,7 +460,7 @@ __global__ void cond_dctx_dir(const float* __restrict__ dctxs_t,
   }
 }
 
-__global__ void cond_gate_bwd(float* __restrict__ dctx_buf,  // re-zeroed
+__global__ void cond_gate_bwd(const float* __restrict__ dctx_buf,
                               const float* __restrict__ ctxs_t,  // gated val
                               const float* __restrict__ Ucon,
                               const float* __restrict__ Wcon,
@@ -487,7 +475,6 @@ __global__ void cond_gate_bwd(float* __restrict__ dctx_buf,  // re-zeroed
     const int c = idx % C;
     const float g = ctxs_t[idx];
     const float dg = dctx_buf[idx] * (1.f - g * g);
-    dctx_buf[idx] = 0.f;  // next step's split-K dual-gemm atomics
     const float dpre = dg * Ucon[c];
     dctxpre_f32[idx] = dpre;
     dctxpre_all_t[idx] = (bf16_t)dpre;
@@ -956,7 +943,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
   auto ddirect_h1 = torch::empty({B, H}, optsF);
   auto ddirect2 = torch::empty({B, H}, optsF);
   auto dctx_dir = torch::empty({B, C}, optsF);
-  auto dctx_buf = torch::zeros({B, C}, optsF);  // split-K atomics
+  auto dctx_buf = torch::empty({B, C}, optsF);
   auto dctxpre_f32 = torch::empty({B, C}, optsF);
   auto daccA = daccA_f.has_value() ? daccA_f->contiguous().to(torch::kFloat32)
                                    : torch::zeros({B, Ts}, optsF);
@@ -1022,8 +1009,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
     // b2+b3 fused launch: dh1 = ddirect_h1 + dstep1 @ [U_1|Ux_1]^T and
     // dctx = dctx_dir + dstepC @ [W_1|Wx_1]^T (independent problems)
     hipLaunchKernelGGL(nats_gru_step_bwd_gemm_dual,
-                       dim3(std::max(ngrpH, ngrpC), 2, 2), dim3(384), 0,
-                       stream,
+                       dim3(std::max(ngrpH, ngrpC), 2), dim3(384), 0, stream,
                        (const bf16_t*)dstep1.data_ptr(),
                        (const bf16_t*)U1cat.data_ptr(),
                        ddirect_h1.data_ptr<float>(),
@@ -1100,8 +1086,8 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        (bf16_t*)dpre2_all.data_ptr() + (long)t * B * 4 * H, B,
                        H);
     // b9: dh_{t-1} = [dpr1|dpu1|dpxl1] @ [U|Ux]^T + passthrough
-    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH, 1, 2), dim3(384),
-                       0, stream, (const bf16_t*)dstep2.data_ptr(),
+    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH), dim3(384), 0,
+                       stream, (const bf16_t*)dstep2.data_ptr(),
                        (const bf16_t*)U2cat.data_ptr(),
                        ddirect2.data_ptr<float>(), dh_carry.data_ptr<float>(),
                        B, H, K3Hpad);

@@ -25,7 +25,7 @@ __global__ void nats_gru2_step_pointwise(
 
 // backward pointwise: dh -> gate preactivation grads. dh_out_t may be null.
 __global__ void nats_gru_step_bwd_pointwise(
-    float* __restrict__ dh_buf, const float* __restrict__ dh_out_t,
+    const float* __restrict__ dh_buf, const float* __restrict__ dh_out_t,
     const bf16_t* __restrict__ saved_t, const bf16_t* __restrict__ xc_t,
     const float* __restrict__ h_prev, const float* __restrict__ mask_t,
     bf16_t* __restrict__ dstep, int ld_dstep, float* __restrict__ ddirect,

@@ -171,8 +171,7 @@ __global__ void nats_gru2_step_pointwise(
 
 // ---------------- backward pointwise ----------------
 __global__ void nats_gru_step_bwd_pointwise(
-    float* __restrict__ dh_buf,   // [B][H] recurrent dh; re-zeroed after
-                                  // read (split-K gemm atomics refill it)
+    const float* __restrict__ dh_buf,   // [B][H] recurrent dh (from t+1)
     const float* __restrict__ dh_out_t, // [B][H] upstream grad (or null)
     const bf16_t* __restrict__ saved_t, // [B][3H] (r,u,pxl)
     const bf16_t* __restrict__ xc_t,    // [B][H]
@@ -189,7 +188,6 @@ __global__ void nats_gru_step_bwd_pointwise(
     const int b = idx / H;
     const int j = idx % H;
     float dh = dh_buf[idx];
-    dh_buf[idx] = 0.f;
     if (dh_out_t != nullptr) dh += dh_out_t[idx];
     const float r = (float)saved_t[(long)b * 3 * H + j];
     const float u = (float)saved_t[(long)b * 3 * H + H + j];
@@ -910,17 +908,9 @@ __device__ __forceinline__ void gru_bwd_gemm_body(const bf16_t* dstep,
   const int m = wave / 3;
   const int ks = wave % 3;
   const int i0 = wg * JB;
-  // optional split-K over grid.z: each z-block takes a K-slab (its 3
-  // wave-chunks inside), atomicAdds its partial into a PRE-ZEROED out,
-  // and z==0 also adds ddirect. gridDim.z==1 keeps the original
-  // write-with-ddirect (callers whose out is not zeroed).
-  const int KSZ = gridDim.z;
-  const int zslab = ((Kpad / KSZ + 31) / 32) * 32;
-  const int zbeg = min(Kpad, (int)blockIdx.z * zslab);
-  const int zend = min(Kpad, zbeg + zslab);
-  const int kchunk = (((zend - zbeg) / 3 + 31) / 32) * 32;
-  const int kbeg = min(zend, zbeg + ks * kchunk);
-  const int kend = min(zend, zbeg + (ks + 1) * kchunk);
+  const int kchunk = ((Kpad / 3 + 31) / 32) * 32;
+  const int kbeg = ks * kchunk;
+  const int kend = min(Kpad, (ks + 1) * kchunk);
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   NATS_MFMA_KLOOP(acc, dstep, 16 * m, Kpad, Wt, i0, Kpad, kbeg, kend);
   {
@@ -936,13 +926,8 @@ __device__ __forceinline__ void gru_bwd_gemm_body(const bf16_t* dstep,
     const int c = idx % JB;
     const int i = i0 + c;
     if (i >= H) continue;
-    const float p = part[0][b][c] + part[1][b][c] + part[2][b][c];
-    if (KSZ == 1) {
-      out[(long)b * H + i] = ddirect[(long)b * H + i] + p;
-    } else {
-      atomicAdd(out + (long)b * H + i,
-                p + (blockIdx.z == 0 ? ddirect[(long)b * H + i] : 0.f));
-    }
+    out[(long)b * H + i] = ddirect[(long)b * H + i] + part[0][b][c] +
+                           part[1][b][c] + part[2][b][c];
   }
 }
 
