@@ -17,6 +17,17 @@
 // (SURVEY §7 "hard parts" (ii)); weight gradients that factor over time
 // (dU_1, dW_1, dW_att, dWc_att, dU_con, ...) are computed as single
 // time-batched GEMMs in python from the per-step buffers saved here.
+//
+// Occupancy design: at dim 1000 an output-tile grid is only ngrpH=63
+// workgroups (a quarter of the 256 CUs), so every per-step GEMM splits
+// its K dimension across extra grid dimensions and the elementwise
+// reductions chunk their serial axis, with fp32 partials summed by the
+// NEXT kernel in the chain rather than an extra pass (gru1/gru2 split-K
+// -> pointwise combine; pstate split-K -> escore/softmax inline sums;
+// escore/scatter A-chunks -> e_buf/daccA atomics). Scratch that is
+// atomically accumulated each step (e_buf, ctxpre_f32) is re-zeroed by
+// its consumer in the same pass, so the steady-state loop launches no
+// memsets. Measured ladder in profiles/README.md.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
