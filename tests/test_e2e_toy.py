@@ -111,3 +111,27 @@ def test_decode_chain(toy_corpus, tmp_path):
     replace_unk_files(src, temp, final)
     r, p, f = score_files(ref, final, 1, "N")
     assert 0.0 <= r <= 1.0 and 0.0 <= f <= 1.0
+
+
+def test_train_profile_and_resume_optimizer(toy_corpus, tmp_path):
+    """profile=True collects per-section timers; resume_optimizer=True
+    writes the .opt.npz sidecar at saveFreq and restores it on reload."""
+    kw = dict(dim_word=10, dim=12, dim_att=6, n_words=64, maxlen=50,
+              batch_size=8, valid_batch_size=8,
+              datasets=[os.path.join(toy_corpus, "toy_train_input.txt"),
+                        os.path.join(toy_corpus, "toy_train_output.txt")],
+              valid_datasets=[
+                  os.path.join(toy_corpus, "toy_validation_input.txt"),
+                  os.path.join(toy_corpus, "toy_validation_output.txt")],
+              dictionary=os.path.join(toy_corpus, "toy_train_input.txt.pkl"),
+              validFreq=100, saveFreq=3, sampleFreq=1000, dispFreq=2,
+              finish_after=6, device="cpu", seed=7,
+              profile=True, resume_optimizer=True)
+    saveto = str(tmp_path / "model.npz")
+    train(saveto=saveto, **kw)
+    assert os.path.exists(saveto + ".opt.npz")
+    # restore must not blow up and must pick up the adadelta accumulators
+    archive = numpy.load(saveto + ".opt.npz")
+    assert any(k.endswith("::rg2") for k in archive.files)
+    err = train(saveto=saveto, reload_=True, **kw)
+    assert numpy.isfinite(err)
