@@ -135,3 +135,29 @@ def test_train_profile_and_resume_optimizer(toy_corpus, tmp_path):
     assert any(k.endswith("::rg2") for k in archive.files)
     err = train(saveto=saveto, reload_=True, **kw)
     assert numpy.isfinite(err)
+
+
+def test_gen_multiprocess_matches_single(toy_corpus, tmp_path):
+    """gen.py's worker-pool path (gen.py:111-126): 2 spawned workers must
+    produce byte-identical output to the single-process path."""
+    from nats_amd.decode.driver import generate_file
+    from nats_amd.engine.checkpoint import save_checkpoint
+
+    opts = default_options(dim_word=10, dim=12, dim_att=6, n_words=43,
+                           maxlen=50)
+    model = NatsModel(opts, seed=13)
+    saveto = str(tmp_path / "model.npz")
+    save_checkpoint(saveto, model.get_params(), [], options=opts)
+    src = str(tmp_path / "src.txt")
+    with open(os.path.join(toy_corpus, "toy_test_input.txt")) as f:
+        lines = f.read().splitlines()[:6]
+    with open(src, "w") as f:
+        f.write("\n".join(lines) + "\n")
+    dic = os.path.join(toy_corpus, "toy_train_input.txt.pkl")
+    out1 = str(tmp_path / "out1.txt")
+    out2 = str(tmp_path / "out2.txt")
+    generate_file(saveto, dic, src, out1, k=3, normalize=True, n_process=1,
+                  verbose=False, maxlen=12, devices=["cpu"])
+    generate_file(saveto, dic, src, out2, k=3, normalize=True, n_process=2,
+                  verbose=False, maxlen=12, devices=["cpu"])
+    assert open(out1).read() == open(out2).read()
