@@ -86,3 +86,73 @@ def test_distraction_rerank_changes_selection_not_costs(tiny_options):
     # all returned costs are sums of -log p along the path: positive
     for c in c0 + c1:
         assert c > 0
+
+
+def _oracle_beam(model, x, k, maxlen, use_unk):
+    """Independent re-derivation of the reference beam loop (nats.py:
+    940-1074) written against f_init/f_next directly — no shared code
+    with decode.beam. Used to cross-check gen_sample's bookkeeping."""
+    import torch
+    init, ctx0 = model.f_init(x)
+    pctx0 = model.project_ctx(ctx0)
+    Ts, _, C = ctx0.shape
+    hyps = [dict(toks=[], cost=0.0, state=init[0],
+                 accC=torch.zeros(C), accA=torch.zeros(Ts))]
+    done = []
+    for step in range(maxlen):
+        if not hyps or len(done) >= k:
+            break
+        B = len(hyps)
+        y = torch.tensor([h["toks"][-1] if h["toks"] else -1 for h in hyps])
+        state = torch.stack([h["state"] for h in hyps])
+        accC = torch.stack([h["accC"] for h in hyps])
+        accA = torch.stack([h["accA"] for h in hyps])
+        probs, _, h2, alpha, ctx_t, accC2, accA2 = model.f_next(
+            y, ctx0.expand(Ts, B, C), None,
+            pctx0.expand(Ts, B, pctx0.shape[2]), state, accC, accA,
+            sample_draw=False)
+        p = probs.double().numpy()
+        if not use_unk:
+            p[:, 1] = 1e-20
+        import numpy as np
+        cand = np.array([h["cost"] for h in hyps])[:, None] - np.log(p)
+        flat = cand.flatten()
+        order = flat.argsort()[:k - len(done)]
+        new = []
+        for r in order:
+            ti, wi = int(r) // p.shape[1], int(r) % p.shape[1]
+            h = dict(toks=hyps[ti]["toks"] + [wi], cost=float(flat[r]),
+                     state=h2[ti], accC=accC2[ti], accA=accA2[ti])
+            if wi == 0:
+                done.append(h)
+            else:
+                new.append(h)
+        hyps = new
+    done += hyps  # flush still-live at maxlen
+    return (sorted(tuple(h["toks"]) for h in done),
+            sorted(round(h["cost"], 4) for h in done))
+
+
+def test_gen_sample_matches_independent_oracle(tiny_options):
+    """Cross-check the production beam (decode/beam.py) against a from-
+    scratch oracle on a model with well-separated probabilities."""
+    import torch
+    from nats_amd.models.distraction import NatsModel
+    from nats_amd.decode.beam import gen_sample
+    model = NatsModel(tiny_options, seed=21).eval()
+    with torch.no_grad():
+        for key in ("ff_logit_lstm_W", "ff_logit_ctx_W", "ff_logit_prev_W",
+                    "ff_logit_W"):
+            model.P[key].mul_(50.0)
+    g = torch.Generator().manual_seed(2)
+    for trial in range(3):
+        x = torch.randint(2, tiny_options["n_words"], (6 + trial, 1),
+                          generator=g)
+        x[-1] = 0
+        with torch.no_grad():
+            s, c, a = gen_sample(model, x, k=4, maxlen=7, stochastic=False,
+                                 use_unk=True)
+            otoks, ocosts = _oracle_beam(model, x, k=4, maxlen=7,
+                                         use_unk=True)
+        assert sorted(map(tuple, s)) == otoks
+        assert sorted(round(v, 4) for v in c) == ocosts
