@@ -161,3 +161,30 @@ def test_gen_multiprocess_matches_single(toy_corpus, tmp_path):
     generate_file(saveto, dic, src, out2, k=3, normalize=True, n_process=2,
                   verbose=False, maxlen=12, devices=["cpu"])
     assert open(out1).read() == open(out2).read()
+
+
+def test_bench_contract(tmp_path):
+    """The driver depends on bench.py's exact CLI + one-JSON-line output
+    contract; pin it on the CPU-capable toy config."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "bench.py"), "--gpus", "1",
+         "--steps", "2", "--warmup", "1", "--config", "toy"],
+        capture_output=True, text=True, timeout=300, cwd=repo)
+    assert out.returncode == 0, out.stderr[-800:]
+    lines = [l for l in out.stdout.splitlines() if l.strip().startswith("{")]
+    assert len(lines) == 1, out.stdout
+    r = json.loads(lines[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in r, key
+    assert r["metric"] == "train_tokens_per_sec"
+    assert r["n_gpus"] == 1 and r["steps"] == 2 and r["warmup"] == 1
+    assert r["data"] == "synthetic" and r["scaling"] == "weak"
+    assert r["value"] > 0 and numpy.isfinite(r["config"]["final_cost"])
+    assert r["config"]["global_batch"] == r["config"]["batch_per_gpu"]
