@@ -20,6 +20,9 @@ if __name__ == "__main__":
     parser.add_argument("-s", type=float, default=0)
     parser.add_argument("-n", action="store_true", default=False)
     parser.add_argument("-c", action="store_true", default=False)
+    parser.add_argument("--maxlen", type=int, default=100,
+                        help="summary length cap (framework extension; "
+                             "the reference hardcodes 100, gen.py:33)")
     parser.add_argument("model", type=str)
     parser.add_argument("dictionary", type=str)
     parser.add_argument("source", type=str)
@@ -29,4 +32,4 @@ if __name__ == "__main__":
     generate_file(args.model, args.dictionary, args.source, args.saveto,
                   k=args.k, normalize=args.n, n_process=args.p,
                   chr_level=args.c, kl_factor=args.l, ctx_factor=args.x,
-                  state_factor=args.s)
+                  state_factor=args.s, maxlen=args.maxlen)
