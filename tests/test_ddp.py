@@ -115,3 +115,29 @@ def test_ddp_ragged_batches_no_deadlock(tmp_path):
     mp.spawn(_ragged_worker, args=(2, port, str(tmp_path)), nprocs=2,
              join=True)
     assert os.path.exists(tmp_path / "ok_ragged")
+
+
+@pytest.mark.timeout(600)
+def test_bench_multirank_torchrun_contract(tmp_path):
+    """The round-end driver launches bench.py via torch.distributed.run
+    with N ranks; pin that exact invocation on CPU (gloo, toy config)."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.pop("RANK", None), env.pop("WORLD_SIZE", None)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29561", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--config", "toy"],
+        capture_output=True, text=True, timeout=540, cwd=repo, env=env)
+    assert out.returncode == 0, out.stderr[-800:]
+    lines = [l for l in out.stdout.splitlines() if l.strip().startswith("{")]
+    assert len(lines) == 1, out.stdout  # rank 0 only
+    r = json.loads(lines[0])
+    assert r["n_gpus"] == 2
+    assert r["config"]["parallelism"] == "dp2"
+    assert r["config"]["global_batch"] == 2 * r["config"]["batch_per_gpu"]
