@@ -188,3 +188,53 @@ def test_bench_contract(tmp_path):
     assert r["data"] == "synthetic" and r["scaling"] == "weak"
     assert r["value"] > 0 and numpy.isfinite(r["config"]["final_cost"])
     assert r["config"]["global_batch"] == r["config"]["batch_per_gpu"]
+
+
+def test_nan_failure_detection(toy_corpus, tmp_path):
+    """NaN/Inf training cost -> hard abort returning the
+    (1., 1., 1.) sentinel (nats.py:1415-1417); NaN validation cost ->
+    raise (the reference drops into ipdb, nats.py:1096)."""
+    import torch
+
+    from nats_amd.engine.validate import pred_probs
+
+    kw = dict(dim_word=10, dim=12, dim_att=6, n_words=64, maxlen=50,
+              batch_size=8, valid_batch_size=8,
+              datasets=[os.path.join(toy_corpus, "toy_train_input.txt"),
+                        os.path.join(toy_corpus, "toy_train_output.txt")],
+              valid_datasets=[
+                  os.path.join(toy_corpus, "toy_validation_input.txt"),
+                  os.path.join(toy_corpus, "toy_validation_output.txt")],
+              dictionary=os.path.join(toy_corpus, "toy_train_input.txt.pkl"),
+              validFreq=1000, saveFreq=1000, sampleFreq=1000, dispFreq=1000,
+              finish_after=2, device="cpu", seed=3,
+              saveto=str(tmp_path / "m.npz"))
+    # inject the NaN via poisoned initial embeddings (adadelta ignores
+    # lrate, so a divergence can't be provoked through the optimizer)
+    import nats_amd.models.distraction as D
+    orig = D.init_params
+
+    def poisoned(options, seed=None):
+        p = orig(options, seed=seed)
+        p["Wemb"] = p["Wemb"] * numpy.nan
+        return p
+
+    D.init_params = poisoned
+    try:
+        out = train(**kw)
+    finally:
+        D.init_params = orig
+    assert out == (1.0, 1.0, 1.0)
+
+    # validation NaN raises
+    opts = default_options(dim_word=10, dim=12, dim_att=6, n_words=64)
+    model = NatsModel(opts, seed=3)
+    with torch.no_grad():
+        model.P["Wemb"].mul_(float("nan"))
+    it = TextIterator(os.path.join(toy_corpus, "toy_validation_input.txt"),
+                      os.path.join(toy_corpus, "toy_validation_output.txt"),
+                      os.path.join(toy_corpus, "toy_train_input.txt.pkl"),
+                      batch_size=4, n_words=64)
+    import pytest as _pytest
+    with _pytest.raises(FloatingPointError):
+        pred_probs(model, it)
