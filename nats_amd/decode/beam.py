@@ -170,8 +170,7 @@ def gen_sample(model, x, k=1, maxlen=30, stochastic=True, argmax=False,
             probs, w_sample, next_state, dec_alphas, ctxs, acc_ctx, \
                 acc_alpha = model.f_next(
                     next_w, ctx, None, pctx, next_state, acc_ctx, acc_alpha,
-                    generator=generator,
-                    sample_draw=stochastic and not argmax)
+                    generator=generator, sample_draw=stochastic)
 
         if stochastic:
             if argmax:
@@ -182,7 +181,10 @@ def gen_sample(model, x, k=1, maxlen=30, stochastic=True, argmax=False,
             sample_score += float(probs[0, nw])
             if nw == 0:
                 break
-            next_w = torch.tensor([nw], dtype=torch.int64, device=device)
+            # The reference feeds the SAMPLED word back into f_next
+            # unconditionally (next_w = ret[1], nats.py:961); argmax only
+            # selects which word is RECORDED (nats.py:965-966).
+            next_w = w_sample[:1].to(torch.int64).reshape(1)
             continue
 
         next_p = probs.float().cpu().numpy()

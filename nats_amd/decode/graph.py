@@ -25,8 +25,10 @@ def _weights_version(model):
     """Monotone tag over the decoder weights: in-place optimizer updates
     bump tensor _version, which must invalidate captured graphs (packed
     weights are baked into the graph at capture — ops/cond_gru.py
-    _step_packed)."""
-    return tuple(p._version for p in model.P.values())
+    _step_packed). The fused adadelta kernel updates through raw pointers
+    instead, so it bumps _nats_update_epoch (ops/optim.py) — include both."""
+    return tuple((p._version, getattr(p, "_nats_update_epoch", 0))
+                 for p in model.P.values())
 
 
 def get_stepper(model, ctx0, pctx0, k):

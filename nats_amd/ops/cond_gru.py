@@ -195,7 +195,8 @@ def _step_packed(P):
     decode runs hundreds of steps against frozen weights, and re-packing
     (zero-fill + scatter + transpose + bf16 cast of every decoder matrix)
     per step dominated the non-graph beam loop."""
-    vers = tuple(P[k]._version for k in _STEP_PACK_KEYS)
+    vers = tuple((P[k]._version, getattr(P[k], "_nats_update_epoch", 0))
+                 for k in _STEP_PACK_KEYS)
     cached = getattr(P, "_nats_step_pack", None)
     if cached is not None and cached[0] == vers:
         return cached[1]

@@ -55,6 +55,13 @@ class FusedAdadelta(Adadelta):
             dtype=torch.int64, device=device)
         ext.fused_adadelta_step(ptrs, sizes_t, chunk_tensor, chunk_off, g2,
                                 float(self.clip_c), self.rho, self.eps)
+        # The fused kernel writes parameters through raw pointers, which
+        # never bumps tensor._version — the invalidation key used by the
+        # packed-weight cache (ops/cond_gru.py _step_packed) and the
+        # hipGraph stepper cache (decode/graph.py). Bump an explicit
+        # update epoch on each parameter so those caches see the change.
+        for p in ps:
+            p._nats_update_epoch = getattr(p, "_nats_update_epoch", 0) + 1
         return torch.sqrt(g2[0])  # pre-clip norm, no host sync
 
     @torch.no_grad()

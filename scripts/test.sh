@@ -15,6 +15,10 @@ TEMP=${TEMP:-./temp.txt}
 GEN=${GEN:-./final.txt}
 REF=${REF:-$ROOT/data/toy_test_output.txt}
 
+# bootstrap the toy corpus on a fresh clone (deterministic, seed=1234)
+python -c "import sys; sys.path.insert(0, 'scripts'); \
+  from train_nats import ensure_toy_corpus; ensure_toy_corpus()"
+
 # generate summaries (beam k=5, 10 workers, length-normalized)
 python scripts/gen.py -n -p 10 -k 5 -l ${KL} -x ${CTX} -s ${STATE} $MODEL $DIC $INPUT $TEMP
 

@@ -20,6 +20,18 @@ DATA = os.environ.get(
 MODELS = os.environ.get("NATS_MODEL_DIR", os.path.join(DATA, "..", "models"))
 
 
+def ensure_toy_corpus(data_dir=DATA):
+    """Generate the deterministic toy corpus + dictionary on first run so
+    the entrypoints work on a fresh clone (the reference ships its toy
+    files in-repo, train_nats.py:22-26; ours are regenerated, seed=1234)."""
+    if not os.path.exists(os.path.join(data_dir, "toy_train_input.txt")) or \
+            not os.path.exists(os.path.join(data_dir,
+                                            "toy_train_input.txt.pkl")):
+        from nats_amd.data.synthetic import make_toy_corpus
+        print("Bootstrapping toy corpus into %s" % data_dir)
+        make_toy_corpus(data_dir)
+
+
 def main(job_id, params):
     print(params)
     validerr = train(
@@ -52,6 +64,7 @@ def main(job_id, params):
 
 if __name__ == "__main__":
     os.makedirs(MODELS, exist_ok=True)
+    ensure_toy_corpus()
     main(0, {
         "model": [os.path.join(MODELS, "model.npz")],
         "dim_word": [120],
