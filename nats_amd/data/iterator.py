@@ -16,25 +16,24 @@ import gzip
 from .dictionary import load_dictionary
 
 
-def fopen(filename, mode="r"):
-    if filename.endswith(".gz"):
-        return gzip.open(filename, mode + "t")
-    return open(filename, mode)
+def _open_text(path):
+    if path.endswith(".gz"):
+        return gzip.open(path, "rt")
+    return open(path, "r")
 
 
 class TextIterator:
-    """Simple bitext iterator (API-compatible rebuild of data_iterator.py)."""
+    """Simple bitext iterator (behavioural rebuild of data_iterator.py)."""
 
     def __init__(self, source, target, dict, batch_size=128, n_words=-1):
-        self.source = fopen(source, "r")
-        self.target = fopen(target, "r")
-        if isinstance(dict, str):
-            self.dict = load_dictionary(dict)
-        else:
-            self.dict = dict
+        self.source = _open_text(source)
+        self.target = _open_text(target)
+        self.dict = load_dictionary(dict) if isinstance(dict, str) else dict
         self.batch_size = batch_size
         self.n_words = n_words
-        self.end_of_data = False
+        # Set when EOF was reached while a partial batch was still
+        # delivered; the following __next__ ends the epoch.
+        self._exhausted = False
 
     def __iter__(self):
         return self
@@ -43,42 +42,42 @@ class TextIterator:
         self.source.seek(0)
         self.target.seek(0)
 
-    def _map_line(self, line):
-        toks = line.strip().split()
-        ids = [self.dict[w] if w in self.dict else 1 for w in toks]
+    def _encode(self, line):
+        """Token line -> id list: dictionary lookup, UNK=1, n_words cap."""
+        ids = [self.dict.get(tok, 1) for tok in line.split()]
         if self.n_words > 0:
-            ids = [w if w < self.n_words else 1 for w in ids]
+            ids = [tok_id if tok_id < self.n_words else 1 for tok_id in ids]
         return ids
 
+    def _read_pair(self):
+        """Next (source, target) id-list pair, or None at end of either
+        file (a ragged tail line in one file is dropped, as in the
+        reference's paired readline loop)."""
+        src_line = self.source.readline()
+        tgt_line = self.target.readline()
+        if src_line == "" or tgt_line == "":
+            return None
+        return self._encode(src_line), self._encode(tgt_line)
+
     def __next__(self):
-        if self.end_of_data:
-            self.end_of_data = False
+        def _end_epoch():
+            self._exhausted = False
             self.reset()
             raise StopIteration
 
-        source = []
-        target = []
-        try:
-            while True:
-                ss = self.source.readline()
-                if ss == "":
-                    raise IOError
-                tt = self.target.readline()
-                if tt == "":
-                    raise IOError
-                source.append(self._map_line(ss))
-                target.append(self._map_line(tt))
-                if len(source) >= self.batch_size or len(target) >= self.batch_size:
-                    break
-        except IOError:
-            self.end_of_data = True
-
-        if len(source) <= 0 or len(target) <= 0:
-            self.end_of_data = False
-            self.reset()
-            raise StopIteration
-
-        return source, target
+        if self._exhausted:
+            _end_epoch()
+        batch_src, batch_tgt = [], []
+        while len(batch_src) < self.batch_size:
+            pair = self._read_pair()
+            if pair is None:
+                self._exhausted = True
+                break
+            batch_src.append(pair[0])
+            batch_tgt.append(pair[1])
+        if not batch_src:
+            _end_epoch()
+        return batch_src, batch_tgt
 
     # py2-style alias kept for API parity with the reference
     next = __next__
