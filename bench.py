@@ -56,6 +56,8 @@ def main():
                     help="per-GPU batch override")
     ap.add_argument("--eager", action="store_true",
                     help="force the eager (no HIP kernels) path")
+    ap.add_argument("--bucket-mb", type=int, default=25,
+                    help="DP gradient all-reduce bucket size (MiB)")
     args = ap.parse_args()
 
     if args.eager:
@@ -72,7 +74,8 @@ def main():
 
     rank, local_rank, world = init_distributed()
     if use_cuda:
-        device = torch.device("cuda", local_rank)
+        device = torch.device("cuda",
+                              local_rank % max(1, torch.cuda.device_count()))
         torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")
@@ -83,7 +86,7 @@ def main():
         clip_c=100.0, maxlen=cfg["src"] + 1,
         enc_depth=cfg.get("enc_depth", 1))
     model = NatsModel(opts, seed=1234).to(device)
-    dp = DataParallelGrads(model.parameters())
+    dp = DataParallelGrads(model.parameters(), bucket_cap_mb=args.bucket_mb)
     dp.broadcast_params()
     opt = build_optimizer("adadelta", list(model.P.items()),
                           lrate=1e-4, clip_c=100.0)

@@ -88,8 +88,9 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
     if device is None:
         device = "cuda" if torch.cuda.is_available() else "cpu"
     rank, local_rank, world = init_distributed()
-    if device == "cuda":
-        device = "cuda:%d" % local_rank if world > 1 else "cuda"
+    if device == "cuda" and world > 1:
+        device = "cuda:%d" % (local_rank %
+                              max(1, torch.cuda.device_count()))
 
     # load dictionary and invert (nats.py:1264-1268)
     worddicts = load_dictionary(dictionary)

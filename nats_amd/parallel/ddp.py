@@ -45,7 +45,11 @@ def init_distributed(backend=None, device=None):
         os.environ.setdefault("MASTER_PORT", "29517")
         dist.init_process_group(backend=backend, rank=rank, world_size=world)
     if backend == "nccl":
-        torch.cuda.set_device(local_rank if device is None else device)
+        if device is None:
+            # modulo mapping: identity on a full node (one rank per GPU);
+            # lets >1 rank share one device for single-GPU RCCL testing
+            device = local_rank % max(1, torch.cuda.device_count())
+        torch.cuda.set_device(device)
     return rank, local_rank, world
 
 
@@ -96,6 +100,14 @@ class DataParallelGrads:
         for p in self.params:
             self._hooks.append(
                 p.register_post_accumulate_grad_hook(self._on_grad_ready))
+
+    def detach(self):
+        """Remove the autograd hooks (e.g. before building a replacement
+        DataParallelGrads over the same parameters — two live instances
+        would each all-reduce every gradient)."""
+        for h in getattr(self, "_hooks", []):
+            h.remove()
+        self._hooks = []
 
     def broadcast_params(self):
         """Broadcast initial parameters from rank 0 (SURVEY §2.4 note)."""

@@ -1,0 +1,205 @@
+#!/usr/bin/env python3
+"""RCCL data-parallel validation probe (VERDICT r1 missing #1).
+
+Run under torchrun with N ranks (they may share one GPU — the device
+mapping is local_rank % device_count), nccl backend (= RCCL on ROCm):
+
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 \
+        --master-addr 127.0.0.1 --master-port 29533 scripts/rccl_probe.py
+
+Proves, over the REAL RCCL stack:
+  * process-group init + broadcast_params from rank 0,
+  * bucketed async gradient all-reduce (DataParallelGrads) produces
+    exactly the mean-of-rank-grads (checked against a single-process
+    two-shard reference computed independently on every rank),
+  * all_agree (MIN-reduce) both ways, all_reduce_scalar (validation path),
+  * a bucket_cap sweep (8/25/100 MiB) timing full train steps at the
+    CNN/DM model shape.
+
+Rank 0 writes gpurun_out/rccl_dp2.json.
+"""
+
+import json
+import os
+import sys
+import time
+
+import numpy
+import torch
+import torch.distributed as dist
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from nats_amd.data.synthetic import synthetic_batch
+from nats_amd.engine.optim import build_optimizer
+from nats_amd.models.distraction import NatsModel, default_options
+from nats_amd.parallel.ddp import DataParallelGrads, init_distributed
+
+
+def main():
+    rank, local_rank, world = init_distributed()
+    assert world > 1, "run under torchrun with --nproc-per-node >= 2"
+    use_cuda = torch.cuda.is_available()
+    device = (torch.device("cuda", torch.cuda.current_device())
+              if use_cuda else torch.device("cpu"))
+    backend = dist.get_backend()
+    try:
+        nccl_ver = ".".join(str(v) for v in torch.cuda.nccl.version())
+    except Exception:
+        nccl_ver = "unknown"
+    report = {
+        "world": world,
+        "backend": str(backend),
+        "rccl_version": nccl_ver,
+        "device_count": torch.cuda.device_count() if use_cuda else 0,
+        "device": str(device),
+        "torch": torch.__version__,
+    }
+
+    # ---- model at the headline CNN/DM shape (tiny on CPU smoke) ----
+    fast = bool(os.environ.get("NATS_PROBE_FAST"))
+    if use_cuda and not fast:
+        cfg = dict(src=800, tgt=100, n_words=30000, dim=1000, dim_word=100,
+                   dim_att=100, batch=20)
+    elif use_cuda:
+        cfg = dict(src=120, tgt=30, n_words=4000, dim=500, dim_word=100,
+                   dim_att=100, batch=8)
+    else:
+        cfg = dict(src=16, tgt=8, n_words=100, dim=16, dim_word=8,
+                   dim_att=8, batch=4)
+    opts = default_options(dim=cfg["dim"], dim_word=cfg["dim_word"],
+                           dim_att=cfg["dim_att"], n_words=cfg["n_words"],
+                           batch_size=cfg["batch"], optimizer="adadelta",
+                           clip_c=100.0, maxlen=cfg["src"] + 1)
+    # rank-dependent seed then broadcast: proves broadcast really runs
+    model = NatsModel(opts, seed=100 + rank).to(device)
+    n_params = sum(p.numel() for p in model.parameters())
+    report["n_params"] = n_params
+
+    dp = DataParallelGrads(model.parameters(), bucket_cap_mb=25)
+    dp.broadcast_params()
+
+    # broadcast check: parameter checksum must now agree across ranks
+    with torch.no_grad():
+        csum = torch.stack([p.float().sum() for p in model.parameters()]).sum()
+    csums = [torch.zeros_like(csum) for _ in range(world)]
+    dist.all_gather(csums, csum)
+    bcast_spread = float(max(abs(c.item() - csums[0].item()) for c in csums))
+    report["broadcast_param_checksum_spread"] = bcast_spread
+    assert bcast_spread == 0.0, "broadcast_params left ranks diverged"
+
+    # ---- gradient exactness over real RCCL ----
+    # every rank computes BOTH shards' grads locally as the oracle, then
+    # the DP path computes its own shard's grads + bucketed all-reduce
+    import contextlib
+
+    def amp_ctx():
+        return (torch.autocast("cuda", dtype=torch.bfloat16) if use_cuda
+                else contextlib.nullcontext())
+
+    shards = []
+    for r in range(world):
+        rng = numpy.random.RandomState(500 + r)
+        shards.append([torch.from_numpy(a).to(device) for a in
+                       synthetic_batch(rng, 4, 40, 12, cfg["n_words"])])
+
+    # oracle: mean over shard means, single process (no collectives).
+    # Each shard gets its OWN backward pass with incoming grad 1.0 —
+    # exactly what each DP rank computes — then grads are divided by
+    # world, mirroring the all-reduce-sum + divide on the DP side.
+    oracle = NatsModel(opts, seed=0).to(device)
+    oracle.set_params({k: v.detach().cpu().numpy()
+                       for k, v in model.P.items()})
+    for sh in shards:
+        with amp_ctx():
+            c = oracle(*sh).mean()
+        c.backward()  # grads accumulate across shards
+    with torch.no_grad():
+        for q in oracle.parameters():
+            if q.grad is not None:
+                q.grad.div_(world)
+
+    # DP path: this rank's shard only
+    for p in model.parameters():
+        p.grad = None
+    with amp_ctx():
+        cost = model(*shards[rank]).mean()
+    cost.backward()
+    dp.finish()
+
+    max_err = 0.0
+    for (k, p), (k2, q) in zip(model.P.items(), oracle.P.items()):
+        assert k == k2
+        if p.grad is None or q.grad is None:
+            assert p.grad is None and q.grad is None, k
+            continue
+        err = float((p.grad - q.grad).abs().max())
+        ref = float(q.grad.abs().max())
+        max_err = max(max_err, err / max(ref, 1e-30))
+    report["grad_allreduce_max_rel_err"] = max_err
+    # tolerance covers fp32 atomic-order nondeterminism in the HIP
+    # backward kernels between the oracle run and the DP run; the
+    # collective itself adds nothing (2-rank sum, same order)
+    assert max_err < 1e-4, "bucketed RCCL all-reduce != oracle mean grad"
+
+    # ---- all_agree / all_reduce_scalar over RCCL ----
+    assert dp.all_agree(True) is True
+    assert dp.all_agree(rank != 0) is False  # one dissenting rank
+    v = dp.all_reduce_scalar(float(rank + 1), average=True)
+    assert abs(v - (world + 1) / 2.0) < 1e-9
+    report["all_agree"] = "ok"
+    report["all_reduce_scalar"] = v
+
+    # ---- bucket_cap sweep at the CNN/DM shape ----
+    rng = numpy.random.RandomState(1234 + rank)
+    x, x_mask, y, y_mask = [
+        torch.from_numpy(a).to(device) for a in synthetic_batch(
+            rng, cfg["batch"], cfg["src"], cfg["tgt"], cfg["n_words"])]
+    sweep = {}
+    dp.detach()
+    for cap in ((25,) if fast else (8, 25, 100)):
+        dps = DataParallelGrads(model.parameters(), bucket_cap_mb=cap)
+        opt = build_optimizer("adadelta", list(model.P.items()),
+                              lrate=1e-4, clip_c=100.0)
+
+        def step():
+            opt.zero_grad()
+            with amp_ctx():
+                c = model(x, x_mask, y, y_mask).mean()
+            c.backward()
+            dps.finish()
+            opt.step()
+
+        for _ in range(3):
+            step()
+        dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        n_steps = 10
+        for _ in range(n_steps):
+            step()
+        dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+        el = time.perf_counter() - t0
+        t = torch.tensor([el], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        sweep["%dMiB" % cap] = {
+            "ms_per_step": 1e3 * float(t.item()) / n_steps,
+            "n_buckets": len(dps.buckets),
+        }
+        dps.detach()
+    report["bucket_sweep"] = sweep
+
+    if rank == 0:
+        os.makedirs("gpurun_out", exist_ok=True)
+        with open("gpurun_out/rccl_dp2.json", "w") as f:
+            json.dump(report, f, indent=1)
+        print(json.dumps(report))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

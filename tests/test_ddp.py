@@ -141,3 +141,35 @@ def test_bench_multirank_torchrun_contract(tmp_path):
     assert r["n_gpus"] == 2
     assert r["config"]["parallelism"] == "dp2"
     assert r["config"]["global_batch"] == 2 * r["config"]["batch_per_gpu"]
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(900)
+def test_rccl_dp2_one_gpu():
+    """Real-RCCL DP validation on hardware: 2 ranks sharing one MI355X
+    (device mapping local_rank % device_count) running the full probe —
+    init, broadcast, bucketed all-reduce grad exactness vs a
+    single-process oracle, all_agree, scalar reduce. VERDICT r1 #1."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR",
+              "MASTER_PORT"):
+        env.pop(k, None)
+    env["NATS_PROBE_FAST"] = "1"
+    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29573", "scripts/rccl_probe.py"],
+        capture_output=True, text=True, timeout=840, cwd=repo, env=env)
+    assert out.returncode == 0, (out.stdout[-800:], out.stderr[-1500:])
+    import json as _json
+    lines = [l for l in out.stdout.splitlines() if l.strip().startswith("{")]
+    assert lines, out.stdout
+    r = _json.loads(lines[-1])
+    assert r["backend"] == "nccl"
+    assert r["world"] == 2
+    assert r["grad_allreduce_max_rel_err"] < 1e-4
