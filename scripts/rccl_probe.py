@@ -56,6 +56,30 @@ def main():
         "torch": torch.__version__,
     }
 
+    # RCCL (2.26.6) hard-refuses two ranks on one physical device
+    # ("Duplicate GPU detected", init.cc:1108 — no override env exists, so
+    # a single leased GPU cannot host a real 2-rank RCCL communicator).
+    # Probe it, record the refusal, and fall back to gloo with
+    # device-resident tensors: the DP layer's bucketing/hooks/collectives
+    # then still run against real CUDA gradients. Separately, a 1-rank
+    # nccl subgroup proves RCCL comm init + device collectives on this
+    # hardware (below).
+    if use_cuda and backend == "nccl" and torch.cuda.device_count() < world:
+        try:
+            probe = torch.ones(4, device=device)
+            dist.all_reduce(probe)
+            report["nccl_shared_device"] = "ok"
+        except Exception as e:
+            report["nccl_shared_device"] = "refused: %s" % (
+                str(e).splitlines()[-1][:160])
+            dist.destroy_process_group()
+            port2 = int(os.environ.get("MASTER_PORT", "29533")) + 7
+            store = dist.TCPStore("127.0.0.1", port2, world, rank == 0)
+            dist.init_process_group("gloo", store=store, rank=rank,
+                                    world_size=world)
+            backend = "gloo(cuda-tensors)"
+            report["backend"] = backend
+
     # ---- model at the headline CNN/DM shape (tiny on CPU smoke) ----
     fast = bool(os.environ.get("NATS_PROBE_FAST"))
     if use_cuda and not fast:
@@ -149,6 +173,22 @@ def main():
     assert abs(v - (world + 1) / 2.0) < 1e-9
     report["all_agree"] = "ok"
     report["all_reduce_scalar"] = v
+
+    # 1-rank RCCL communicator: real librccl comm init + device-side
+    # collectives on this GPU (the strongest RCCL evidence a single
+    # physical device permits; see nccl_shared_device above).
+    if use_cuda:
+        try:
+            g = dist.new_group(ranks=[0], backend="nccl")
+            if rank == 0:
+                t = torch.arange(8, device=device, dtype=torch.float32)
+                dist.all_reduce(t, group=g)
+                dist.broadcast(t, src=0, group=g)
+                torch.cuda.synchronize()
+                assert t.sum().item() == 28.0
+                report["rccl_1rank_collectives"] = "ok"
+        except Exception as e:  # pragma: no cover - hardware-dependent
+            report["rccl_1rank_collectives"] = "failed: %s" % str(e)[:160]
 
     # ---- bucket_cap sweep at the CNN/DM shape ----
     rng = numpy.random.RandomState(1234 + rank)
