@@ -58,6 +58,8 @@ def main():
                     help="force the eager (no HIP kernels) path")
     ap.add_argument("--bucket-mb", type=int, default=25,
                     help="DP gradient all-reduce bucket size (MiB)")
+    ap.add_argument("--no-graph", action="store_true",
+                    help="disable whole-step hipGraph capture")
     args = ap.parse_args()
 
     if args.eager:
@@ -101,7 +103,17 @@ def main():
     breakdown = os.environ.get("NATS_BENCH_BREAKDOWN")
     bd = {"fwd": 0.0, "bwd": 0.0, "opt": 0.0, "n": 0}
 
+    # whole-step hipGraph capture (one replay per step instead of ~2000
+    # launches); falls back to eager if capture fails
+    graph_step = None
+    if use_cuda and not args.no_graph and not breakdown and not args.eager:
+        from nats_amd.utils.step_graph import GraphedStepCache
+        gcache = GraphedStepCache(model, opt, dp if world > 1 else None)
+        graph_step = gcache.get(x, x_mask, y, y_mask)
+
     def step():
+        if graph_step is not None:
+            return graph_step.step(x, x_mask, y, y_mask)
         opt.zero_grad()
         if breakdown:
             torch.cuda.synchronize()
@@ -188,6 +200,7 @@ def main():
                 "seq_len": cfg["src"],
                 "optimizer": "adadelta+clip100",
                 "parallelism": "dp%d" % world,
+                "step_graph": graph_step is not None,
                 "final_cost": float(last_cost.detach()),
             },
         }

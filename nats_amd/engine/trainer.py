@@ -59,7 +59,7 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
           validFreq=1000, saveFreq=1000, sampleFreq=100, datasets=[],
           valid_datasets=[], dictionary="", use_dropout=False, reload_=False,
           verbose=False, device=None, seed=None, enc_depth=1,
-          profile=False, resume_optimizer=False):
+          profile=False, resume_optimizer=False, step_graph=None):
     """Train the distraction model; returns final validation error.
 
     Signature (and defaults) mirror nats.py:1230-1257; `device`/`seed`/
@@ -76,6 +76,12 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
     dispFreq. ``resume_optimizer=True`` also saves/loads the optimizer
     sidecar ``<saveto>.opt.npz`` with each checkpoint (default off: the
     reference restarts adadelta accumulators from zero on resume).
+
+    ``step_graph``: None (auto: on for CUDA when compatible) / True /
+    False — capture fwd+bwd+optimizer into one hipGraph per input shape
+    (utils/step_graph.py; the analogue of the reference's compiled
+    f_grad_shared+f_update pair being one host call). Ragged shapes
+    beyond the cache limit run eager.
     """
     logging.basicConfig(
         level=logging.DEBUG,
@@ -135,6 +141,16 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
         from ..utils import StepTimer
         step_timer = StepTimer()
 
+    # whole-step hipGraph capture (auto-on for CUDA; incompatible with
+    # decay_c — the weight-decay term is added outside the closure — and
+    # pointless with the per-section profiler's syncs)
+    gcache = None
+    use_graph = step_graph if step_graph is not None else True
+    if (use_graph and str(device).startswith("cuda") and decay_c == 0.0
+            and not profile):
+        from ..utils.step_graph import GraphedStepCache
+        gcache = GraphedStepCache(model, opt, dp if world > 1 else None)
+
     history_errs = []
     if reload_history is not None:
         print("Reload history error")
@@ -178,28 +194,34 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
             ud_start = time.time()
             x, x_mask, y, y_mask = _to_device((x, x_mask, y, y_mask), device)
 
-            opt.zero_grad()
-            if step_timer is not None:
-                step_timer.start("forward")
-            cost_vec = model(x, x_mask, y, y_mask)
-            cost = cost_vec.mean()
-            if decay_c > 0.0:
-                weight_decay = sum((p ** 2).sum() for p in model.parameters())
-                cost = cost + decay_c * weight_decay
-            if step_timer is not None:
-                step_timer.stop("forward")
-                step_timer.start("backward")
-            cost.backward()
-            if step_timer is not None:
-                step_timer.stop("backward")
-                step_timer.start("allreduce")
-            dp.finish()
-            if step_timer is not None:
-                step_timer.stop("allreduce")
-                step_timer.start("optimizer")
-            norm_g = opt.step()
-            if step_timer is not None:
-                step_timer.stop("optimizer")
+            gstep = gcache.get(x, x_mask, y, y_mask) if gcache else None
+            if gstep is not None:
+                cost = gstep.step(x, x_mask, y, y_mask)
+                norm_g = None
+            else:
+                opt.zero_grad()
+                if step_timer is not None:
+                    step_timer.start("forward")
+                cost_vec = model(x, x_mask, y, y_mask)
+                cost = cost_vec.mean()
+                if decay_c > 0.0:
+                    weight_decay = sum((p ** 2).sum()
+                                       for p in model.parameters())
+                    cost = cost + decay_c * weight_decay
+                if step_timer is not None:
+                    step_timer.stop("forward")
+                    step_timer.start("backward")
+                cost.backward()
+                if step_timer is not None:
+                    step_timer.stop("backward")
+                    step_timer.start("allreduce")
+                dp.finish()
+                if step_timer is not None:
+                    step_timer.stop("allreduce")
+                    step_timer.start("optimizer")
+                norm_g = opt.step()
+                if step_timer is not None:
+                    step_timer.stop("optimizer")
             cost_val = float(cost.detach())
             ud = time.time() - ud_start
 
@@ -211,7 +233,7 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
             if numpy.mod(uidx, dispFreq) == 0:
                 logger.debug("Epoch {0} Update {1} Cost {2} UD {3}".format(
                     eidx, uidx, cost_val, ud))
-                if verbose and clip_c > 0:
+                if verbose and clip_c > 0 and norm_g is not None:
                     logger.debug("Grad {0}".format(float(norm_g)))
                 if step_timer is not None:
                     logger.debug("Step breakdown:\n%s", step_timer.report())

@@ -408,6 +408,11 @@ __global__ void cond_gru1_step_pointwise(
 
 // ---------------- backward kernels ----------------
 
+// Fused per-step backward prologue: the GRU_1 pointwise, the independent
+// dctx passthrough (upstream readout grad + acc-chain, formerly its own
+// cond_dctx_dir launch), and the per-step re-zero of dot_buf (consumed by
+// last step's scatter/reduce before this kernel runs — stream order) all
+// share one grid-stride index space.
 __global__ void cond_gru1_bwd_pointwise(
     const float* __restrict__ dh_carry,   // [B][H]
     const float* __restrict__ dh2_all_t,  // [B][H] or null
@@ -420,10 +425,30 @@ __global__ void cond_gru1_bwd_pointwise(
     int ldK3,
     float* __restrict__ ddirect_h1,       // [B][H]
     bf16_t* __restrict__ dpre1_t,         // [B][4H]
+    const float* __restrict__ dctxs_t,    // [B][C] or null (upstream)
+    const float* __restrict__ daccC,      // [B][C] (pre-update, read)
+    float* __restrict__ dctx_dir,         // [B][C] out
+    int C,
+    float* __restrict__ dot_buf,          // [B] (zeroed for this step)
     int B, int H) {
   const long total = (long)B * H;
-  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
+  const long total_all = total + (long)B * C + B;
+  for (long idx2 = blockIdx.x * blockDim.x + threadIdx.x; idx2 < total_all;
+       idx2 += (long)gridDim.x * blockDim.x) {
+    if (idx2 >= total) {
+      const long e = idx2 - total;
+      if (e < (long)B * C) {
+        const int b = e / C;
+        const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
+        float v = mm * daccC[e];
+        if (dctxs_t != nullptr) v += dctxs_t[e];
+        dctx_dir[e] = v;
+      } else {
+        dot_buf[e - (long)B * C] = 0.f;
+      }
+      continue;
+    }
+    const long idx = idx2;
     const int b = idx / H;
     const int j = idx % H;
     float dh2 = dh_carry[idx];
@@ -455,41 +480,127 @@ __global__ void cond_gru1_bwd_pointwise(
   }
 }
 
-__global__ void cond_dctx_dir(const float* __restrict__ dctxs_t,
-                              const float* __restrict__ daccC,
-                              const float* __restrict__ mask_t,
-                              float* __restrict__ dctx_dir, int B, int C) {
-  const long total = (long)B * C;
-  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    const int b = idx / C;
-    const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
-    float v = mm * daccC[idx];
-    if (dctxs_t != nullptr) v += dctxs_t[idx];
-    dctx_dir[idx] = v;
+// Dual recurrent-GEMM with the distraction-gate backward fused into the
+// context side's epilogue. grid.y == 0: dh1 = ddirect_h1 + dstep1 @
+// [U_1|Ux_1]^T (plain). grid.y == 1: the former dctx_buf value (dctx_dir
+// + dstepC @ [W_1|Wx_1]^T) never touches memory — the gate backward
+// (tanh', Ucon/Wcon splits, daccC accumulate; nats.py:545-546 reverse)
+// is applied in-register, eliminating the separate cond_gate_bwd pass.
+__global__ __launch_bounds__(384) void cond_bwd_gemm_dual_gate(
+    const bf16_t* __restrict__ dstep1, const bf16_t* __restrict__ U1cat,
+    const float* __restrict__ ddirect_h1, float* __restrict__ dh1_buf,
+    int H, const bf16_t* __restrict__ dstepC,
+    const bf16_t* __restrict__ W1cat, const float* __restrict__ dctx_dir,
+    int C, int Kpad,
+    const float* __restrict__ ctxs_t,   // [B][C] gated value
+    const float* __restrict__ Ucon, const float* __restrict__ Wcon,
+    float* __restrict__ daccC,          // [B][C] in/out
+    float* __restrict__ dctxpre_f32,    // [B][C] out
+    bf16_t* __restrict__ dctxpre_all_t, // [B][C] out
+    bf16_t* __restrict__ dgate_all_t,   // [B][C] out
+    int B) {
+  __shared__ float part[3][32][JB + 1];
+  const bool ctx_side = (blockIdx.y == 1);
+  const int N = ctx_side ? C : H;
+  if ((int)blockIdx.x * JB >= N) return;
+  const bf16_t* dstep = ctx_side ? dstepC : dstep1;
+  const bf16_t* Wt = ctx_side ? W1cat : U1cat;
+
+  const int wg = blockIdx.x;
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int m = wave / 3;
+  const int ks = wave % 3;
+  const int i0 = wg * JB;
+  const int kchunk = ((Kpad / 3 + 31) / 32) * 32;
+  const int kbeg = ks * kchunk;
+  const int kend = min(Kpad, (ks + 1) * kchunk);
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  NATS_MFMA_KLOOP(acc, dstep, 16 * m, Kpad, Wt, i0, Kpad, kbeg, kend);
+  {
+    const int lane = threadIdx.x & (NATS_WAVE - 1);
+    const int col = lane & 15;
+    const int rbase = 16 * m + (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) part[ks][rbase + i][col] = acc[i];
+  }
+  __syncthreads();
+
+  for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+    const int b = idx / JB;
+    const int cc = idx % JB;
+    const int i = i0 + cc;
+    if (i >= N) continue;
+    const long bi = (long)b * N + i;
+    const float sum = part[0][b][cc] + part[1][b][cc] + part[2][b][cc];
+    if (!ctx_side) {
+      dh1_buf[bi] = ddirect_h1[bi] + sum;
+    } else {
+      const float v = dctx_dir[bi] + sum;  // former dctx_buf value
+      const float g = ctxs_t[bi];
+      const float dg = v * (1.f - g * g);
+      const float dpre = dg * Ucon[i];
+      dctxpre_f32[bi] = dpre;
+      dctxpre_all_t[bi] = (bf16_t)dpre;
+      dgate_all_t[bi] = (bf16_t)dg;
+      daccC[bi] += dg * Wcon[i];
+    }
   }
 }
 
-__global__ void cond_gate_bwd(const float* __restrict__ dctx_buf,
-                              const float* __restrict__ ctxs_t,  // gated val
-                              const float* __restrict__ Ucon,
-                              const float* __restrict__ Wcon,
-                              float* __restrict__ daccC,          // in/out
-                              float* __restrict__ dctxpre_f32,    // [B][C]
-                              bf16_t* __restrict__ dctxpre_all_t, // [B][C]
-                              bf16_t* __restrict__ dgate_all_t,   // [B][C]
-                              int B, int C) {
-  const long total = (long)B * C;
-  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    const int c = idx % C;
-    const float g = ctxs_t[idx];
-    const float dg = dctx_buf[idx] * (1.f - g * g);
-    const float dpre = dg * Ucon[c];
-    dctxpre_f32[idx] = dpre;
-    dctxpre_all_t[idx] = (bf16_t)dpre;
-    dgate_all_t[idx] = (bf16_t)dg;
-    daccC[idx] += dg * Wcon[c];
+// A-fragment built from an UNPADDED fp32 [Rows][Kcols] matrix with
+// zero-fill outside (rows >= Rows, k >= Kcols) — lets the dh1 += dpstate
+// @ W_att^T GEMM consume the atomically-reduced fp32 dpstate directly,
+// without the former cond_dpstate_cast pass.
+__device__ __forceinline__ bf16x8 frag_a_f32pad(const float* A, int row0,
+                                                int ld, int k0, int Rows,
+                                                int Kcols) {
+  const int lane = threadIdx.x & (NATS_WAVE - 1);
+  const int r = row0 + (lane & 15);
+  const int kb = k0 + (lane >> 4) * 8;
+  bf16x8 v;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    const int k = kb + i;
+    v[i] = (bf16_t)((r < Rows && k < Kcols) ? A[(long)r * ld + k] : 0.f);
+  }
+  return v;
+}
+
+__global__ __launch_bounds__(384) void cond_dh1_att_gemm(
+    const float* __restrict__ dpstate_t,  // [B][A] fp32 (atomic-reduced)
+    const bf16_t* __restrict__ WattB,     // [ngrpH*16][Apad]
+    float* __restrict__ dh1_buf,          // [B][H] in/out (+=)
+    int B, int H, int A, int Apad) {
+  __shared__ float part[3][32][JB + 1];
+  const int wg = blockIdx.x;
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int m = wave / 3;
+  const int ks = wave % 3;
+  const int i0 = wg * JB;
+  const int kchunk = ((Apad / 3 + 31) / 32) * 32;
+  const int kbeg = ks * kchunk;
+  const int kend = min(Apad, (ks + 1) * kchunk);
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int k = kbeg; k + 32 <= kend; k += 32) {
+    bf16x8 a = frag_a_f32pad(dpstate_t, 16 * m, A, k, B, A);
+    bf16x8 bfr = frag_bt_rowmajor(WattB, i0, Apad, k);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
+  }
+  {
+    const int lane = threadIdx.x & (NATS_WAVE - 1);
+    const int col = lane & 15;
+    const int rbase = 16 * m + (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) part[ks][rbase + i][col] = acc[i];
+  }
+  __syncthreads();
+  for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+    const int b = idx / JB;
+    const int cc = idx % JB;
+    const int i = i0 + cc;
+    if (i >= H) continue;
+    dh1_buf[(long)b * H + i] +=
+        part[0][b][cc] + part[1][b][cc] + part[2][b][cc];
   }
 }
 
@@ -510,56 +621,45 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_dalpha(
   const int wave = threadIdx.x / NATS_WAVE;
   const int lane = threadIdx.x & (NATS_WAVE - 1);
   const int s = blockIdx.y * 4 + wave;
-  if (s >= Ts) return;
-  const bf16_t* crow = ctx_bf + ((long)s * B + b) * C;
-  const float* drow = dctxpre_f32 + (long)b * C;
-  float part = 0.f;
-  const int C8 = C & ~7;
-  for (int c = lane * 8; c < C8; c += NATS_WAVE * 8) {
-    bf16x8 v = *(const bf16x8*)(crow + c);
-    const float4 d0 = *(const float4*)(drow + c);
-    const float4 d1 = *(const float4*)(drow + c + 4);
-    part += (float)v[0] * d0.x + (float)v[1] * d0.y + (float)v[2] * d0.z +
-            (float)v[3] * d0.w + (float)v[4] * d1.x + (float)v[5] * d1.y +
-            (float)v[6] * d1.z + (float)v[7] * d1.w;
-  }
-  for (int c = C8 + lane; c < C; c += NATS_WAVE) {
-    part += (float)crow[c] * drow[c];
-  }
+  // softmax-backward dot: block-local tree over the 4 waves' (alpha*dal),
+  // ONE atomicAdd per block into dot_buf[b] (~Ts/4 adds spread over B
+  // addresses — the per-s variant serialized ~Ts adds on one address at
+  // ~120us/step, and a separate cond_attn_bwd_dot pass cost a full
+  // launch + Ts*B re-read; this fold removes that kernel).
+  __shared__ float wdot[4];
+  float dal = 0.f;
+  if (s < Ts) {
+    const bf16_t* crow = ctx_bf + ((long)s * B + b) * C;
+    const float* drow = dctxpre_f32 + (long)b * C;
+    float part = 0.f;
+    const int C8 = C & ~7;
+    for (int c = lane * 8; c < C8; c += NATS_WAVE * 8) {
+      bf16x8 v = *(const bf16x8*)(crow + c);
+      const float4 d0 = *(const float4*)(drow + c);
+      const float4 d1 = *(const float4*)(drow + c + 4);
+      part += (float)v[0] * d0.x + (float)v[1] * d0.y + (float)v[2] * d0.z +
+              (float)v[3] * d0.w + (float)v[4] * d1.x + (float)v[5] * d1.y +
+              (float)v[6] * d1.z + (float)v[7] * d1.w;
+    }
+    for (int c = C8 + lane; c < C; c += NATS_WAVE) {
+      part += (float)crow[c] * drow[c];
+    }
 #pragma unroll
-  for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
-    part += __shfl_down(part, off);
-  if (lane == 0) {
-    float dal = part;
-    const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
-    dal += mm * daccA[(long)b * Ts + s];
-    if (dalphas_t != nullptr) dal += dalphas_t[(long)b * Ts + s];
-    dal_buf[(long)s * B + b] = dal;
-    // NOTE: dot(alpha, dal) is computed by cond_attn_bwd_dot — a per-s
-    // atomicAdd here serializes ~Ts adds on one address (~120us/step).
+    for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
+      part += __shfl_down(part, off);
+    if (lane == 0) {
+      dal = part;
+      const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
+      dal += mm * daccA[(long)b * Ts + s];
+      if (dalphas_t != nullptr) dal += dalphas_t[(long)b * Ts + s];
+      dal_buf[(long)s * B + b] = dal;
+    }
   }
-}
-
-// softmax-backward dot: dot_buf[b] = sum_s alpha[b,s] * dal[s,b]
-__global__ __launch_bounds__(256) void cond_attn_bwd_dot(
-    const float* __restrict__ alphas_t, const float* __restrict__ dal_buf,
-    float* __restrict__ dot_buf, int B, int Ts) {
-  const int b = blockIdx.x;
-  __shared__ float red[256 / NATS_WAVE];
-  float part = 0.f;
-  for (int s = threadIdx.x; s < Ts; s += blockDim.x)
-    part += alphas_t[(long)b * Ts + s] * dal_buf[(long)s * B + b];
-#pragma unroll
-  for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
-    part += __shfl_down(part, off);
-  if ((threadIdx.x & (NATS_WAVE - 1)) == 0)
-    red[threadIdx.x / NATS_WAVE] = part;
+  if (lane == 0)
+    wdot[wave] = (s < Ts) ? alphas_t[(long)b * Ts + s] * dal : 0.f;
   __syncthreads();
-  if (threadIdx.x == 0) {
-    float S = 0.f;
-    for (int w = 0; w < (int)blockDim.x / NATS_WAVE; ++w) S += red[w];
-    dot_buf[b] = S;
-  }
+  if (threadIdx.x == 0)
+    atomicAdd(dot_buf + b, wdot[0] + wdot[1] + wdot[2] + wdot[3]);
 }
 
 // attention backward, stage 2 (grid (B, ceil(Ts/256))): softmax backward,
@@ -647,7 +747,6 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_reduce(
     const bf16_t* __restrict__ pc_buf,      // [B][A][Tpad8]
     const float* __restrict__ Uatt,
     float* __restrict__ dpstate_t,          // [B][A]
-    bf16_t* __restrict__ dstep_att,         // [32][Apad]
     float* __restrict__ gdDwei,             // [A] (atomic)
     float* __restrict__ gdUatt,             // [A] (atomic)
     float* __restrict__ gdcatt,             // [1] (atomic)
@@ -716,19 +815,6 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_reduce(
     atomicAdd(&dpstate_t[(long)b * A + i], sps);
     atomicAdd(&gdDwei[i], sdw);
     atomicAdd(&gdUatt[i], sua);
-  }
-}
-
-// bf16 copy of the reduced dpstate into the padded GEMM operand
-__global__ void cond_dpstate_cast(const float* __restrict__ dpstate_t,
-                                  bf16_t* __restrict__ dstep_att, int Apad,
-                                  int B, int A) {
-  const long total = (long)B * A;
-  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    const int b = idx / A;
-    const int i = idx % A;
-    dstep_att[(long)b * Apad + i] = (bf16_t)dpstate_t[idx];
   }
 }
 
@@ -953,7 +1039,6 @@ std::vector<torch::Tensor> cond_gru_bwd(
   auto ddirect_h1 = torch::empty({B, H}, optsF);
   auto ddirect2 = torch::empty({B, H}, optsF);
   auto dctx_dir = torch::empty({B, C}, optsF);
-  auto dctx_buf = torch::empty({B, C}, optsF);
   auto dctxpre_f32 = torch::empty({B, C}, optsF);
   auto daccA = daccA_f.has_value() ? daccA_f->contiguous().to(torch::kFloat32)
                                    : torch::zeros({B, Ts}, optsF);
@@ -968,7 +1053,6 @@ std::vector<torch::Tensor> cond_gru_bwd(
   auto dstep1 = torch::zeros({32, K3Hpad}, optsB);
   auto dstepC = torch::zeros({32, K3Hpad}, optsB);
   auto dstep2 = torch::zeros({32, K3Hpad}, optsB);
-  auto dstep_att = torch::zeros({32, Apad32}, optsB);
   auto init_f = init_state.contiguous().to(torch::kFloat32);
 
   const float* mask_all = nullptr;
@@ -993,15 +1077,16 @@ std::vector<torch::Tensor> cond_gru_bwd(
   auto stream = at::cuda::getCurrentCUDAStream().stream();
   auto dh2_c = dh2_all.contiguous().to(torch::kFloat32);
   const int pwH = (int)std::min<long>(512, (((long)B * H) + 255) / 256);
-  const int pwC = (int)std::min<long>(512, (((long)B * C) + 255) / 256);
+  const int pwHC =
+      (int)std::min<long>(512, (((long)B * (H + C) + B) + 255) / 256);
 
   for (int t = T - 1; t >= 0; --t) {
     const float* mt = mask_all ? mask_all + (long)t * B : nullptr;
     const float* h2prev =
         (t == 0) ? init_f.data_ptr<float>()
                  : h2_all.data_ptr<float>() + (long)(t - 1) * B * H;
-    // b1: GRU_1 pointwise
-    hipLaunchKernelGGL(cond_gru1_bwd_pointwise, dim3(pwH), dim3(256), 0,
+    // b1 (fused): GRU_1 pointwise + dctx passthrough + dot_buf re-zero
+    hipLaunchKernelGGL(cond_gru1_bwd_pointwise, dim3(pwHC), dim3(256), 0,
                        stream, dh_carry.data_ptr<float>(),
                        dh2_c.data_ptr<float>() + (long)t * B * H,
                        (const bf16_t*)saved1.data_ptr() + (long)t * B * 4 * H,
@@ -1009,35 +1094,30 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        bx1.data_ptr<float>(), mt,
                        (bf16_t*)dstep1.data_ptr(), (bf16_t*)dstepC.data_ptr(),
                        K3Hpad, ddirect_h1.data_ptr<float>(),
-                       (bf16_t*)dpre1_all.data_ptr() + (long)t * B * 4 * H, B,
-                       H);
-    // b0: dctx passthrough (readout grad + acc chain)
-    hipLaunchKernelGGL(cond_dctx_dir, dim3(pwC), dim3(256), 0, stream,
+                       (bf16_t*)dpre1_all.data_ptr() + (long)t * B * 4 * H,
                        dctxs_p ? dctxs_p + (long)t * B * C : nullptr,
-                       daccC.data_ptr<float>(), mt,
-                       dctx_dir.data_ptr<float>(), B, C);
-    // b2+b3 fused launch: dh1 = ddirect_h1 + dstep1 @ [U_1|Ux_1]^T and
-    // dctx = dctx_dir + dstepC @ [W_1|Wx_1]^T (independent problems)
-    hipLaunchKernelGGL(nats_gru_step_bwd_gemm_dual,
+                       daccC.data_ptr<float>(), dctx_dir.data_ptr<float>(), C,
+                       dot_buf.data_ptr<float>(), B, H);
+    // b2+b3+b4 fused launch: dh1 = ddirect_h1 + dstep1 @ [U_1|Ux_1]^T and
+    // the dctx GEMM with the distraction-gate backward applied in the
+    // epilogue (the former separate cond_gate_bwd pass)
+    hipLaunchKernelGGL(cond_bwd_gemm_dual_gate,
                        dim3(std::max(ngrpH, ngrpC), 2), dim3(384), 0, stream,
                        (const bf16_t*)dstep1.data_ptr(),
                        (const bf16_t*)U1cat.data_ptr(),
                        ddirect_h1.data_ptr<float>(),
-                       dh1_buf.data_ptr<float>(), H, K3Hpad,
+                       dh1_buf.data_ptr<float>(), H,
                        (const bf16_t*)dstepC.data_ptr(),
                        (const bf16_t*)W1cat.data_ptr(),
-                       dctx_dir.data_ptr<float>(), dctx_buf.data_ptr<float>(),
-                       C, K3Hpad, B);
-    // b4: distraction gate backward
-    hipLaunchKernelGGL(cond_gate_bwd, dim3(pwC), dim3(256), 0, stream,
-                       dctx_buf.data_ptr<float>(),
+                       dctx_dir.data_ptr<float>(), C, K3Hpad,
                        ctxs_all.data_ptr<float>() + (long)t * B * C,
                        Ucon.data_ptr<float>(), Wcon.data_ptr<float>(),
                        daccC.data_ptr<float>(), dctxpre_f32.data_ptr<float>(),
                        (bf16_t*)dctxpre_all.data_ptr() + (long)t * B * C,
-                       (bf16_t*)dgate_all.data_ptr() + (long)t * B * C, B, C);
-    // b5: attention backward — wave-per-(b,s) dalpha + softmax-bwd dot,
-    // the (b,s)-parallel scatter (dpctx/daccA/pc_buf), and the per-(b,i)
+                       (bf16_t*)dgate_all.data_ptr() + (long)t * B * C, B);
+    // b5: attention backward — wave-per-(b,s) dalpha with the softmax-bwd
+    // dot folded in (block tree + one atomic per block), then the
+    // (b,s)-parallel scatter (dpctx/daccA/pc_buf) and the per-(b,i)
     // s-contiguous reduce for dpstate/dD_wei/dU_att/dc_att
     hipLaunchKernelGGL(cond_attn_bwd_dalpha, dim3(B, cdiv_i(Ts, 4)),
                        dim3(256), 0, stream,
@@ -1048,10 +1128,6 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        dalpha_p ? dalpha_p + (long)t * B * Ts : nullptr,
                        dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
                        B, Ts, C);
-    hipLaunchKernelGGL(cond_attn_bwd_dot, dim3(B), dim3(256), 0, stream,
-                       alphas_all.data_ptr<float>() + (long)t * B * Ts,
-                       dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
-                       B, Ts);
     hipLaunchKernelGGL(cond_attn_bwd_scatter,
                        dim3(B, cdiv_i(Ts, 256), A >= 32 ? 8 : (A >= 16 ? 4 : 1)),
                        dim3(256), 0, stream,
@@ -1072,20 +1148,15 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        (const bf16_t*)pc_buf.data_ptr(),
                        Uatt.data_ptr<float>(),
                        dpstate_all.data_ptr<float>() + (long)t * B * A,
-                       (bf16_t*)dstep_att.data_ptr(),
                        gdDwei.data_ptr<float>(), gdUatt.data_ptr<float>(),
                        gdcatt.data_ptr<float>(), B, Ts, A, Apad32, Tpad8,
                        RSCH);
-    hipLaunchKernelGGL(cond_dpstate_cast, dim3(cdiv_i(B * A, 256)), dim3(256),
-                       0, stream,
+    // b6a: dh1 += dpstate @ W_att^T — A fragments converted from the
+    // fp32 dpstate in-register (no cond_dpstate_cast pass)
+    hipLaunchKernelGGL(cond_dh1_att_gemm, dim3(ngrpH), dim3(384), 0, stream,
                        dpstate_all.data_ptr<float>() + (long)t * B * A,
-                       (bf16_t*)dstep_att.data_ptr(), Apad32, B, A);
-    // b6a: dh1 += dpstate @ W_att^T (in-place add via ddirect aliasing)
-    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH), dim3(384), 0,
-                       stream, (const bf16_t*)dstep_att.data_ptr(),
                        (const bf16_t*)WattB.data_ptr(),
-                       dh1_buf.data_ptr<float>(), dh1_buf.data_ptr<float>(),
-                       B, H, Apad32);
+                       dh1_buf.data_ptr<float>(), B, H, A, Apad32);
     // b8: GRU_2 pointwise (dh1 -> gate preact grads)
     hipLaunchKernelGGL(nats_gru_step_bwd_pointwise, dim3(pwH), dim3(256), 0,
                        stream, dh1_buf.data_ptr<float>(), nullptr,

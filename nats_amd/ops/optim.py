@@ -46,13 +46,20 @@ class FusedAdadelta(Adadelta):
         g2.zero_()
         rg2 = [self.state[k]["rg2"] for k, _ in self.named]
         ru2 = [self.state[k]["ru2"] for k, _ in self.named]
-        # pointer table rebuilt each step (grad storages change)
-        ptrs = torch.tensor(
-            [p.data_ptr() for p in ps] +
-            [p.grad.data_ptr() for p in ps] +
-            [t.data_ptr() for t in rg2] +
-            [t.data_ptr() for t in ru2],
-            dtype=torch.int64, device=device)
+        # pointer table: rebuilt only when a storage moved (grads keep
+        # their buffers across steps under a steady allocator, and MUST
+        # for hipGraph capture — an H2D upload of a fresh pageable tensor
+        # is not capturable, and replays need stable addresses anyway)
+        host_ptrs = ([p.data_ptr() for p in ps] +
+                     [p.grad.data_ptr() for p in ps] +
+                     [t.data_ptr() for t in rg2] +
+                     [t.data_ptr() for t in ru2])
+        cached = getattr(self, "_ptr_cache", None)
+        if cached is not None and cached[0] == host_ptrs:
+            ptrs = cached[1]
+        else:
+            ptrs = torch.tensor(host_ptrs, dtype=torch.int64, device=device)
+            self._ptr_cache = (host_ptrs, ptrs)
         ext.fused_adadelta_step(ptrs, sizes_t, chunk_tensor, chunk_off, g2,
                                 float(self.clip_c), self.rho, self.eps)
         # The fused kernel writes parameters through raw pointers, which
