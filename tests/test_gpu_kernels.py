@@ -16,6 +16,33 @@ def ext():
     return m
 
 
+def assert_grad_close(a, b, name="", rel=0.06, floor_frac=1e-3):
+    """Scale-aware gradient comparison.
+
+    Global-max normalization (rel = max|a-b| / max|a|) lets a large
+    relative error on a small-magnitude COLUMN hide under the tensor's
+    largest column. Instead: normalize per output column (last axis —
+    each column of a weight grad is an independent sum over (T,B)), with
+    a small absolute floor of floor_frac * global_max covering bf16
+    swamping where large contributions cancel. floor_frac=1e-3 is ~60x
+    tighter than the old global-max bound for the smallest columns.
+    """
+    a2 = a.reshape(-1, a.shape[-1]) if a.dim() > 1 else a.reshape(1, -1)
+    b2 = b.reshape(-1, b.shape[-1]) if b.dim() > 1 else b.reshape(1, -1)
+    col_scale = a2.abs().amax(dim=0)
+    floor = floor_frac * a2.abs().max().clamp_min(1e-6)
+    col_err = (a2 - b2).abs().amax(dim=0)
+    bound = rel * col_scale + floor
+    bad = col_err > bound
+    assert not bool(bad.any()), (
+        "%s: %d/%d columns out of bound; worst col err %.3e vs bound %.3e "
+        "(col scale %.3e, floor %.3e)" % (
+            name, int(bad.sum()), bad.numel(),
+            float(col_err[bad].max()) if bad.any() else 0.0,
+            float(bound[bad].min()) if bad.any() else 0.0,
+            float(col_scale[bad.argmax()]), float(floor)))
+
+
 def test_mfma_fragment_layout(ext):
     """Transpose-detecting self-test of the MFMA lane maps (asymmetric B)."""
     torch.manual_seed(0)
@@ -74,10 +101,7 @@ def test_gru_scan_backward(ext):
     (h_hip * loss_w.cuda()).sum().backward()
 
     for r, h, name in zip(ref_in, hip_in, ["xg", "xc", "U", "Ux"]):
-        a, b = r.grad, h.grad.cpu().float()
-        denom = a.abs().max().clamp_min(1e-3)
-        rel = (a - b).abs().max() / denom
-        assert rel < 0.06, (name, float(rel))
+        assert_grad_close(r.grad, h.grad.cpu().float(), name, rel=0.06)
 
 
 @pytest.mark.parametrize("V", [120, 3001, 30000])
@@ -239,15 +263,9 @@ def test_cond_gru_backward(ext):
     hg, hi = run("cuda", cond_gru_scan_hip)
 
     for k in rg:
-        a, b = rg[k], hg[k]
-        denom = a.abs().max().clamp_min(1e-4)
-        rel = (a - b).abs().max() / denom
-        assert rel < 0.08, (k, float(rel))
+        assert_grad_close(rg[k], hg[k], k, rel=0.08)
     for k in ri:
-        a, b = ri[k], hi[k]
-        denom = a.abs().max().clamp_min(1e-4)
-        rel = (a - b).abs().max() / denom
-        assert rel < 0.08, (k, float(rel))
+        assert_grad_close(ri[k], hi[k], k, rel=0.08)
 
 
 def test_cond_gru_one_step(ext):
@@ -301,9 +319,7 @@ def test_gru_scan_bidir(ext):
     for r, h, name in zip(ref_in, hip_in,
                           ["xg0", "xc0", "U0", "Ux0", "xg1", "xc1", "U1",
                            "Ux1"]):
-        a, b = r.grad, h.grad.cpu().float()
-        rel = (a - b).abs().max() / a.abs().max().clamp_min(1e-3)
-        assert rel < 0.08, (name, float(rel))
+        assert_grad_close(r.grad, h.grad.cpu().float(), name, rel=0.08)
 
 
 def test_graph_decode_matches_plain(ext):
