@@ -102,9 +102,11 @@ class NatsModel(nn.Module):
                     numpy.ascontiguousarray(params[k]), dtype=self.P[k].dtype))
 
     # -- graph pieces ------------------------------------------------------
-    def embed(self, ids):
-        """Wemb gather; ids (T,B) or (B,) int64."""
-        return self.P["Wemb"][ids]
+    def embed(self, ids, shift=False):
+        """Wemb gather; ids (T,B) or (B,) int64. shift=True fuses the
+        decoder's shift-right-with-zero-BOS row (nats.py:730-734) into
+        the gather kernel (ops/hip/embed.hip)."""
+        return ops.embed_gather(self.P["Wemb"], ids, shift=shift)
 
     def encode(self, x, x_mask=None):
         """Bidirectional encode + decoder init state.
@@ -167,9 +169,7 @@ class NatsModel(nn.Module):
         ctx, init_state = self.encode(x, x_mask)
         pctx = self.project_ctx(ctx)
 
-        emb = self.embed(y)
-        emb_shifted = torch.zeros_like(emb)
-        emb_shifted[1:] = emb[:-1]
+        emb_shifted = self.embed(y, shift=True)
         yg, yc = self._dec_inputs(emb_shifted)
 
         h2s, ctxs, alphas, _, _ = ops.cond_gru_scan(

@@ -143,6 +143,22 @@ def cond_gru_step(h_prev, x_g, x_c, ctx, ctx_mask, pctx, acc_ctx, acc_alpha, P):
                                acc_ctx, acc_alpha, P)
 
 
+def embed_gather(Wemb, ids, shift=False):
+    """Embedding gather, optionally fused with the decoder's
+    shift-right-by-one (BOS row = zeros). Semantics: nats.py:700-701
+    (plain) and 730-734 (shifted target embedding).
+    """
+    if Wemb.dtype == torch.float32 and _use_hip(Wemb):
+        from .embed import embed_gather_hip
+        return embed_gather_hip(Wemb, ids, shift=shift)
+    emb = Wemb[ids.clamp_min(0)]
+    if shift and ids.dim() == 2:
+        shifted = torch.zeros_like(emb)
+        shifted[1:] = emb[:-1]
+        return shifted
+    return emb
+
+
 def softmax_xent(logits, targets):
     """Per-position NLL of a softmax over the vocabulary.
 

@@ -59,7 +59,16 @@ std::vector<torch::Tensor> cond_gru_bwd(
     torch::Tensor bx1, torch::Tensor Dwei, torch::Tensor Uatt,
     torch::Tensor Ucon, torch::Tensor Wcon);
 
+torch::Tensor embed_gather(torch::Tensor Wemb, torch::Tensor ids,
+                           long shift_rows);
+torch::Tensor embed_scatter_add(torch::Tensor dout, torch::Tensor ids,
+                                long V, long shift_rows);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("embed_gather", &embed_gather,
+        "embedding gather (+fused decoder shift)");
+  m.def("embed_scatter_add", &embed_scatter_add,
+        "embedding backward scatter-add");
   m.def("gru_scan_fwd", &gru_scan_fwd, "fused GRU scan forward");
   m.def("gru_scan_bwd", &gru_scan_bwd, "fused GRU scan backward");
   m.def("gru_scan_fwd_bidir", &gru_scan_fwd_bidir,

@@ -19,6 +19,7 @@ SOURCES = [
     os.path.join(HERE, "mfma_test.hip"),
     os.path.join(HERE, "optim.hip"),
     os.path.join(HERE, "rerank.hip"),
+    os.path.join(HERE, "embed.hip"),
 ]
 
 
