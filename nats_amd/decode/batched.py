@@ -27,7 +27,8 @@ from .beam import distraction_penalties_gpu
 
 @torch.no_grad()
 def gen_sample_batched(model, xs, k=1, maxlen=30, use_unk=False,
-                       kl_factor=0.0, ctx_factor=0.0, state_factor=0.0):
+                       kl_factor=0.0, ctx_factor=0.0, state_factor=0.0,
+                       use_graph=True):
     """Beam-decode a list of sources jointly.
 
     xs: list of (T_i, 1) int64 tensors (same device as model).
@@ -80,6 +81,14 @@ def gen_sample_batched(model, xs, k=1, maxlen=30, use_unk=False,
     ctx_mask_s = src_mask
     pctx_pad = model.project_ctx(ctx_pad)
 
+    # hipGraph-captured f_next for the shared step (fixed row budget S*k;
+    # the per-step sentence gather runs inside the graph)
+    stepper = None
+    if use_graph and device.type == "cuda":
+        from .graph import get_batched_stepper
+        stepper = get_batched_stepper(model, ctx_pad.float(), ctx_mask_s,
+                                      pctx_pad.float(), S * k)
+
     NEG_UNK = math.log(1e-20)
 
     for ii in range(max(maxlens)):
@@ -102,13 +111,21 @@ def gen_sample_batched(model, xs, k=1, maxlen=30, use_unk=False,
         state = torch.cat([dstate[i]["state"] for i in alive])
         acc_c = torch.cat([dstate[i]["acc_c"] for i in alive])
         acc_a = torch.cat([dstate[i]["acc_a"] for i in alive])
-        ctx_b = ctx_pad[:, sent_idx]
-        cmask_b = ctx_mask_s[:, sent_idx]
-        pctx_b = pctx_pad[:, sent_idx]
+        if stepper is None:
+            ctx_b = ctx_pad[:, sent_idx]
+            cmask_b = ctx_mask_s[:, sent_idx]
+            pctx_b = pctx_pad[:, sent_idx]
 
-        probs, _, h2, alpha, ctx_t, acc_c, acc_a = model.f_next(
-            y, ctx_b, cmask_b, pctx_b, state, acc_c, acc_a,
-            sample_draw=False)
+        if stepper is not None:
+            rows = y.shape[0]
+            outs = stepper.step(sent_idx, y, state.float(), acc_c.float(),
+                                acc_a.float())
+            probs, h2, alpha, ctx_t, acc_c, acc_a = [
+                o[:rows] for o in outs]
+        else:
+            probs, _, h2, alpha, ctx_t, acc_c, acc_a = model.f_next(
+                y, ctx_b, cmask_b, pctx_b, state, acc_c, acc_a,
+                sample_draw=False)
         V = probs.shape[1]
 
         # ---- phase 1: per-sentence top-k ON DEVICE, one host transfer ----

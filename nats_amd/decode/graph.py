@@ -50,6 +50,102 @@ def get_stepper(model, ctx0, pctx0, k):
     return st
 
 
+def get_batched_stepper(model, ctx_pad, ctx_mask, pctx_pad, R):
+    """Cached BatchedGraphStepper per (model, Ts, S, R) — reused across
+    same-shaped micro-batches (serving); invalidated on weight updates."""
+    ver = _weights_version(model)
+    per_model = _STEPPER_CACHE.setdefault(model, {})
+    if per_model.get("__ver__") != ver:
+        per_model.clear()
+        per_model["__ver__"] = ver
+    key = ("batched", int(ctx_pad.shape[0]), int(ctx_pad.shape[1]), int(R))
+    st = per_model.get(key)
+    if st is None:
+        st = BatchedGraphStepper(model, ctx_pad, ctx_mask, pctx_pad, R)
+        per_model[key] = st
+    else:
+        st.set_context(ctx_pad, ctx_mask, pctx_pad)
+    return st
+
+
+class BatchedGraphStepper:
+    """hipGraph-captured f_next for the batched multi-sentence beam
+    (decode/batched.py — the production serving path).
+
+    Static row budget R (= S*k): live rows occupy a prefix, dead rows are
+    padded with row 0's state (harmless — callers slice [:rows]). The
+    per-step context gather ctx_pad[:, sent_idx] happens INSIDE the
+    graph from a static sent_idx buffer, so hypothesis->sentence routing
+    changes never force recapture."""
+
+    def __init__(self, model, ctx_pad, ctx_mask, pctx_pad, R):
+        self.model = model
+        self.R = R
+        device = ctx_pad.device
+        Ts, S, C = ctx_pad.shape
+        H = model.options["dim"]
+        # own the static buffers (a caller's tensor may alias freed memory
+        # by the next micro-batch)
+        self.ctx_pad = ctx_pad.contiguous().clone()
+        self.ctx_mask = ctx_mask.float().contiguous().clone()
+        self.pctx_pad = pctx_pad.contiguous().clone()
+        self.sent_idx = torch.zeros(R, dtype=torch.int64, device=device)
+        self.y_in = torch.zeros(R, dtype=torch.int64, device=device)
+        self.state_in = torch.zeros(R, H, device=device)
+        self.accC_in = torch.zeros(R, C, device=device)
+        self.accA_in = torch.zeros(R, Ts, device=device)
+        self.graph = None
+        self.outs = None
+
+    def set_context(self, ctx_pad, ctx_mask, pctx_pad):
+        self.ctx_pad.copy_(ctx_pad)
+        self.ctx_mask.copy_(ctx_mask)
+        self.pctx_pad.copy_(pctx_pad)
+
+    def _run(self):
+        ctx_b = self.ctx_pad.index_select(1, self.sent_idx)
+        cmask_b = self.ctx_mask.index_select(1, self.sent_idx)
+        pctx_b = self.pctx_pad.index_select(1, self.sent_idx)
+        return self.model.f_next(self.y_in, ctx_b, cmask_b, pctx_b,
+                                 self.state_in, self.accC_in, self.accA_in,
+                                 sample_draw=False)
+
+    def capture(self):
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._run()
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            outs = self._run()
+            self.outs = tuple(o.clone() if isinstance(o, torch.Tensor) else o
+                              for o in outs)
+        return self
+
+    def step(self, sent_idx, y, state, acc_ctx, acc_alpha):
+        """rows = len(y) (<= R) live hypothesis rows; returns (probs, h2,
+        alpha, ctx_t, accC, accA), each R rows — callers slice [:rows]."""
+        rows = y.shape[0]
+        self.sent_idx[:rows] = sent_idx
+        self.y_in[:rows] = y
+        self.state_in[:rows] = state
+        self.accC_in[:rows] = acc_ctx
+        self.accA_in[:rows] = acc_alpha
+        if rows < self.R:
+            self.sent_idx[rows:] = sent_idx[0]
+            self.y_in[rows:] = y[0]
+            self.state_in[rows:] = state[0]
+            self.accC_in[rows:] = acc_ctx[0]
+            self.accA_in[rows:] = acc_alpha[0]
+        if self.graph is None:
+            self.capture()
+        self.graph.replay()
+        probs, _, h2, alpha, ctx_t, accC, accA = self.outs
+        return probs, h2, alpha, ctx_t, accC, accA
+
+
 class GraphDecodeStepper:
     """Replayable f_next for beam search (argmax/beam mode, no sampling)."""
 
