@@ -148,7 +148,9 @@ def embed_gather(Wemb, ids, shift=False):
     shift-right-by-one (BOS row = zeros). Semantics: nats.py:700-701
     (plain) and 730-734 (shifted target embedding).
     """
-    if Wemb.dtype == torch.float32 and _use_hip(Wemb):
+    if (Wemb.dtype == torch.float32
+            and not os.environ.get("NATS_EMBED_EAGER")
+            and _use_hip(Wemb)):
         from .embed import embed_gather_hip
         return embed_gather_hip(Wemb, ids, shift=shift)
     emb = Wemb[ids.clamp_min(0)]

@@ -621,45 +621,59 @@ __global__ __launch_bounds__(256) void cond_attn_bwd_dalpha(
   const int wave = threadIdx.x / NATS_WAVE;
   const int lane = threadIdx.x & (NATS_WAVE - 1);
   const int s = blockIdx.y * 4 + wave;
-  // softmax-backward dot: block-local tree over the 4 waves' (alpha*dal),
-  // ONE atomicAdd per block into dot_buf[b] (~Ts/4 adds spread over B
-  // addresses — the per-s variant serialized ~Ts adds on one address at
-  // ~120us/step, and a separate cond_attn_bwd_dot pass cost a full
-  // launch + Ts*B re-read; this fold removes that kernel).
-  __shared__ float wdot[4];
-  float dal = 0.f;
-  if (s < Ts) {
-    const bf16_t* crow = ctx_bf + ((long)s * B + b) * C;
-    const float* drow = dctxpre_f32 + (long)b * C;
-    float part = 0.f;
-    const int C8 = C & ~7;
-    for (int c = lane * 8; c < C8; c += NATS_WAVE * 8) {
-      bf16x8 v = *(const bf16x8*)(crow + c);
-      const float4 d0 = *(const float4*)(drow + c);
-      const float4 d1 = *(const float4*)(drow + c + 4);
-      part += (float)v[0] * d0.x + (float)v[1] * d0.y + (float)v[2] * d0.z +
-              (float)v[3] * d0.w + (float)v[4] * d1.x + (float)v[5] * d1.y +
-              (float)v[6] * d1.z + (float)v[7] * d1.w;
-    }
-    for (int c = C8 + lane; c < C; c += NATS_WAVE) {
-      part += (float)crow[c] * drow[c];
-    }
-#pragma unroll
-    for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
-      part += __shfl_down(part, off);
-    if (lane == 0) {
-      dal = part;
-      const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
-      dal += mm * daccA[(long)b * Ts + s];
-      if (dalphas_t != nullptr) dal += dalphas_t[(long)b * Ts + s];
-      dal_buf[(long)s * B + b] = dal;
-    }
+  if (s >= Ts) return;
+  const bf16_t* crow = ctx_bf + ((long)s * B + b) * C;
+  const float* drow = dctxpre_f32 + (long)b * C;
+  float part = 0.f;
+  const int C8 = C & ~7;
+  for (int c = lane * 8; c < C8; c += NATS_WAVE * 8) {
+    bf16x8 v = *(const bf16x8*)(crow + c);
+    const float4 d0 = *(const float4*)(drow + c);
+    const float4 d1 = *(const float4*)(drow + c + 4);
+    part += (float)v[0] * d0.x + (float)v[1] * d0.y + (float)v[2] * d0.z +
+            (float)v[3] * d0.w + (float)v[4] * d1.x + (float)v[5] * d1.y +
+            (float)v[6] * d1.z + (float)v[7] * d1.w;
   }
-  if (lane == 0)
-    wdot[wave] = (s < Ts) ? alphas_t[(long)b * Ts + s] * dal : 0.f;
+  for (int c = C8 + lane; c < C; c += NATS_WAVE) {
+    part += (float)crow[c] * drow[c];
+  }
+#pragma unroll
+  for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
+    part += __shfl_down(part, off);
+  if (lane == 0) {
+    float dal = part;
+    const float mm = (mask_t != nullptr) ? mask_t[b] : 1.f;
+    dal += mm * daccA[(long)b * Ts + s];
+    if (dalphas_t != nullptr) dal += dalphas_t[(long)b * Ts + s];
+    dal_buf[(long)s * B + b] = dal;
+    // NOTE: the dot(alpha, dal) fold was MEASURED OUT here twice — a
+    // per-s atomicAdd serializes ~Ts adds on one address (~120us/step,
+    // round 1), and a per-BLOCK atomic tree still serialized Ts/4 adds
+    // per address (dalpha 9us -> 47.9us/call, prof_r2). The separate
+    // cond_attn_bwd_dot pass (~3us) stays.
+  }
+}
+
+// softmax-backward dot: dot_buf[b] = sum_s alpha[b,s] * dal[s,b]
+__global__ __launch_bounds__(256) void cond_attn_bwd_dot(
+    const float* __restrict__ alphas_t, const float* __restrict__ dal_buf,
+    float* __restrict__ dot_buf, int B, int Ts) {
+  const int b = blockIdx.x;
+  __shared__ float red[256 / NATS_WAVE];
+  float part = 0.f;
+  for (int s = threadIdx.x; s < Ts; s += blockDim.x)
+    part += alphas_t[(long)b * Ts + s] * dal_buf[(long)s * B + b];
+#pragma unroll
+  for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
+    part += __shfl_down(part, off);
+  if ((threadIdx.x & (NATS_WAVE - 1)) == 0)
+    red[threadIdx.x / NATS_WAVE] = part;
   __syncthreads();
-  if (threadIdx.x == 0)
-    atomicAdd(dot_buf + b, wdot[0] + wdot[1] + wdot[2] + wdot[3]);
+  if (threadIdx.x == 0) {
+    float S = 0.f;
+    for (int w = 0; w < (int)blockDim.x / NATS_WAVE; ++w) S += red[w];
+    dot_buf[b] = S;
+  }
 }
 
 // attention backward, stage 2 (grid (B, ceil(Ts/256))): softmax backward,
@@ -1128,6 +1142,10 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        dalpha_p ? dalpha_p + (long)t * B * Ts : nullptr,
                        dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
                        B, Ts, C);
+    hipLaunchKernelGGL(cond_attn_bwd_dot, dim3(B), dim3(256), 0, stream,
+                       alphas_all.data_ptr<float>() + (long)t * B * Ts,
+                       dal_buf.data_ptr<float>(), dot_buf.data_ptr<float>(),
+                       B, Ts);
     hipLaunchKernelGGL(cond_attn_bwd_scatter,
                        dim3(B, cdiv_i(Ts, 256), A >= 32 ? 8 : (A >= 16 ? 4 : 1)),
                        dim3(256), 0, stream,
