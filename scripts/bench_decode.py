@@ -30,7 +30,7 @@ def run(model, xs, k, maxlen, use_graph, lam, batched=False):
             outs = gen_sample_batched(model, xs[base:base + sb], k=k,
                                       maxlen=maxlen, use_unk=True,
                                       kl_factor=lam, ctx_factor=lam,
-                                      state_factor=lam)
+                                      state_factor=lam, use_graph=use_graph)
             for sample, _, _ in outs:
                 n_tokens += sum(len(s) for s in sample)
     else:
@@ -72,6 +72,9 @@ def main():
                 ("beam+distraction+batched", False, 0.5, True)]
     if device == "cuda":
         variants.append(("beam+hipgraph", True, 0.0, False))
+        variants.append(("beam+batched+hipgraph", True, 0.0, True))
+        variants.append(("beam+distraction+batched+hipgraph", True, 0.5,
+                         True))
     for name, graph, lam, batched in variants:
         run(model, xs[:2], args.k, args.maxlen, graph, lam, batched)
         sps, ntok = run(model, xs, args.k, args.maxlen, graph, lam, batched)
