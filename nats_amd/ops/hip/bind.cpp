@@ -28,6 +28,7 @@ torch::Tensor softmax_ce_bwd(torch::Tensor logits, torch::Tensor targets,
                              torch::Tensor stats, torch::Tensor dnll);
 torch::Tensor mfma_gemm_bt(torch::Tensor A, torch::Tensor Bt);
 double barrier_bench(int nwg_x, int nwg_y, int iters);
+double barrier_bench_xcd(int nwg, int nxcd, int iters);
 torch::Tensor rerank_penalties(torch::Tensor hist_a, torch::Tensor hist_c,
                                torch::Tensor hist_s, torch::Tensor cur_a,
                                torch::Tensor cur_c, torch::Tensor cur_s,
@@ -79,6 +80,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_ce_bwd", &softmax_ce_bwd, "fused softmax+CE backward");
   m.def("mfma_gemm_bt", &mfma_gemm_bt, "MFMA layout self-test GEMM");
   m.def("barrier_bench", &barrier_bench, "grid barrier us/iteration");
+  m.def("barrier_bench_xcd", &barrier_bench_xcd,
+        "XCD-constrained grid barrier us/iteration");
   m.def("rerank_penalties", &rerank_penalties,
         "fused decode-time distraction rerank reductions");
   m.def("fused_adadelta_step", &fused_adadelta_step,
