@@ -72,10 +72,27 @@ class GraphedTrainStep:
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
             self.cost = self._full_step()
+        # self-check: one replay must MOVE the parameters (a capture that
+        # silently drops the backward/optimizer dependency would
+        # otherwise train nothing while reporting healthy throughput)
+        with torch.no_grad():
+            csum0 = float(torch.stack(
+                [p.float().abs().sum() for p in
+                 self.model.parameters()]).sum())
+        self.graph.replay()
+        torch.cuda.synchronize()
+        with torch.no_grad():
+            csum1 = float(torch.stack(
+                [p.float().abs().sum() for p in
+                 self.model.parameters()]).sum())
         with torch.no_grad():
             for p, snap in zip(self.model.parameters(), param_snap):
                 p.copy_(snap)
         self.opt.load_state_dict(opt_snap)
+        if csum1 == csum0:
+            raise RuntimeError(
+                "graph replay left parameters unchanged — capture did not "
+                "wire the backward/optimizer update")
 
     def step(self, x, x_mask, y, y_mask):
         """Copy inputs into the static buffers and replay. Returns the

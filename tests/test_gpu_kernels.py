@@ -34,13 +34,15 @@ def assert_grad_close(a, b, name="", rel=0.06, floor_frac=1e-3):
     col_err = (a2 - b2).abs().amax(dim=0)
     bound = rel * col_scale + floor
     bad = col_err > bound
-    assert not bool(bad.any()), (
-        "%s: %d/%d columns out of bound; worst col err %.3e vs bound %.3e "
-        "(col scale %.3e, floor %.3e)" % (
-            name, int(bad.sum()), bad.numel(),
-            float(col_err[bad].max()) if bad.any() else 0.0,
-            float(bound[bad].min()) if bad.any() else 0.0,
-            float(col_scale[bad.argmax()]), float(floor)))
+    if bool(bad.any()):
+        worst = int((col_err - bound)[bad].argmax())
+        idx = bad.nonzero().flatten()[worst]
+        raise AssertionError(
+            "%s: %d/%d columns out of bound; worst col %d err %.3e vs "
+            "bound %.3e (col scale %.3e, floor %.3e)" % (
+                name, int(bad.sum()), bad.numel(), int(idx),
+                float(col_err[idx]), float(bound[idx]),
+                float(col_scale[idx]), float(floor)))
 
 
 def test_mfma_fragment_layout(ext):
