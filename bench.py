@@ -104,11 +104,16 @@ def main():
     bd = {"fwd": 0.0, "bwd": 0.0, "opt": 0.0, "n": 0}
 
     # whole-step hipGraph capture (one replay per step instead of ~2000
-    # launches); falls back to eager if capture fails
+    # launches); self-checked at capture, falls back to eager on failure.
+    # world==1 only: collective capture over RCCL is unproven on this
+    # stack and a capture hang would take the whole scaling run with it.
+    # Measured effect at the CNN/DM shape: neutral (41.8 vs 41.9 ms) —
+    # the step is kernel-bound; kept for the launch-overhead headroom.
     graph_step = None
-    if use_cuda and not args.no_graph and not breakdown and not args.eager:
+    if (use_cuda and world == 1 and not args.no_graph and not breakdown
+            and not args.eager):
         from nats_amd.utils.step_graph import GraphedStepCache
-        gcache = GraphedStepCache(model, opt, dp if world > 1 else None)
+        gcache = GraphedStepCache(model, opt, None)
         graph_step = gcache.get(x, x_mask, y, y_mask)
 
     def step():

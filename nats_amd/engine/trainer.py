@@ -145,7 +145,7 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
     # decay_c — the weight-decay term is added outside the closure — and
     # pointless with the per-section profiler's syncs)
     gcache = None
-    use_graph = step_graph if step_graph is not None else True
+    use_graph = step_graph if step_graph is not None else (world == 1)
     if (use_graph and str(device).startswith("cuda") and decay_c == 0.0
             and not profile):
         from ..utils.step_graph import GraphedStepCache

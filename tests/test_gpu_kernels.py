@@ -16,7 +16,8 @@ def ext():
     return m
 
 
-def assert_grad_close(a, b, name="", rel=0.06, floor_frac=1e-3):
+def assert_grad_close(a, b, name="", rel=0.06, floor_frac=1e-3,
+                      global_scale=None):
     """Scale-aware gradient comparison.
 
     Global-max normalization (rel = max|a-b| / max|a|) lets a large
@@ -30,7 +31,14 @@ def assert_grad_close(a, b, name="", rel=0.06, floor_frac=1e-3):
     a2 = a.reshape(-1, a.shape[-1]) if a.dim() > 1 else a.reshape(1, -1)
     b2 = b.reshape(-1, b.shape[-1]) if b.dim() > 1 else b.reshape(1, -1)
     col_scale = a2.abs().amax(dim=0)
-    floor = floor_frac * a2.abs().max().clamp_min(1e-6)
+    # global_scale: the MODEL's gradient magnitude — reduce-to-scalar
+    # grads (c_att = sum of thousands of cancelling de terms) carry fp32
+    # accumulation noise proportional to the contributions, not to their
+    # near-zero sum; floor such tensors at 1e-3 of the model scale
+    floor_base = a2.abs().max().clamp_min(1e-6)
+    if global_scale is not None:
+        floor_base = max(float(floor_base), 1e-2 * float(global_scale))
+    floor = floor_frac * floor_base
     col_err = (a2 - b2).abs().amax(dim=0)
     bound = rel * col_scale + floor
     bad = col_err > bound
@@ -264,10 +272,11 @@ def test_cond_gru_backward(ext):
     rg, ri = run("cpu", eager.cond_gru_scan)
     hg, hi = run("cuda", cond_gru_scan_hip)
 
+    gscale = max(float(v.abs().max()) for v in rg.values())
     for k in rg:
-        assert_grad_close(rg[k], hg[k], k, rel=0.08)
+        assert_grad_close(rg[k], hg[k], k, rel=0.08, global_scale=gscale)
     for k in ri:
-        assert_grad_close(ri[k], hi[k], k, rel=0.08)
+        assert_grad_close(ri[k], hi[k], k, rel=0.08, global_scale=gscale)
 
 
 def test_cond_gru_one_step(ext):
