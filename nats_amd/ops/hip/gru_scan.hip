@@ -550,20 +550,61 @@ struct GruPersistFwd {
   const float* h0;     // [B][H]
 };
 
+// XCD-preferred direction claim (profiles/barrier_xcd.json: a 63-WG
+// barrier costs 5.97us with blocks spread over all 8 XCDs but 3.66us
+// confined to 4 — and confinement also keeps each direction's h-exchange
+// lines in fewer L2s). The grid is overprovisioned (gridDim.x = 256, all
+// resident at <=150KB LDS); each block claims a (direction, column-tile)
+// slot with direction 0 preferred on XCDs 0-3 and direction 1 on 4-7; a
+// block whose preferred side is full backs off briefly, then steals from
+// the other side (correct under any placement — the barrier census is
+// placement-independent); unclaimed blocks exit. Greedy claiming has NO
+// census spin, so there is no new hang mode.
+__device__ __forceinline__ int nats_claim_dir_slot(unsigned* claim,
+                                                   int ngrp) {
+  __shared__ int sh_sel;
+  if (threadIdx.x == 0) {
+    const unsigned xcc = nats_xcc_id();
+    int dir = (xcc < 4) ? 0 : 1;
+    int slot = -1;
+    const unsigned s0 = __hip_atomic_fetch_add(
+        claim + dir, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    if (s0 < (unsigned)ngrp) {
+      slot = (int)s0;
+    } else {
+      // preferred side full: let the other side's own blocks claim first
+      for (int i = 0; i < 6; ++i) __builtin_amdgcn_s_sleep(127);
+      const int od = 1 - dir;
+      const unsigned s1 = __hip_atomic_fetch_add(
+          claim + od, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      if (s1 < (unsigned)ngrp) {
+        dir = od;
+        slot = (int)s1;
+      }
+    }
+    sh_sel = (slot < 0) ? -1 : (dir << 16 | slot);
+  }
+  __syncthreads();
+  return sh_sel;
+}
+
 __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
     GruPersistFwd p0, GruPersistFwd p1, int T, int B, int H, int Hpad,
-    unsigned* sync, unsigned nwg, int unsafe_nobarrier) {
+    unsigned* sync, unsigned nwg, int ngrp, int unsafe_nobarrier) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* upk_lds = (bf16_t*)smem;                       // [3*16][Hpad] swz
   float(*pre)[32][JB + 1] =
       (float(*)[32][JB + 1])(smem + (long)3 * JB * Hpad * 2);
 
-  const GruPersistFwd& p = (blockIdx.y == 0) ? p0 : p1;
-  const int wg = blockIdx.x;
+  const int sel = nats_claim_dir_slot(sync + 2 * NATS_SYNC_WORDS, ngrp);
+  if (sel < 0) return;
+  const int dirsel = sel >> 16;
+  const int wg = sel & 0xffff;
+  const GruPersistFwd& p = (dirsel == 0) ? p0 : p1;
   // per-direction barrier: the two directions are data-independent, so
-  // each grid.y half syncs only its own gridDim.x blocks (own sync slab)
-  sync += (long)blockIdx.y * NATS_SYNC_WORDS;
-  nwg = gridDim.x;
+  // each direction syncs only its own ngrp blocks (own sync slab)
+  sync += (long)dirsel * NATS_SYNC_WORDS;
+  nwg = (unsigned)ngrp;
   stage_weights_lds(upk_lds, p.Upk + (long)wg * 3 * JB * Hpad, 3 * JB, Hpad);
   __syncthreads();
   NatsBarrierCtx bctx;
@@ -705,16 +746,19 @@ struct GruPersistBwd {
 
 __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
     GruPersistBwd p0, GruPersistBwd p1, int T, int B, int H, int K3pad,
-    unsigned* sync, unsigned nwg) {
+    unsigned* sync, unsigned nwg, int ngrp) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* ub_lds = (bf16_t*)smem;  // [16][K3pad] swizzled
   float(*part)[32][JB + 1] =
       (float(*)[32][JB + 1])(smem + (long)JB * K3pad * 2);
 
-  const GruPersistBwd& p = (blockIdx.y == 0) ? p0 : p1;
-  const int wg = blockIdx.x;
-  sync += (long)blockIdx.y * NATS_SYNC_WORDS;
-  nwg = gridDim.x;
+  const int sel = nats_claim_dir_slot(sync + 2 * NATS_SYNC_WORDS, ngrp);
+  if (sel < 0) return;
+  const int dirsel = sel >> 16;
+  const int wg = sel & 0xffff;
+  const GruPersistBwd& p = (dirsel == 0) ? p0 : p1;
+  sync += (long)dirsel * NATS_SYNC_WORDS;
+  nwg = (unsigned)ngrp;
   stage_weights_lds(ub_lds, p.Ubwd + (long)wg * JB * K3pad, JB, K3pad);
   __syncthreads();
   NatsBarrierCtx bctx;
@@ -1048,7 +1092,8 @@ std::vector<torch::Tensor> gru_scan_fwd_bidir(
   const bool persistent = (2 * ngrp <= 192) && (smem_fwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
-    auto sync = torch::zeros({2 * NATS_SYNC_WORDS}, xg0.options().dtype(torch::kInt32));
+    auto sync = torch::zeros({2 * NATS_SYNC_WORDS + 4},
+                             xg0.options().dtype(torch::kInt32));
     unsigned* sync_p = (unsigned*)sync.data_ptr<int>();
     GruPersistFwd p0{(const bf16_t*)xg0.data_ptr(),
                      (const bf16_t*)xc0.data_ptr(),
@@ -1067,9 +1112,12 @@ std::vector<torch::Tensor> gru_scan_fwd_bidir(
                      (bf16_t*)saved1.data_ptr(),
                      h00.data_ptr<float>()};
     const int unsafe = getenv("NATS_UNSAFE_NOBARRIER") != nullptr;
-    hipLaunchKernelGGL(nats_gru_persistent_fwd, dim3(ngrp, 2), dim3(384),
+    // overprovisioned 1-D grid: blocks self-select a (direction, tile)
+    // slot with XCD preference (see nats_claim_dir_slot); all 256 fit
+    // resident (smem gate is 150KB -> >=1 block/CU)
+    hipLaunchKernelGGL(nats_gru_persistent_fwd, dim3(256), dim3(384),
                        smem_fwd, stream, p0, p1, T, B, H, Hpad, sync_p,
-                       (unsigned)(2 * ngrp), unsafe);
+                       (unsigned)(2 * ngrp), ngrp, unsafe);
     HIP_CHECK(hipGetLastError());
     return {h_all0, saved0, h_all1, saved1};
   }
@@ -1150,7 +1198,8 @@ std::vector<torch::Tensor> gru_scan_bwd_bidir(
   const bool persistent = (2 * ngrp <= 192) && (smem_bwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
-    auto sync = torch::zeros({2 * NATS_SYNC_WORDS}, dh_out0.options().dtype(torch::kInt32));
+    auto sync = torch::zeros({2 * NATS_SYNC_WORDS + 4},
+                             dh_out0.options().dtype(torch::kInt32));
     unsigned* sync_p = (unsigned*)sync.data_ptr<int>();
     GruPersistBwd p0{dh0c.data_ptr<float>(),
                      h_all0.data_ptr<float>(),
@@ -1172,9 +1221,9 @@ std::vector<torch::Tensor> gru_scan_bwd_bidir(
                      dd + (long)B * H,
                      (bf16_t*)dpre1.data_ptr(),
                      h00.data_ptr<float>()};
-    hipLaunchKernelGGL(nats_gru_persistent_bwd, dim3(ngrp, 2), dim3(384),
+    hipLaunchKernelGGL(nats_gru_persistent_bwd, dim3(256), dim3(384),
                        smem_bwd, stream, p0, p1, T, B, H, K3pad, sync_p,
-                       (unsigned)(2 * ngrp));
+                       (unsigned)(2 * ngrp), ngrp);
     // final dh0 per direction (dstep(0) lives in ping-pong slot 0)
     hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrp), dim3(384), 0,
                        stream, dsp + 0 * 2 * ds,
