@@ -561,25 +561,33 @@ struct GruPersistFwd {
 // placement-independent); unclaimed blocks exit. Greedy claiming has NO
 // census spin, so there is no new hang mode.
 __device__ __forceinline__ int nats_claim_dir_slot(unsigned* claim,
-                                                   int ngrp) {
+                                                   int ngrp, int xpd) {
   __shared__ int sh_sel;
   if (threadIdx.x == 0) {
     const unsigned xcc = nats_xcc_id();
-    int dir = (xcc < 4) ? 0 : 1;
+    // xpd = preferred XCDs per direction (4: halves of the chip; 2:
+    // tighter locality, XCDs 0-1 / 2-3 with 4-7 unpreferred)
+    int dir = (xcc < (unsigned)xpd) ? 0
+              : (xcc < (unsigned)(2 * xpd)) ? 1 : -1;
     int slot = -1;
-    const unsigned s0 = __hip_atomic_fetch_add(
-        claim + dir, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    if (s0 < (unsigned)ngrp) {
-      slot = (int)s0;
-    } else {
-      // preferred side full: let the other side's own blocks claim first
+    if (dir >= 0) {
+      const unsigned s0 = __hip_atomic_fetch_add(
+          claim + dir, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      if (s0 < (unsigned)ngrp) slot = (int)s0;
+    }
+    if (slot < 0) {
+      // unpreferred block (or preferred side full): let the preferred
+      // blocks claim first, then fill leftovers on either side
       for (int i = 0; i < 6; ++i) __builtin_amdgcn_s_sleep(127);
-      const int od = 1 - dir;
-      const unsigned s1 = __hip_atomic_fetch_add(
-          claim + od, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-      if (s1 < (unsigned)ngrp) {
-        dir = od;
-        slot = (int)s1;
+      const int first = (dir >= 0) ? 1 - dir : (int)(xcc & 1);
+      for (int k = 0; k < 2 && slot < 0; ++k) {
+        const int d2 = k == 0 ? first : 1 - first;
+        const unsigned s1 = __hip_atomic_fetch_add(
+            claim + d2, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        if (s1 < (unsigned)ngrp) {
+          dir = d2;
+          slot = (int)s1;
+        }
       }
     }
     sh_sel = (slot < 0) ? -1 : (dir << 16 | slot);
@@ -590,13 +598,15 @@ __device__ __forceinline__ int nats_claim_dir_slot(unsigned* claim,
 
 __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
     GruPersistFwd p0, GruPersistFwd p1, int T, int B, int H, int Hpad,
-    unsigned* sync, unsigned nwg, int ngrp, int unsafe_nobarrier) {
+    unsigned* sync, unsigned nwg, int ngrp, int xpd,
+    int unsafe_nobarrier) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* upk_lds = (bf16_t*)smem;                       // [3*16][Hpad] swz
   float(*pre)[32][JB + 1] =
       (float(*)[32][JB + 1])(smem + (long)3 * JB * Hpad * 2);
 
-  const int sel = nats_claim_dir_slot(sync + 2 * NATS_SYNC_WORDS, ngrp);
+  const int sel = nats_claim_dir_slot(sync + 2 * NATS_SYNC_WORDS, ngrp,
+                                      xpd);
   if (sel < 0) return;
   const int dirsel = sel >> 16;
   const int wg = sel & 0xffff;
@@ -746,13 +756,14 @@ struct GruPersistBwd {
 
 __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
     GruPersistBwd p0, GruPersistBwd p1, int T, int B, int H, int K3pad,
-    unsigned* sync, unsigned nwg, int ngrp) {
+    unsigned* sync, unsigned nwg, int ngrp, int xpd) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* ub_lds = (bf16_t*)smem;  // [16][K3pad] swizzled
   float(*part)[32][JB + 1] =
       (float(*)[32][JB + 1])(smem + (long)JB * K3pad * 2);
 
-  const int sel = nats_claim_dir_slot(sync + 2 * NATS_SYNC_WORDS, ngrp);
+  const int sel = nats_claim_dir_slot(sync + 2 * NATS_SYNC_WORDS, ngrp,
+                                      xpd);
   if (sel < 0) return;
   const int dirsel = sel >> 16;
   const int wg = sel & 0xffff;
@@ -1115,9 +1126,11 @@ std::vector<torch::Tensor> gru_scan_fwd_bidir(
     // overprovisioned 1-D grid: blocks self-select a (direction, tile)
     // slot with XCD preference (see nats_claim_dir_slot); all 256 fit
     // resident (smem gate is 150KB -> >=1 block/CU)
+    const char* xpd_env = getenv("NATS_XPD");
+    const int xpd = xpd_env ? atoi(xpd_env) : 4;
     hipLaunchKernelGGL(nats_gru_persistent_fwd, dim3(256), dim3(384),
                        smem_fwd, stream, p0, p1, T, B, H, Hpad, sync_p,
-                       (unsigned)(2 * ngrp), ngrp, unsafe);
+                       (unsigned)(2 * ngrp), ngrp, xpd, unsafe);
     HIP_CHECK(hipGetLastError());
     return {h_all0, saved0, h_all1, saved1};
   }
@@ -1221,9 +1234,11 @@ std::vector<torch::Tensor> gru_scan_bwd_bidir(
                      dd + (long)B * H,
                      (bf16_t*)dpre1.data_ptr(),
                      h00.data_ptr<float>()};
+    const char* xpd_env = getenv("NATS_XPD");
+    const int xpd = xpd_env ? atoi(xpd_env) : 4;
     hipLaunchKernelGGL(nats_gru_persistent_bwd, dim3(256), dim3(384),
                        smem_bwd, stream, p0, p1, T, B, H, K3pad, sync_p,
-                       (unsigned)(2 * ngrp), ngrp);
+                       (unsigned)(2 * ngrp), ngrp, xpd);
     // final dh0 per direction (dstep(0) lives in ping-pong slot 0)
     hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrp), dim3(384), 0,
                        stream, dsp + 0 * 2 * ds,
