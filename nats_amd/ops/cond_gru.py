@@ -21,13 +21,24 @@ def _ceil(x, m):
 
 def _pack_rows(mat, out_rows, out_cols):
     """[rows,K] -> zero-padded [out_rows, out_cols] bf16 contiguous."""
+    if mat.is_cuda and mat.dtype == torch.float32:
+        ext = _hip_ext()
+        if ext is not None:
+            # fused pad(+transpose) kernel; .t() views pack without copy
+            if mat.dim() == 2 and not mat.is_contiguous() and                     mat.t().is_contiguous():
+                return ext.pack_pad(mat.t(), out_rows, out_cols, True)
+            return ext.pack_pad(mat.contiguous(), out_rows, out_cols, False)
     return _pad_to(mat, out_rows, out_cols).to(torch.bfloat16).contiguous()
 
 
 def pack_gru1_weights(U_1, W_1, Ux_1, Wx_1, Hpad, Cpad):
     """[ngrp*4*16, K1] with per-group rows [r2|u2|pxa|pxb]; K layout
     [h1-block (Hpad) | ctx-block (Cpad)], zero where an output group does
-    not consume that operand block."""
+    not consume that operand block. One fused kernel on GPU."""
+    if U_1.is_cuda and U_1.dtype == torch.float32:
+        ext = _hip_ext()
+        if ext is not None:
+            return ext.pack_gru1_weights(U_1, W_1, Ux_1, Wx_1, Hpad, Cpad)
     H = Ux_1.shape[0]
     C = W_1.shape[0]
     ngrp = (H + JB - 1) // JB
@@ -101,8 +112,14 @@ class CondGRUScanFn(torch.autograd.Function):
         U1cat = pack_bwd_weights(U_1.float(), Ux_1.float())
         U2cat = pack_bwd_weights(U.float(), Ux.float())
         K3Hpad = U1cat.shape[1]
-        W1cat = _pack_rows(torch.cat([W_1.float(), Wx_1.float()], dim=1),
-                           _ceil(C, 16), K3Hpad)
+        ext2 = _hip_ext()
+        if W_1.is_cuda and ext2 is not None:
+            W1cat = ext2.pack_cat2(W_1.float(), Wx_1.float(),
+                                   _ceil(C, 16), K3Hpad)
+        else:
+            W1cat = _pack_rows(
+                torch.cat([W_1.float(), Wx_1.float()], dim=1),
+                _ceil(C, 16), K3Hpad)
         Apad32 = _ceil(A, 32)
         WattB = _pack_rows(W_att.float(), _ceil(H, 16), Apad32)
 

@@ -22,7 +22,14 @@ def _pad_to(x, rows, cols):
 
 def pack_fwd_weights(U, Ux):
     """[ngrp*3*16, Hpad] bf16: per group g rows = [r-cols | u-cols | x-cols]
-    of output columns [g*16,(g+1)*16), transposed (row = output col)."""
+    of output columns [g*16,(g+1)*16), transposed (row = output col).
+    On GPU one fused kernel (ops/hip/pack.hip) replaces the 6-op torch
+    chain (pad x3, stack, permute-contiguous, cast) — these packs run
+    every training step (weights change every update)."""
+    if U.is_cuda and U.dtype == torch.float32:
+        ext = _hip_ext()
+        if ext is not None:
+            return ext.pack_fwd_weights(U, Ux)
     H = Ux.shape[1]
     ngrp = (H + JB - 1) // JB
     rows = ngrp * JB
@@ -42,6 +49,10 @@ def pack_bwd_weights(U, Ux):
     rows = ngrp * JB
     K3 = 3 * H
     K3pad = ((K3 + 31) // 32) * 32
+    if U.is_cuda and U.dtype == torch.float32:
+        ext = _hip_ext()
+        if ext is not None:
+            return ext.pack_cat2(U, Ux, rows, K3pad)
     cat = torch.cat([U, Ux], dim=1)                   # (H, 3H)
     return _pad_to(cat, rows, K3pad).to(torch.bfloat16).contiguous()
 

@@ -62,10 +62,22 @@ std::vector<torch::Tensor> cond_gru_bwd(
 
 torch::Tensor embed_gather(torch::Tensor Wemb, torch::Tensor ids,
                            long shift_rows);
+torch::Tensor pack_fwd_weights_hip(torch::Tensor U, torch::Tensor Ux);
+torch::Tensor pack_cat2_hip(torch::Tensor A, torch::Tensor B, long R, long K);
+torch::Tensor pack_pad_hip(torch::Tensor src, long R, long K, bool transpose);
+torch::Tensor pack_gru1_weights_hip(torch::Tensor U_1, torch::Tensor W_1,
+                                    torch::Tensor Ux_1, torch::Tensor Wx_1,
+                                    long Hpad, long Cpad);
 torch::Tensor embed_scatter_add(torch::Tensor dout, torch::Tensor ids,
                                 long V, long shift_rows);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("pack_fwd_weights", &pack_fwd_weights_hip,
+        "fused forward-scan weight pack");
+  m.def("pack_cat2", &pack_cat2_hip, "fused [A|B] zero-padded bf16 pack");
+  m.def("pack_pad", &pack_pad_hip, "fused pad(+transpose) bf16 pack");
+  m.def("pack_gru1_weights", &pack_gru1_weights_hip,
+        "fused GRU_1 4-group operand pack");
   m.def("embed_gather", &embed_gather,
         "embedding gather (+fused decoder shift)");
   m.def("embed_scatter_add", &embed_scatter_add,

@@ -20,6 +20,7 @@ SOURCES = [
     os.path.join(HERE, "optim.hip"),
     os.path.join(HERE, "rerank.hip"),
     os.path.join(HERE, "embed.hip"),
+    os.path.join(HERE, "pack.hip"),
 ]
 
 
