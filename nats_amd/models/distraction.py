@@ -23,6 +23,19 @@ from torch import nn
 from .. import ops
 from .init import init_params, param_shapes
 
+_SIDE_STREAMS = {}
+
+
+def _side_stream(device):
+    """Per-device side HIP stream for forward work that overlaps the
+    persistent encoder scans (decoder input prep)."""
+    key = device.index or 0
+    s = _SIDE_STREAMS.get(key)
+    if s is None:
+        s = torch.cuda.Stream(device=device)
+        _SIDE_STREAMS[key] = s
+    return s
+
 
 def default_options(**overrides):
     """Hyperparameter schema mirroring train()'s keyword defaults
@@ -165,12 +178,30 @@ class NatsModel(nn.Module):
         return logit @ P["ff_logit_W"] + P["ff_logit_b"]
 
     def forward(self, x, x_mask, y, y_mask):
-        """Training graph -> per-sequence NLL (B,) (nats.py:658-772)."""
+        """Training graph -> per-sequence NLL (B,) (nats.py:658-772).
+
+        The decoder's input preparation (target embedding + hoisted input
+        GEMMs, nats.py:730-734/487-492) is independent of the encoder, so
+        on GPU it runs on a SIDE HIP stream concurrent with the
+        persistent encoder scans (which occupy ~126 of 256 CUs);
+        autograd replays backward on the recording streams."""
+        side = None
+        if x.is_cuda:
+            side = _side_stream(x.device)
+            side.wait_stream(torch.cuda.current_stream())
+        if side is not None:
+            with torch.cuda.stream(side):
+                emb_shifted = self.embed(y, shift=True)
+                yg, yc = self._dec_inputs(emb_shifted)
         ctx, init_state = self.encode(x, x_mask)
         pctx = self.project_ctx(ctx)
-
-        emb_shifted = self.embed(y, shift=True)
-        yg, yc = self._dec_inputs(emb_shifted)
+        if side is None:
+            emb_shifted = self.embed(y, shift=True)
+            yg, yc = self._dec_inputs(emb_shifted)
+        else:
+            torch.cuda.current_stream().wait_stream(side)
+            for t in (emb_shifted, yg, yc):
+                t.record_stream(torch.cuda.current_stream())
 
         h2s, ctxs, alphas, _, _ = ops.cond_gru_scan(
             yg, yc, y_mask, init_state, ctx, x_mask, pctx, self.P)

@@ -11,6 +11,15 @@ import sys
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
+# pre-tuned hipBLASLt algorithms for the readout/weight-grad GEMMs
+# (read-only; see bench.py)
+_TUNED = os.path.join(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))), "nats_amd", "ops", "tunableop_gfx950.csv")
+if os.path.exists(_TUNED) and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ:
+    os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
+    os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
+    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = _TUNED
+
 from nats_amd.engine.trainer import train
 
 DATA = os.environ.get(
