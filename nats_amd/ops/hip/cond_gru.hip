@@ -86,7 +86,22 @@ __global__ __launch_bounds__(256) void cond_attn_escore(
     const float* __restrict__ catt_p,    // [1]
     float* __restrict__ e_buf,           // [Ts][B]
     int B, int Ts, int A) {
+  // combined pstate + Dwei/Uatt staged in LDS once per block: the inner
+  // i-loop otherwise issues PS_KS+2 extra L2 loads per element for every
+  // one of this block's 256 s-positions
+  extern __shared__ float sm_e[];  // [A] ps, [A] Dwei, [A] Uatt
+  float* sm_ps = sm_e;
+  float* sm_dw = sm_e + A;
+  float* sm_ua = sm_e + 2 * A;
   const int b = blockIdx.x;
+  for (int i = threadIdx.x; i < A; i += blockDim.x) {
+    float ps = 0.f;
+    for (int k = 0; k < PS_KS; ++k) ps += ps_part[((long)k * 32 + b) * A + i];
+    sm_ps[i] = ps;
+    sm_dw[i] = Dwei[i];
+    sm_ua[i] = Uatt[i];
+  }
+  __syncthreads();
   const int s = blockIdx.y * blockDim.x + threadIdx.x;
   if (s >= Ts) return;
   const int ACH = gridDim.z;
@@ -98,11 +113,9 @@ __global__ __launch_bounds__(256) void cond_attn_escore(
   const float* prow = pctx + ((long)s * B + b) * A;
   float e = (blockIdx.z == 0) ? catt_p[0] : 0.f;
   for (int i = ibeg; i < iend; ++i) {
-    float ps = 0.f;
-    for (int k = 0; k < PS_KS; ++k) ps += ps_part[((long)k * 32 + b) * A + i];
-    e += tanhf(prow[i] + ps + accAu * Dwei[i]) * Uatt[i];
+    e += tanhf(prow[i] + sm_ps[i] + accAu * sm_dw[i]) * sm_ua[i];
   }
-  if (ACH == 1 && PS_KS >= 0) {
+  if (ACH == 1) {
     e_buf[(long)s * B + b] += e;  // single chunk: still additive (zeroed)
   } else {
     atomicAdd(e_buf + (long)s * B + b, e);
@@ -948,7 +961,7 @@ std::vector<torch::Tensor> cond_gru_fwd(
     // 3) attention scores (s- and A-parallel) + softmax + acc_alpha
     hipLaunchKernelGGL(cond_attn_escore,
                        dim3(B, cdiv_i(Ts, 256), A >= 16 ? 4 : 1), dim3(256),
-                       0, stream, pctx.data_ptr<float>(),
+                       3 * A * sizeof(float), stream, pctx.data_ptr<float>(),
                        ps_part.data_ptr<float>(), PS_KS,
                        accA.data_ptr<float>(), Dwei.data_ptr<float>(),
                        Uatt.data_ptr<float>(), catt.data_ptr<float>(),
