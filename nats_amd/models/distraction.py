@@ -23,19 +23,6 @@ from torch import nn
 from .. import ops
 from .init import init_params, param_shapes
 
-_SIDE_STREAMS = {}
-
-
-def _side_stream(device):
-    """Per-device side HIP stream for forward work that overlaps the
-    persistent encoder scans (decoder input prep)."""
-    key = device.index or 0
-    s = _SIDE_STREAMS.get(key)
-    if s is None:
-        s = torch.cuda.Stream(device=device)
-        _SIDE_STREAMS[key] = s
-    return s
-
 
 def default_options(**overrides):
     """Hyperparameter schema mirroring train()'s keyword defaults
@@ -180,28 +167,17 @@ class NatsModel(nn.Module):
     def forward(self, x, x_mask, y, y_mask):
         """Training graph -> per-sequence NLL (B,) (nats.py:658-772).
 
-        The decoder's input preparation (target embedding + hoisted input
-        GEMMs, nats.py:730-734/487-492) is independent of the encoder, so
-        on GPU it runs on a SIDE HIP stream concurrent with the
-        persistent encoder scans (which occupy ~126 of 256 CUs);
-        autograd replays backward on the recording streams."""
-        side = None
-        if x.is_cuda:
-            side = _side_stream(x.device)
-            side.wait_stream(torch.cuda.current_stream())
-        if side is not None:
-            with torch.cuda.stream(side):
-                emb_shifted = self.embed(y, shift=True)
-                yg, yc = self._dec_inputs(emb_shifted)
+        NOTE (measured negative, round 2): running the decoder input prep
+        (embed + hoisted GEMMs) on a side stream concurrent with the
+        persistent encoder scans REGRESSED the step 39.2 -> 41.2 ms —
+        the scans are latency-critical (one grid barrier per timestep)
+        and any co-resident work inflates every barrier interval; the
+        idle CUs are not free capacity. Reverted to sequential."""
         ctx, init_state = self.encode(x, x_mask)
         pctx = self.project_ctx(ctx)
-        if side is None:
-            emb_shifted = self.embed(y, shift=True)
-            yg, yc = self._dec_inputs(emb_shifted)
-        else:
-            torch.cuda.current_stream().wait_stream(side)
-            for t in (emb_shifted, yg, yc):
-                t.record_stream(torch.cuda.current_stream())
+
+        emb_shifted = self.embed(y, shift=True)
+        yg, yc = self._dec_inputs(emb_shifted)
 
         h2s, ctxs, alphas, _, _ = ops.cond_gru_scan(
             yg, yc, y_mask, init_state, ctx, x_mask, pctx, self.P)
