@@ -14,11 +14,20 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 # pre-tuned hipBLASLt algorithms for the readout/weight-grad GEMMs
 # (read-only; see bench.py)
 _TUNED = os.path.join(os.path.dirname(os.path.dirname(
-    os.path.abspath(__file__))), "nats_amd", "ops", "tunableop_gfx950.csv")
+    os.path.abspath(__file__))),
+                      "nats_amd", "ops", "tunableop_gfx950.csv")
 if os.path.exists(_TUNED) and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ:
+    # TunableOp substitutes the device ordinal into %d (or appends it
+    # when absent), so stage per-ordinal copies of the committed table
+    import shutil
+    import tempfile
+    _tdir = tempfile.mkdtemp(prefix="nats_tunableop")
+    for _i in range(8):
+        shutil.copy(_TUNED, os.path.join(_tdir, "tuned_%d.csv" % _i))
     os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
     os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
-    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = _TUNED
+    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = os.path.join(
+        _tdir, "tuned_%d.csv")
 
 from nats_amd.engine.trainer import train
 
