@@ -613,3 +613,40 @@ def test_batched_decode_graph_matches_ungraphed(ext):
         assert sorted(map(tuple, ps)) == sorted(map(tuple, gs))
         numpy.testing.assert_allclose(sorted(pc), sorted(gc), rtol=2e-3,
                                       atol=1e-3)
+
+
+def test_pack_kernels_match_torch(ext):
+    """pack.hip fused packers vs the torch chains they replaced (the
+    packed layouts are consumed by every MFMA kernel — a layout bug
+    would corrupt training silently)."""
+    from nats_amd.ops import gru as g
+    from nats_amd.ops import cond_gru as cg
+    torch.manual_seed(11)
+    H, C, A = 72, 144, 20  # deliberately non-multiples of 32
+    U = torch.randn(H, 2 * H, device="cuda")
+    Ux = torch.randn(H, H, device="cuda")
+    U_1 = torch.randn(H, 2 * H, device="cuda")
+    W_1 = torch.randn(C, 2 * H, device="cuda")
+    Ux_1 = torch.randn(H, H, device="cuda")
+    Wx_1 = torch.randn(C, H, device="cuda")
+    W_att = torch.randn(H, A, device="cuda")
+    Hpad = (H + 31) // 32 * 32
+    Cpad = (C + 31) // 32 * 32
+    Apad = (A + 31) // 32 * 32
+
+    # CPU inputs always take the torch pack path (reference)
+    ref_fwd = g.pack_fwd_weights(U.cpu(), Ux.cpu())
+    ref_bwd = g.pack_bwd_weights(U.cpu(), Ux.cpu())
+    ref_g1 = cg.pack_gru1_weights(U_1.cpu(), W_1.cpu(), Ux_1.cpu(),
+                                  Wx_1.cpu(), Hpad, Cpad)
+    ref_att_t = cg._pack_rows(W_att.cpu().t(), (A + 15) // 16 * 16, Hpad)
+    ref_att = cg._pack_rows(W_att.cpu(), (H + 15) // 16 * 16, Apad)
+
+    torch.testing.assert_close(g.pack_fwd_weights(U, Ux).cpu(), ref_fwd)
+    torch.testing.assert_close(g.pack_bwd_weights(U, Ux).cpu(), ref_bwd)
+    torch.testing.assert_close(
+        cg.pack_gru1_weights(U_1, W_1, Ux_1, Wx_1, Hpad, Cpad).cpu(), ref_g1)
+    torch.testing.assert_close(
+        cg._pack_rows(W_att.t(), (A + 15) // 16 * 16, Hpad).cpu(), ref_att_t)
+    torch.testing.assert_close(
+        cg._pack_rows(W_att, (H + 15) // 16 * 16, Apad).cpu(), ref_att)
