@@ -77,11 +77,12 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
     sidecar ``<saveto>.opt.npz`` with each checkpoint (default off: the
     reference restarts adadelta accumulators from zero on resume).
 
-    ``step_graph``: None (auto: on for CUDA when compatible) / True /
-    False — capture fwd+bwd+optimizer into one hipGraph per input shape
-    (utils/step_graph.py; the analogue of the reference's compiled
-    f_grad_shared+f_update pair being one host call). Ragged shapes
-    beyond the cache limit run eager.
+    ``step_graph=True`` (EXPERIMENTAL, default off): capture
+    fwd+bwd+optimizer into one hipGraph per input shape
+    (utils/step_graph.py). Replays were observed to intermittently stop
+    applying parameter updates on some machines (profiles/README.md), so
+    the reliable eager path is the default; the measured benefit when
+    healthy is under 1%.
     """
     logging.basicConfig(
         level=logging.DEBUG,
@@ -141,11 +142,13 @@ def train(dim_word=100, dim=1000, dim_att=100, encoder="gru",
         from ..utils import StepTimer
         step_timer = StepTimer()
 
-    # whole-step hipGraph capture (auto-on for CUDA; incompatible with
+    # whole-step hipGraph capture (EXPERIMENTAL opt-in; incompatible with
     # decay_c — the weight-decay term is added outside the closure — and
     # pointless with the per-section profiler's syncs)
     gcache = None
-    use_graph = step_graph if step_graph is not None else (world == 1)
+    # default OFF: graph replays intermittently stopped applying updates
+    # on some boxes (see profiles/README.md); opt in via step_graph=True
+    use_graph = bool(step_graph)
     if (use_graph and str(device).startswith("cuda") and decay_c == 0.0
             and not profile):
         from ..utils.step_graph import GraphedStepCache
