@@ -136,8 +136,9 @@ class CondGRUScanFn(torch.autograd.Function):
             U_att.float().reshape(-1).contiguous(),
             U_con.float().reshape(-1).contiguous(),
             W_con.float().reshape(-1).contiguous())
-        (dpre1_all, dpre2_all, dctxpre_all, dgate_all, dpstate_all,
-         dpctx_acc, gdDwei, gdUatt, gdcatt, dh_carry, daccC, daccA) = outs
+        (dpre1_all, dpre2_all, dctxpre_all, gdUcon, dpstate_all,
+         dpctx_acc, gdDwei, gdUatt, gdcatt, dh_carry, daccC, daccA,
+         gdWcon) = outs
 
         TB = T * B
         h1f = h1_all.reshape(TB, H).to(torch.bfloat16)
@@ -154,9 +155,10 @@ class CondGRUScanFn(torch.autograd.Function):
         dW_att = (h1f.t() @
                   dpstate_all.reshape(TB, A).to(torch.bfloat16)).float()
 
-        dgf = dgate_all.float()
-        dU_con = (dgf * ctxpre_all.float()).sum(dim=(0, 1)).reshape(C, 1)
-        dW_con = (dgf * accC_used.float()).sum(dim=(0, 1)).reshape(C, 1)
+        # dU_con/dW_con accumulated in the gate epilogue (one atomic per
+        # column per step) — no (T,B,C) re-streams here
+        dU_con = gdUcon.reshape(C, 1)
+        dW_con = gdWcon.reshape(C, 1)
 
         # GRU_2 (same algebra as the encoder scan backward)
         dyg = dpre2_all[..., :2 * H]

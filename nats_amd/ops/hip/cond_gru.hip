@@ -510,10 +510,21 @@ __global__ __launch_bounds__(384) void cond_bwd_gemm_dual_gate(
     float* __restrict__ daccC,          // [B][C] in/out
     float* __restrict__ dctxpre_f32,    // [B][C] out
     bf16_t* __restrict__ dctxpre_all_t, // [B][C] out
-    bf16_t* __restrict__ dgate_all_t,   // [B][C] out
+    const bf16_t* __restrict__ ctxpre_t,    // [B][C] (saved pre-gate sum)
+    const bf16_t* __restrict__ accC_used_t, // [B][C] (saved acc state)
+    float* __restrict__ gdUcon,         // [C] (+=, one atomic/col/step)
+    float* __restrict__ gdWcon,         // [C]
     int B) {
+  // dU_con[c] = sum_{t,b} dg*ctxpre and dW_con[c] = sum dg*accC_used
+  // fold in here (block-local b-reduction + one atomic per column per
+  // step) — the torch expression re-streamed three (T,B,C) buffers
+  __shared__ float ucon_s[JB], wcon_s[JB];
   __shared__ float part[3][32][JB + 1];
   const bool ctx_side = (blockIdx.y == 1);
+  if (ctx_side && threadIdx.x < JB) {
+    ucon_s[threadIdx.x] = 0.f;
+    wcon_s[threadIdx.x] = 0.f;
+  }
   const int N = ctx_side ? C : H;
   if ((int)blockIdx.x * JB >= N) return;
   const bf16_t* dstep = ctx_side ? dstepC : dstep1;
@@ -554,8 +565,16 @@ __global__ __launch_bounds__(384) void cond_bwd_gemm_dual_gate(
       const float dpre = dg * Ucon[i];
       dctxpre_f32[bi] = dpre;
       dctxpre_all_t[bi] = (bf16_t)dpre;
-      dgate_all_t[bi] = (bf16_t)dg;
       daccC[bi] += dg * Wcon[i];
+      atomicAdd(ucon_s + cc, dg * (float)ctxpre_t[bi]);
+      atomicAdd(wcon_s + cc, dg * (float)accC_used_t[bi]);
+    }
+  }
+  if (ctx_side) {
+    __syncthreads();
+    if (threadIdx.x < JB && i0 + (int)threadIdx.x < N) {
+      atomicAdd(gdUcon + i0 + threadIdx.x, ucon_s[threadIdx.x]);
+      atomicAdd(gdWcon + i0 + threadIdx.x, wcon_s[threadIdx.x]);
     }
   }
 }
@@ -1054,7 +1073,8 @@ std::vector<torch::Tensor> cond_gru_bwd(
   auto dpre1_all = torch::empty({T, B, 4 * H}, optsB);
   auto dpre2_all = torch::empty({T, B, 4 * H}, optsB);
   auto dctxpre_all = torch::empty({T, B, C}, optsB);
-  auto dgate_all = torch::empty({T, B, C}, optsB);
+  auto gdUcon = torch::zeros({C}, optsF);
+  auto gdWcon = torch::zeros({C}, optsF);
   auto dpstate_all = torch::zeros({T, B, A}, optsF);  // reduce atomic-adds
   auto dpctx_acc = torch::zeros({Ts, B, A}, optsF);
   auto gdDwei = torch::zeros({A}, optsF);
@@ -1141,7 +1161,10 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        Ucon.data_ptr<float>(), Wcon.data_ptr<float>(),
                        daccC.data_ptr<float>(), dctxpre_f32.data_ptr<float>(),
                        (bf16_t*)dctxpre_all.data_ptr() + (long)t * B * C,
-                       (bf16_t*)dgate_all.data_ptr() + (long)t * B * C, B);
+                       (const bf16_t*)ctxpre_all.data_ptr() + (long)t * B * C,
+                       (const bf16_t*)accC_used.data_ptr() + (long)t * B * C,
+                       gdUcon.data_ptr<float>(), gdWcon.data_ptr<float>(),
+                       B);
     // b5: attention backward — wave-per-(b,s) dalpha with the softmax-bwd
     // dot folded in (block tree + one atomic per block), then the
     // (b,s)-parallel scatter (dpctx/daccA/pc_buf) and the per-(b,i)
@@ -1205,6 +1228,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        B, H, K3Hpad);
   }
   HIP_CHECK(hipGetLastError());
-  return {dpre1_all, dpre2_all, dctxpre_all, dgate_all, dpstate_all,
-          dpctx_acc, gdDwei, gdUatt, gdcatt, dh_carry, daccC, daccA};
+  return {dpre1_all, dpre2_all, dctxpre_all, gdUcon, dpstate_all,
+          dpctx_acc, gdDwei, gdUatt, gdcatt, dh_carry, daccC, daccA,
+          gdWcon};
 }
