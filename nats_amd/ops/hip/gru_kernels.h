@@ -76,10 +76,4 @@ __global__ __launch_bounds__(384) void nats_gru_step_bwd_fused_bidir(
     GruBwdArgs a0, GruBwdArgs a1, const bf16_t* Ubwd0, const bf16_t* Ubwd1,
     int B, int H, int Kpad);
 
-// two INDEPENDENT recurrent-GEMM problems in one launch (grid.y selects;
-// used for the decoder's dh1 and dctx backward GEMMs, which have no
-// mutual dependency but would otherwise serialize on the stream)
-__global__ __launch_bounds__(384) void nats_gru_step_bwd_gemm_dual(
-    const bf16_t* dstepA, const bf16_t* WtA, const float* ddirA, float* outA,
-    int HA, int KpadA, const bf16_t* dstepB, const bf16_t* WtB,
-    const float* ddirB, float* outB, int HB, int KpadB, int B);
+

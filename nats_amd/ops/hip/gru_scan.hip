@@ -952,52 +952,6 @@ __global__ __launch_bounds__(384) void nats_gru_step_bwd_gemm(
 }
 
 
-__device__ __forceinline__ void gru_bwd_gemm_body(const bf16_t* dstep,
-                                                  const bf16_t* Wt,
-                                                  const float* ddirect,
-                                                  float* out, int B, int H,
-                                                  int Kpad, int wg) {
-  __shared__ float part[3][32][JB + 1];
-  const int wave = threadIdx.x / NATS_WAVE;
-  const int m = wave / 3;
-  const int ks = wave % 3;
-  const int i0 = wg * JB;
-  const int kchunk = ((Kpad / 3 + 31) / 32) * 32;
-  const int kbeg = ks * kchunk;
-  const int kend = min(Kpad, (ks + 1) * kchunk);
-  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  NATS_MFMA_KLOOP(acc, dstep, 16 * m, Kpad, Wt, i0, Kpad, kbeg, kend);
-  {
-    const int lane = threadIdx.x & (NATS_WAVE - 1);
-    const int col = lane & 15;
-    const int rbase = 16 * m + (lane >> 4) * 4;
-#pragma unroll
-    for (int i = 0; i < 4; ++i) part[ks][rbase + i][col] = acc[i];
-  }
-  __syncthreads();
-  for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
-    const int b = idx / JB;
-    const int c = idx % JB;
-    const int i = i0 + c;
-    if (i >= H) continue;
-    out[(long)b * H + i] = ddirect[(long)b * H + i] + part[0][b][c] +
-                           part[1][b][c] + part[2][b][c];
-  }
-}
-
-__global__ __launch_bounds__(384) void nats_gru_step_bwd_gemm_dual(
-    const bf16_t* dstepA, const bf16_t* WtA, const float* ddirA, float* outA,
-    int HA, int KpadA, const bf16_t* dstepB, const bf16_t* WtB,
-    const float* ddirB, float* outB, int HB, int KpadB, int B) {
-  if (blockIdx.y == 0) {
-    if ((int)blockIdx.x * JB < HA)
-      gru_bwd_gemm_body(dstepA, WtA, ddirA, outA, B, HA, KpadA, blockIdx.x);
-  } else {
-    if ((int)blockIdx.x * JB < HB)
-      gru_bwd_gemm_body(dstepB, WtB, ddirB, outB, B, HB, KpadB, blockIdx.x);
-  }
-}
-
 // ---------------- host drivers ----------------
 
 static inline int cdiv(int a, int b) { return (a + b - 1) / b; }
