@@ -132,20 +132,6 @@ def gen_sample_batched(model, xs, k=1, maxlen=30, use_unk=False,
         logp = probs.float().log()
         if not use_unk:
             logp[:, 1] = NEG_UNK
-        # distraction penalties batched over ALL alive sentences' rows in
-        # one fused-reduction call (histories share n_hist: every live
-        # hypothesis has one entry per elapsed step) — the per-sentence
-        # calls cost S kernel launches per step
-        pen_all = None
-        if ii > 0 and any_lambda:
-            has = [dstate[i]["ha"] for i in alive]
-            if all(h is not None for h in has):
-                pen_all = distraction_penalties_gpu(
-                    torch.cat(has, dim=1),
-                    torch.cat([dstate[i]["hc"] for i in alive], dim=1),
-                    torch.cat([dstate[i]["hs"] for i in alive], dim=1),
-                    alpha.float(), ctx_t.float(), h2.float(),
-                    kl_factor, ctx_factor, state_factor)
         sel_parts = []   # per alive sentence: (ranks (want,), costs (want,))
         row0 = 0
         for i in alive:
@@ -157,9 +143,7 @@ def gen_sample_batched(model, xs, k=1, maxlen=30, use_unk=False,
             cand_flat = cand.flatten()
             want = k - st[i]["dead"]
             sel_flat = cand_flat
-            if pen_all is not None:
-                sel_flat = (cand + pen_all[sl][:, None]).flatten()
-            elif ii > 0 and any_lambda and dstate[i]["ha"] is not None:
+            if ii > 0 and any_lambda and dstate[i]["ha"] is not None:
                 pen = distraction_penalties_gpu(
                     dstate[i]["ha"], dstate[i]["hc"], dstate[i]["hs"],
                     alpha[sl].float(), ctx_t[sl].float(), h2[sl].float(),
