@@ -478,46 +478,35 @@ __device__ __forceinline__ bool nats_grid_barrier(unsigned* sync,
 
 #define NATS_SYNC_WORDS 27
 
-// stage a [rows][Kpad] bf16 slice into LDS with an 8-BYTE-granularity
-// (row&15)<<3 XOR swizzle: 16 rows spread over all 16 8B slots of the
-// 128B bank period, so the paired ds_read_b64 fragment loads below are
-// conflict-free (the earlier 16B-granularity swizzle could only reach 8
-// slots — an inherent 2-way conflict for ds_read_b128; pmc_r2 showed
-// 335M conflict cycles in the backward scan). The LDS row stride is
-// Kpad rounded up to 64 elements (= the 128B bank period): row bases
-// share bank phase, only the swizzle separates them, and the XOR
-// (bits 3-6) stays inside the padded row for any Kpad.
+// stage a [rows][Kpad] bf16 slice into LDS with the (row&15)<<4 byte-XOR
+// swizzle and an LDS row stride of ld elements (ld = Kpad + 8 pads each
+// row by 16B, breaking the bank-period alignment of 2KB/6KB rows that
+// left the XOR spread 2-way-conflicted — pmc_r2: 335M conflict cycles
+// in the backward scan). The XOR flips byte bits 4-7 only; Kpad*2 is a
+// multiple of 256, so swizzled offsets stay inside the row.
 __device__ __forceinline__ void stage_weights_lds(bf16_t* lds,
                                                   const bf16_t* src,
                                                   int rows, int Kpad,
                                                   int ld) {
-  const long total8 = (long)rows * Kpad * 2 / 8;  // 8B chunks
-  for (long idx = threadIdx.x; idx < total8; idx += blockDim.x) {
-    long byte = idx * 8;
+  const long total16 = (long)rows * Kpad * 2 / 16;  // 16B chunks
+  for (long idx = threadIdx.x; idx < total16; idx += blockDim.x) {
+    long byte = idx * 16;
     const int row = (int)(byte / ((long)Kpad * 2));
     const long within = byte - (long)row * Kpad * 2;
     const long dst =
-        (long)row * ld * 2 + (within ^ (long)((row & 15) << 3));
-    *(uint2*)((char*)lds + dst) = *(const uint2*)((const char*)src + byte);
+        (long)row * ld * 2 + (within ^ (long)((row & 15) << 4));
+    *(uint4*)((char*)lds + dst) = *(const uint4*)((const char*)src + byte);
   }
 }
-
-typedef __bf16 bf16x4_t __attribute__((ext_vector_type(4)));
 
 __device__ __forceinline__ bf16x8 frag_bt_lds_swz(const bf16_t* lds, int row,
                                                   int ld, int k) {
   const int lane = threadIdx.x & (NATS_WAVE - 1);
   const int r = row + (lane & 15);
-  const long base = (long)r * ld * 2;
-  const long w = ((long)k + (lane >> 4) * 8) * 2;  // 16B-aligned
-  const long sw = (long)((r & 15) << 3);
-  const bf16x4_t lo = *(const bf16x4_t*)((const char*)lds + base + (w ^ sw));
-  const bf16x4_t hi =
-      *(const bf16x4_t*)((const char*)lds + base + ((w + 8) ^ sw));
-  bf16x8 out;
-  out[0] = lo[0]; out[1] = lo[1]; out[2] = lo[2]; out[3] = lo[3];
-  out[4] = hi[0]; out[5] = hi[1]; out[6] = hi[2]; out[7] = hi[3];
-  return out;
+  const long within = ((long)k + (lane >> 4) * 8) * 2;
+  const long byte =
+      (long)r * ld * 2 + (within ^ (long)((r & 15) << 4));
+  return *(const bf16x8*)((const char*)lds + byte);
 }
 
 // swizzled-LDS MFMA K-loop (B operand from LDS, A from global), 4-deep on
@@ -620,7 +609,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
     unsigned* sync, unsigned nwg, int ngrp, int xpd,
     int unsafe_nobarrier) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const int ldw = (Hpad + 63) / 64 * 64;
+  const int ldw = Hpad + 8;  // padded LDS row stride (bank stagger)
   bf16_t* upk_lds = (bf16_t*)smem;                       // [3*16][ldw] swz
   float(*pre)[32][JB + 1] =
       (float(*)[32][JB + 1])(smem + (long)3 * JB * ldw * 2);
@@ -779,7 +768,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
     GruPersistBwd p0, GruPersistBwd p1, int T, int B, int H, int K3pad,
     unsigned* sync, unsigned nwg, int ngrp, int xpd) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const int ldw = (K3pad + 63) / 64 * 64;
+  const int ldw = K3pad + 8;  // padded LDS row stride (bank stagger)
   bf16_t* ub_lds = (bf16_t*)smem;  // [16][ldw] swizzled
   float(*part)[32][JB + 1] =
       (float(*)[32][JB + 1])(smem + (long)JB * ldw * 2);
@@ -1074,8 +1063,8 @@ std::vector<torch::Tensor> gru_scan_fwd_bidir(
   bf16_t* hbf = (bf16_t*)h_bf.data_ptr();
 
   // persistent path: weight slices LDS-resident across all T steps
-  const size_t smem_fwd = (size_t)3 * JB * ((Hpad + 63) / 64 * 64) * 2 +
-                          sizeof(float) * 3 * 32 * (JB + 1);
+  const size_t smem_fwd =
+      (size_t)3 * JB * (Hpad + 8) * 2 + sizeof(float) * 3 * 32 * (JB + 1);
   const bool persistent = (2 * ngrp <= 192) && (smem_fwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
@@ -1184,8 +1173,8 @@ std::vector<torch::Tensor> gru_scan_bwd_bidir(
   auto dh0c = dh_out0.contiguous().to(torch::kFloat32);
   auto dh1c = dh_out1.contiguous().to(torch::kFloat32);
 
-  const size_t smem_bwd = (size_t)JB * ((K3pad + 63) / 64 * 64) * 2 +
-                          sizeof(float) * 3 * 32 * (JB + 1);
+  const size_t smem_bwd =
+      (size_t)JB * (K3pad + 8) * 2 + sizeof(float) * 3 * 32 * (JB + 1);
   const bool persistent = (2 * ngrp <= 192) && (smem_bwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
