@@ -479,33 +479,25 @@ __device__ __forceinline__ bool nats_grid_barrier(unsigned* sync,
 #define NATS_SYNC_WORDS 27
 
 // stage a [rows][Kpad] bf16 slice into LDS with the (row&15)<<4 byte-XOR
-// swizzle and an LDS row stride of ld elements (ld = Kpad + 8 pads each
-// row by 16B, breaking the bank-period alignment of 2KB/6KB rows that
-// left the XOR spread 2-way-conflicted — pmc_r2: 335M conflict cycles
-// in the backward scan). The XOR flips byte bits 4-7 only; Kpad*2 is a
-// multiple of 256, so swizzled offsets stay inside the row.
+// swizzle; read back with the same XOR (write+read swizzled together).
 __device__ __forceinline__ void stage_weights_lds(bf16_t* lds,
                                                   const bf16_t* src,
-                                                  int rows, int Kpad,
-                                                  int ld) {
+                                                  int rows, int Kpad) {
   const long total16 = (long)rows * Kpad * 2 / 16;  // 16B chunks
   for (long idx = threadIdx.x; idx < total16; idx += blockDim.x) {
     long byte = idx * 16;
     const int row = (int)(byte / ((long)Kpad * 2));
-    const long within = byte - (long)row * Kpad * 2;
-    const long dst =
-        (long)row * ld * 2 + (within ^ (long)((row & 15) << 4));
+    const long dst = byte ^ (long)((row & 15) << 4);
     *(uint4*)((char*)lds + dst) = *(const uint4*)((const char*)src + byte);
   }
 }
 
 __device__ __forceinline__ bf16x8 frag_bt_lds_swz(const bf16_t* lds, int row,
-                                                  int ld, int k) {
+                                                  int Kpad, int k) {
   const int lane = threadIdx.x & (NATS_WAVE - 1);
   const int r = row + (lane & 15);
-  const long within = ((long)k + (lane >> 4) * 8) * 2;
-  const long byte =
-      (long)r * ld * 2 + (within ^ (long)((r & 15) << 4));
+  long byte = ((long)r * Kpad + k + (lane >> 4) * 8) * 2;
+  byte ^= (long)((r & 15) << 4);
   return *(const bf16x8*)((const char*)lds + byte);
 }
 
@@ -609,10 +601,9 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
     unsigned* sync, unsigned nwg, int ngrp, int xpd,
     int unsafe_nobarrier) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const int ldw = Hpad + 8;  // padded LDS row stride (bank stagger)
-  bf16_t* upk_lds = (bf16_t*)smem;                       // [3*16][ldw] swz
+  bf16_t* upk_lds = (bf16_t*)smem;                       // [3*16][Hpad] swz
   float(*pre)[32][JB + 1] =
-      (float(*)[32][JB + 1])(smem + (long)3 * JB * ldw * 2);
+      (float(*)[32][JB + 1])(smem + (long)3 * JB * Hpad * 2);
 
   const int sel = nats_claim_dir_slot(sync + 2 * NATS_SYNC_WORDS, ngrp,
                                       xpd);
@@ -624,8 +615,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
   // each direction syncs only its own ngrp blocks (own sync slab)
   sync += (long)dirsel * NATS_SYNC_WORDS;
   nwg = (unsigned)ngrp;
-  stage_weights_lds(upk_lds, p.Upk + (long)wg * 3 * JB * Hpad, 3 * JB, Hpad,
-                    ldw);
+  stage_weights_lds(upk_lds, p.Upk + (long)wg * 3 * JB * Hpad, 3 * JB, Hpad);
   __syncthreads();
   NatsBarrierCtx bctx;
   if (!nats_barrier_init(sync, nwg, bctx)) {
@@ -688,7 +678,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
 
     const bf16_t* h_bf_in = p.h_bf + (t % 2) * hb;
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    NATS_MFMA_KLOOP_LDSB(acc, h_bf_in, 16 * m, Hpad, upk_lds, g * JB, ldw,
+    NATS_MFMA_KLOOP_LDSB(acc, h_bf_in, 16 * m, Hpad, upk_lds, g * JB, Hpad,
                          0, Hpad);
     {
       const int lane = threadIdx.x & (NATS_WAVE - 1);
@@ -768,10 +758,9 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
     GruPersistBwd p0, GruPersistBwd p1, int T, int B, int H, int K3pad,
     unsigned* sync, unsigned nwg, int ngrp, int xpd) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const int ldw = K3pad + 8;  // padded LDS row stride (bank stagger)
-  bf16_t* ub_lds = (bf16_t*)smem;  // [16][ldw] swizzled
+  bf16_t* ub_lds = (bf16_t*)smem;  // [16][K3pad] swizzled
   float(*part)[32][JB + 1] =
-      (float(*)[32][JB + 1])(smem + (long)JB * ldw * 2);
+      (float(*)[32][JB + 1])(smem + (long)JB * K3pad * 2);
 
   const int sel = nats_claim_dir_slot(sync + 2 * NATS_SYNC_WORDS, ngrp,
                                       xpd);
@@ -781,7 +770,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
   const GruPersistBwd& p = (dirsel == 0) ? p0 : p1;
   sync += (long)dirsel * NATS_SYNC_WORDS;
   nwg = (unsigned)ngrp;
-  stage_weights_lds(ub_lds, p.Ubwd + (long)wg * JB * K3pad, JB, K3pad, ldw);
+  stage_weights_lds(ub_lds, p.Ubwd + (long)wg * JB * K3pad, JB, K3pad);
   __syncthreads();
   NatsBarrierCtx bctx;
   if (!nats_barrier_init(sync, nwg, bctx)) {
@@ -851,7 +840,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
 
     const bf16_t* dstep_in = p.dstep + ((t + 1) % 2) * ds;
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    NATS_MFMA_KLOOP_LDSB(acc, dstep_in, 16 * m, K3pad, ub_lds, 0, ldw,
+    NATS_MFMA_KLOOP_LDSB(acc, dstep_in, 16 * m, K3pad, ub_lds, 0, K3pad,
                          kbeg, kend);
     {
       const int lane = threadIdx.x & (NATS_WAVE - 1);
@@ -1064,7 +1053,7 @@ std::vector<torch::Tensor> gru_scan_fwd_bidir(
 
   // persistent path: weight slices LDS-resident across all T steps
   const size_t smem_fwd =
-      (size_t)3 * JB * (Hpad + 8) * 2 + sizeof(float) * 3 * 32 * (JB + 1);
+      (size_t)3 * JB * Hpad * 2 + sizeof(float) * 3 * 32 * (JB + 1);
   const bool persistent = (2 * ngrp <= 192) && (smem_fwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
@@ -1174,7 +1163,7 @@ std::vector<torch::Tensor> gru_scan_bwd_bidir(
   auto dh1c = dh_out1.contiguous().to(torch::kFloat32);
 
   const size_t smem_bwd =
-      (size_t)JB * (K3pad + 8) * 2 + sizeof(float) * 3 * 32 * (JB + 1);
+      (size_t)JB * K3pad * 2 + sizeof(float) * 3 * 32 * (JB + 1);
   const bool persistent = (2 * ngrp <= 192) && (smem_bwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
