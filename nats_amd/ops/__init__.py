@@ -91,7 +91,13 @@ def gru_scan_bidir(xg0, xc0, mask0, U0, Ux0, xg1, xc1, mask1, U1, Ux1):
     if _use_hip(xg0, U0):
         from .gru import gru_scan_bidir_hip
         B = xg0.shape[1]
-        if B <= MAX_KERNEL_BATCH:
+        # the persistent scans take up to 64 rows natively (two 32-row
+        # chunk jobs concurrent in one launch — same barrier count as 32)
+        direct_max = MAX_KERNEL_BATCH
+        if _hip_ext() is not None and _hip_ext().gru_persistent_ok(
+                U0.shape[0]):
+            direct_max = 2 * MAX_KERNEL_BATCH
+        if B <= direct_max:
             return gru_scan_bidir_hip(xg0, xc0, mask0, U0, Ux0, xg1, xc1,
                                       mask1, U1, Ux1)
         outs = [gru_scan_bidir_hip(

@@ -28,6 +28,7 @@ torch::Tensor softmax_ce_bwd(torch::Tensor logits, torch::Tensor targets,
                              torch::Tensor stats, torch::Tensor dnll);
 torch::Tensor mfma_gemm_bt(torch::Tensor A, torch::Tensor Bt);
 double barrier_bench(int nwg_x, int nwg_y, int iters);
+bool gru_persistent_ok(long H);
 double barrier_bench_xcd(int nwg, int nxcd, int iters);
 torch::Tensor rerank_penalties(torch::Tensor hist_a, torch::Tensor hist_c,
                                torch::Tensor hist_s, torch::Tensor cur_a,
@@ -92,6 +93,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_ce_bwd", &softmax_ce_bwd, "fused softmax+CE backward");
   m.def("mfma_gemm_bt", &mfma_gemm_bt, "MFMA layout self-test GEMM");
   m.def("barrier_bench", &barrier_bench, "grid barrier us/iteration");
+  m.def("gru_persistent_ok", &gru_persistent_ok,
+        "persistent scan available for this hidden size");
   m.def("barrier_bench_xcd", &barrier_bench_xcd,
         "XCD-constrained grid barrier us/iteration");
   m.def("rerank_penalties", &rerank_penalties,

@@ -539,15 +539,22 @@ __device__ __forceinline__ bf16x8 frag_bt_lds_swz(const bf16_t* lds, int row,
     }                                                                        \
   } while (0)
 
+// One persistent-scan job = one (direction, batch-chunk): pointers are
+// pre-offset to the chunk's first row and strides skip the FULL batch
+// per timestep, so jobs of a B>32 call share the (T, Btot, ...) output
+// tensors without copies. Up to 4 jobs run concurrently in one launch
+// (2 directions x 2 chunks of the 32-row MFMA tiling).
 struct GruPersistFwd {
-  const bf16_t* xg;    // [T][B][2H]
-  const bf16_t* xc;    // [T][B][H]
-  const float* mask;   // [T][B] or null
+  const bf16_t* xg;    // -> [t][b][2H] rows of this chunk
+  const bf16_t* xc;
+  const float* mask;   // or null
   const bf16_t* Upk;   // [ngrp*3*16][Hpad]
-  float* h_all;        // [T][B][H]
-  bf16_t* h_bf;        // [2][32][Hpad] ping-pong
-  bf16_t* saved;       // [T][B][3H]
-  const float* h0;     // [B][H]
+  float* h_all;
+  bf16_t* h_bf;        // [2][32][Hpad] ping-pong (per job)
+  bf16_t* saved;
+  const float* h0;     // [B][H] chunk rows (contiguous)
+  int B;               // rows in this chunk (<= 32)
+  long sxg, sxc, smask, sh, ssaved;  // per-t element strides
 };
 
 // XCD-preferred direction claim (profiles/barrier_xcd.json: a 63-WG
@@ -560,61 +567,65 @@ struct GruPersistFwd {
 // the other side (correct under any placement — the barrier census is
 // placement-independent); unclaimed blocks exit. Greedy claiming has NO
 // census spin, so there is no new hang mode.
-__device__ __forceinline__ int nats_claim_dir_slot(unsigned* claim,
-                                                   int ngrp, int xpd) {
+__device__ __forceinline__ int nats_claim_job_slot(unsigned* claim,
+                                                   int ngrp, int njobs,
+                                                   int xpj) {
   __shared__ int sh_sel;
   if (threadIdx.x == 0) {
     const unsigned xcc = nats_xcc_id();
-    // xpd = preferred XCDs per direction (4: halves of the chip; 2:
-    // tighter locality, XCDs 0-1 / 2-3 with 4-7 unpreferred)
-    int dir = (xcc < (unsigned)xpd) ? 0
-              : (xcc < (unsigned)(2 * xpd)) ? 1 : -1;
+    // xpj = preferred XCDs per job; jobs partition the first njobs*xpj
+    // XCDs, blocks elsewhere have no preference
+    int job = (xcc < (unsigned)(njobs * xpj)) ? (int)(xcc / xpj) : -1;
     int slot = -1;
-    if (dir >= 0) {
+    if (job >= 0) {
       const unsigned s0 = __hip_atomic_fetch_add(
-          claim + dir, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+          claim + job, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
       if (s0 < (unsigned)ngrp) slot = (int)s0;
     }
     if (slot < 0) {
-      // unpreferred block (or preferred side full): let the preferred
-      // blocks claim first, then fill leftovers on either side
+      // unpreferred block (or preferred job full): let the preferred
+      // blocks claim first, then fill leftovers round-robin
       for (int i = 0; i < 6; ++i) __builtin_amdgcn_s_sleep(127);
-      const int first = (dir >= 0) ? 1 - dir : (int)(xcc & 1);
-      for (int k = 0; k < 2 && slot < 0; ++k) {
-        const int d2 = k == 0 ? first : 1 - first;
+      const int first = (job >= 0) ? (job + 1) % njobs
+                                   : (int)(xcc % (unsigned)njobs);
+      for (int k = 0; k < njobs && slot < 0; ++k) {
+        const int d2 = (first + k) % njobs;
         const unsigned s1 = __hip_atomic_fetch_add(
             claim + d2, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         if (s1 < (unsigned)ngrp) {
-          dir = d2;
+          job = d2;
           slot = (int)s1;
         }
       }
     }
-    sh_sel = (slot < 0) ? -1 : (dir << 16 | slot);
+    sh_sel = (slot < 0) ? -1 : (job << 16 | slot);
   }
   __syncthreads();
   return sh_sel;
 }
 
 __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
-    GruPersistFwd p0, GruPersistFwd p1, int T, int B, int H, int Hpad,
-    unsigned* sync, unsigned nwg, int ngrp, int xpd,
+    GruPersistFwd p0, GruPersistFwd p1, GruPersistFwd p2, GruPersistFwd p3,
+    int T, int H, int Hpad, unsigned* sync, int ngrp, int njobs, int xpd,
     int unsafe_nobarrier) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* upk_lds = (bf16_t*)smem;                       // [3*16][Hpad] swz
   float(*pre)[32][JB + 1] =
       (float(*)[32][JB + 1])(smem + (long)3 * JB * Hpad * 2);
 
-  const int sel = nats_claim_dir_slot(sync + 2 * NATS_SYNC_WORDS, ngrp,
-                                      xpd);
+  const int sel = nats_claim_job_slot(sync + (long)njobs * NATS_SYNC_WORDS,
+                                      ngrp, njobs, xpd);
   if (sel < 0) return;
-  const int dirsel = sel >> 16;
+  const int jobsel = sel >> 16;
   const int wg = sel & 0xffff;
-  const GruPersistFwd& p = (dirsel == 0) ? p0 : p1;
-  // per-direction barrier: the two directions are data-independent, so
-  // each direction syncs only its own ngrp blocks (own sync slab)
-  sync += (long)dirsel * NATS_SYNC_WORDS;
-  nwg = (unsigned)ngrp;
+  const GruPersistFwd& p = (jobsel == 0) ? p0
+                           : (jobsel == 1) ? p1
+                           : (jobsel == 2) ? p2 : p3;
+  const int B = p.B;
+  // per-job barrier: jobs (direction x batch-chunk) are data-independent
+  // so each syncs only its own ngrp blocks (own sync slab)
+  sync += (long)jobsel * NATS_SYNC_WORDS;
+  unsigned nwg = (unsigned)ngrp;
   stage_weights_lds(upk_lds, p.Upk + (long)wg * 3 * JB * Hpad, 3 * JB, Hpad);
   __syncthreads();
   NatsBarrierCtx bctx;
@@ -650,8 +661,8 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
 
   for (int t = 0; t < T; ++t) {
     // prefetch this step's inputs; latency hides under the MFMA phase
-    const bf16_t* xg_t = p.xg + (long)t * B * 2 * H;
-    const bf16_t* xc_t = p.xc + (long)t * B * H;
+    const bf16_t* xg_t = p.xg + (long)t * p.sxg;
+    const bf16_t* xc_t = p.xc + (long)t * p.sxc;
     float pf_xr[2], pf_xu[2], pf_xc[2];
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
@@ -689,8 +700,8 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
     }
     __syncthreads();
 
-    const float* mask_t = p.mask ? p.mask + (long)t * B : nullptr;
-    float* h_out = p.h_all + (long)t * B * H;
+    const float* mask_t = p.mask ? p.mask + (long)t * p.smask : nullptr;
+    float* h_out = p.h_all + (long)t * p.sh;
     bf16_t* h_bf_out = p.h_bf + ((t + 1) % 2) * hb;
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
@@ -717,7 +728,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
       pend_u[it] = u;
       pend_px[it] = px;
     }
-    pend_saved = p.saved + (long)t * B * 3 * H;
+    pend_saved = p.saved + (long)t * p.ssaved;
     have_pend = true;
     if (unsafe_nobarrier) {  // TIMING EXPERIMENTS ONLY (racy!)
       __syncthreads();
@@ -742,34 +753,39 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
 }
 
 struct GruPersistBwd {
-  const float* dh_out;  // [T][B][H]
-  const float* h_all;   // [T][B][H]
-  const bf16_t* saved;  // [T][B][3H]
-  const bf16_t* xc;     // [T][B][H]
-  const float* mask;    // [T][B] or null
+  const float* dh_out;  // -> [t][b][H] rows of this chunk
+  const float* h_all;
+  const bf16_t* saved;
+  const bf16_t* xc;
+  const float* mask;    // or null
   const bf16_t* Ubwd;   // [ngrp*16][K3pad]
-  bf16_t* dstep;        // [2][32][K3pad] ping-pong (zeroed)
-  float* ddirect;       // [B][H] (zeroed)
-  bf16_t* dpre;         // [T][B][4H]
-  const float* h0;      // [B][H]
+  bf16_t* dstep;        // [2][32][K3pad] ping-pong (zeroed, per job)
+  float* ddirect;       // [B][H] (zeroed, per job)
+  bf16_t* dpre;
+  const float* h0;      // [B][H] chunk rows (contiguous)
+  int B;
+  long sdh, sh, ssaved, sxc, smask, sdpre;  // per-t element strides
 };
 
 __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
-    GruPersistBwd p0, GruPersistBwd p1, int T, int B, int H, int K3pad,
-    unsigned* sync, unsigned nwg, int ngrp, int xpd) {
+    GruPersistBwd p0, GruPersistBwd p1, GruPersistBwd p2, GruPersistBwd p3,
+    int T, int H, int K3pad, unsigned* sync, int ngrp, int njobs, int xpd) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16_t* ub_lds = (bf16_t*)smem;  // [16][K3pad] swizzled
   float(*part)[32][JB + 1] =
       (float(*)[32][JB + 1])(smem + (long)JB * K3pad * 2);
 
-  const int sel = nats_claim_dir_slot(sync + 2 * NATS_SYNC_WORDS, ngrp,
-                                      xpd);
+  const int sel = nats_claim_job_slot(sync + (long)njobs * NATS_SYNC_WORDS,
+                                      ngrp, njobs, xpd);
   if (sel < 0) return;
-  const int dirsel = sel >> 16;
+  const int jobsel = sel >> 16;
   const int wg = sel & 0xffff;
-  const GruPersistBwd& p = (dirsel == 0) ? p0 : p1;
-  sync += (long)dirsel * NATS_SYNC_WORDS;
-  nwg = (unsigned)ngrp;
+  const GruPersistBwd& p = (jobsel == 0) ? p0
+                           : (jobsel == 1) ? p1
+                           : (jobsel == 2) ? p2 : p3;
+  const int B = p.B;
+  sync += (long)jobsel * NATS_SYNC_WORDS;
+  unsigned nwg = (unsigned)ngrp;
   stage_weights_lds(ub_lds, p.Ubwd + (long)wg * JB * K3pad, JB, K3pad);
   __syncthreads();
   NatsBarrierCtx bctx;
@@ -806,11 +822,11 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
 
   for (int t = T - 1; t >= 0; --t) {
     // prefetch this step's pointwise inputs under the MFMA phase
-    const bf16_t* saved_t = p.saved + (long)t * B * 3 * H;
-    const bf16_t* xc_t = p.xc + (long)t * B * H;
+    const bf16_t* saved_t = p.saved + (long)t * p.ssaved;
+    const bf16_t* xc_t = p.xc + (long)t * p.sxc;
     const float* h_prev =
-        (t == 0) ? p.h0 : (p.h_all + (long)(t - 1) * B * H);
-    const float* dh_out_t = p.dh_out + (long)t * B * H;
+        (t == 0) ? p.h0 : (p.h_all + (long)(t - 1) * p.sh);
+    const float* dh_out_t = p.dh_out + (long)t * p.sdh;
     float pf_r[2], pf_u[2], pf_px[2], pf_xc[2], pf_hp[2], pf_dho[2];
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
@@ -851,7 +867,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
     }
     __syncthreads();
 
-    const float* mask_t = p.mask ? p.mask + (long)t * B : nullptr;
+    const float* mask_t = p.mask ? p.mask + (long)t * p.smask : nullptr;
     bf16_t* dstep_out = p.dstep + (t % 2) * ds;
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
@@ -886,7 +902,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_bwd(
       pend[it][2] = dpx;
       pend[it][3] = dpxl;
     }
-    pend_dpre = p.dpre + (long)t * B * 4 * H;
+    pend_dpre = p.dpre + (long)t * p.sdpre;
     have_pend = true;
     if (!nats_grid_barrier(sync, (unsigned)(T - t), bctx)) {
       if (threadIdx.x == 0) p.ddirect[0] = __builtin_nanf("");
@@ -1022,7 +1038,7 @@ std::vector<torch::Tensor> gru_scan_fwd_bidir(
     torch::Tensor Upk0, torch::Tensor xg1, torch::Tensor xc1,
     c10::optional<torch::Tensor> mask1, torch::Tensor Upk1) {
   const int T = xg0.size(0), B = xg0.size(1), H = xc0.size(2);
-  TORCH_CHECK(B <= 32, "gru_scan: batch per step must be <= 32");
+  TORCH_CHECK(B <= 64, "bidir gru_scan: batch must be <= 64");
   const int Hpad = Upk0.size(1);
   const int ngrp = cdiv(H, JB);
 
@@ -1057,40 +1073,62 @@ std::vector<torch::Tensor> gru_scan_fwd_bidir(
   const bool persistent = (2 * ngrp <= 192) && (smem_fwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
-    auto sync = torch::zeros({2 * NATS_SYNC_WORDS + 4},
+    // jobs = directions x 32-row batch chunks (B <= 64), all concurrent
+    // in ONE launch with per-job barriers — a 64-row batch costs the
+    // same T barrier intervals as a 32-row one instead of 2x
+    const int nchunk = cdiv(B, 32);
+    TORCH_CHECK(nchunk <= 2, "persistent bidir scan: B must be <= 64");
+    const int njobs = 2 * nchunk;
+    auto sync = torch::zeros({njobs * NATS_SYNC_WORDS + 8},
                              xg0.options().dtype(torch::kInt32));
     unsigned* sync_p = (unsigned*)sync.data_ptr<int>();
-    GruPersistFwd p0{(const bf16_t*)xg0.data_ptr(),
-                     (const bf16_t*)xc0.data_ptr(),
-                     m0,
-                     (const bf16_t*)Upk0.data_ptr(),
-                     h_all0.data_ptr<float>(),
-                     hbf + 0 * 2 * hb,
-                     (bf16_t*)saved0.data_ptr(),
-                     h00.data_ptr<float>()};
-    GruPersistFwd p1{(const bf16_t*)xg1.data_ptr(),
-                     (const bf16_t*)xc1.data_ptr(),
-                     m1,
-                     (const bf16_t*)Upk1.data_ptr(),
-                     h_all1.data_ptr<float>(),
-                     hbf + 1 * 2 * hb,
-                     (bf16_t*)saved1.data_ptr(),
-                     h00.data_ptr<float>()};
+    auto h_bfj = torch::zeros({njobs, 2, 32, Hpad}, optsB);
+    bf16_t* hbfj = (bf16_t*)h_bfj.data_ptr();
+    GruPersistFwd jobs[4];
+    for (int j = 0; j < njobs; ++j) {
+      const int dir = j & 1;            // interleave so chunk pairs of a
+      const int ch = j >> 1;            // direction sit on far XCD sets
+      const int a = ch * 32;
+      const int Bj = std::min(32, B - a);
+      const torch::Tensor& xg = dir == 0 ? xg0 : xg1;
+      const torch::Tensor& xc = dir == 0 ? xc0 : xc1;
+      const torch::Tensor& Upk = dir == 0 ? Upk0 : Upk1;
+      torch::Tensor& h_all = dir == 0 ? h_all0 : h_all1;
+      torch::Tensor& saved = dir == 0 ? saved0 : saved1;
+      const float* mj = dir == 0 ? m0 : m1;
+      jobs[j] = GruPersistFwd{
+          (const bf16_t*)xg.data_ptr() + (long)a * 2 * H,
+          (const bf16_t*)xc.data_ptr() + (long)a * H,
+          mj ? mj + a : nullptr,
+          (const bf16_t*)Upk.data_ptr(),
+          h_all.data_ptr<float>() + (long)a * H,
+          hbfj + (long)j * 2 * hb,
+          (bf16_t*)saved.data_ptr() + (long)a * 3 * H,
+          h00.data_ptr<float>() + (long)a * H,
+          Bj,
+          (long)B * 2 * H, (long)B * H, (long)B, (long)B * H,
+          (long)B * 3 * H};
+    }
+    for (int j = njobs; j < 4; ++j) jobs[j] = jobs[0];
     const int unsafe = getenv("NATS_UNSAFE_NOBARRIER") != nullptr;
-    // overprovisioned 1-D grid: blocks self-select a (direction, tile)
-    // slot with XCD preference (see nats_claim_dir_slot); all 256 fit
+    // overprovisioned 1-D grid: blocks self-select a (job, tile) slot
+    // with XCD preference (see nats_claim_job_slot); all 256 fit
     // resident (smem gate is 150KB -> >=1 block/CU)
     const char* xpd_env = getenv("NATS_XPD");
-    // 2 XCDs (64 CUs) when the per-direction grid fits comfortably
-    // (LCSTS ngrp=32: +2.5% measured); 4 when 63 WGs would sit 1/CU
-    const int xpd = xpd_env ? atoi(xpd_env) : (ngrp <= 48 ? 2 : 4);
+    // 2 XCDs (64 CUs) per job when its grid fits comfortably (LCSTS
+    // ngrp=32: +2.5% measured); 4 when 63 WGs would sit 1/CU
+    const int xpd = xpd_env ? atoi(xpd_env)
+                            : (njobs == 4 ? 2 : (ngrp <= 48 ? 2 : 4));
     hipLaunchKernelGGL(nats_gru_persistent_fwd, dim3(256), dim3(384),
-                       smem_fwd, stream, p0, p1, T, B, H, Hpad, sync_p,
-                       (unsigned)(2 * ngrp), ngrp, xpd, unsafe);
+                       smem_fwd, stream, jobs[0], jobs[1], jobs[2], jobs[3],
+                       T, H, Hpad, sync_p, ngrp, njobs, xpd, unsafe);
     HIP_CHECK(hipGetLastError());
     return {h_all0, saved0, h_all1, saved1};
   }
 
+  TORCH_CHECK(B <= 32,
+              "bidir scan: B > 32 requires the persistent path "
+              "(gru_persistent_ok)");
   for (int t = 0; t < T; ++t) {
     GruFwdArgs a0{
         hbf + 0 * 2 * hb + (t % 2) * hb,
@@ -1167,49 +1205,72 @@ std::vector<torch::Tensor> gru_scan_bwd_bidir(
   const bool persistent = (2 * ngrp <= 192) && (smem_bwd <= 150 * 1024) &&
                           (getenv("NATS_NO_PERSISTENT") == nullptr);
   if (persistent) {
-    auto sync = torch::zeros({2 * NATS_SYNC_WORDS + 4},
+    const int nchunk = cdiv(B, 32);
+    TORCH_CHECK(nchunk <= 2, "persistent bidir scan: B must be <= 64");
+    const int njobs = 2 * nchunk;
+    auto sync = torch::zeros({njobs * NATS_SYNC_WORDS + 8},
                              dh_out0.options().dtype(torch::kInt32));
     unsigned* sync_p = (unsigned*)sync.data_ptr<int>();
-    GruPersistBwd p0{dh0c.data_ptr<float>(),
-                     h_all0.data_ptr<float>(),
-                     (const bf16_t*)saved0.data_ptr(),
-                     (const bf16_t*)xc0.data_ptr(),
-                     m0,
-                     (const bf16_t*)Ubwd0.data_ptr(),
-                     dsp + 0 * 2 * ds,
-                     dd + 0,
-                     (bf16_t*)dpre0.data_ptr(),
-                     h00.data_ptr<float>()};
-    GruPersistBwd p1{dh1c.data_ptr<float>(),
-                     h_all1.data_ptr<float>(),
-                     (const bf16_t*)saved1.data_ptr(),
-                     (const bf16_t*)xc1.data_ptr(),
-                     m1,
-                     (const bf16_t*)Ubwd1.data_ptr(),
-                     dsp + 1 * 2 * ds,
-                     dd + (long)B * H,
-                     (bf16_t*)dpre1.data_ptr(),
-                     h00.data_ptr<float>()};
+    auto dstepj = torch::zeros({njobs, 2, 32, K3pad}, optsB);
+    auto ddirj = torch::zeros({njobs, 32, H}, optsF);
+    bf16_t* dspj = (bf16_t*)dstepj.data_ptr();
+    float* ddj = ddirj.data_ptr<float>();
+    GruPersistBwd jobs[4];
+    for (int j = 0; j < njobs; ++j) {
+      const int dir = j & 1;
+      const int ch = j >> 1;
+      const int a = ch * 32;
+      const int Bj = std::min(32, B - a);
+      const torch::Tensor& dhc = dir == 0 ? dh0c : dh1c;
+      const torch::Tensor& h_all = dir == 0 ? h_all0 : h_all1;
+      const torch::Tensor& saved = dir == 0 ? saved0 : saved1;
+      const torch::Tensor& xc = dir == 0 ? xc0 : xc1;
+      const torch::Tensor& Ubwd = dir == 0 ? Ubwd0 : Ubwd1;
+      torch::Tensor& dpre = dir == 0 ? dpre0 : dpre1;
+      const float* mj = dir == 0 ? m0 : m1;
+      jobs[j] = GruPersistBwd{
+          dhc.data_ptr<float>() + (long)a * H,
+          h_all.data_ptr<float>() + (long)a * H,
+          (const bf16_t*)saved.data_ptr() + (long)a * 3 * H,
+          (const bf16_t*)xc.data_ptr() + (long)a * H,
+          mj ? mj + a : nullptr,
+          (const bf16_t*)Ubwd.data_ptr(),
+          dspj + (long)j * 2 * ds,
+          ddj + (long)j * 32 * H,
+          (bf16_t*)dpre.data_ptr() + (long)a * 4 * H,
+          h00.data_ptr<float>() + (long)a * H,
+          Bj,
+          (long)B * H, (long)B * H, (long)B * 3 * H, (long)B * H,
+          (long)B, (long)B * 4 * H};
+    }
+    for (int j = njobs; j < 4; ++j) jobs[j] = jobs[0];
     const char* xpd_env = getenv("NATS_XPD");
-    // 2 XCDs (64 CUs) when the per-direction grid fits comfortably
-    // (LCSTS ngrp=32: +2.5% measured); 4 when 63 WGs would sit 1/CU
-    const int xpd = xpd_env ? atoi(xpd_env) : (ngrp <= 48 ? 2 : 4);
+    const int xpd = xpd_env ? atoi(xpd_env)
+                            : (njobs == 4 ? 2 : (ngrp <= 48 ? 2 : 4));
     hipLaunchKernelGGL(nats_gru_persistent_bwd, dim3(256), dim3(384),
-                       smem_bwd, stream, p0, p1, T, B, H, K3pad, sync_p,
-                       (unsigned)(2 * ngrp), ngrp, xpd);
-    // final dh0 per direction (dstep(0) lives in ping-pong slot 0)
-    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrp), dim3(384), 0,
-                       stream, dsp + 0 * 2 * ds,
-                       (const bf16_t*)Ubwd0.data_ptr(), dd + 0,
-                       dh0_0.data_ptr<float>(), B, H, K3pad);
-    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrp), dim3(384), 0,
-                       stream, dsp + 1 * 2 * ds,
-                       (const bf16_t*)Ubwd1.data_ptr(), dd + (long)B * H,
-                       dh0_1.data_ptr<float>(), B, H, K3pad);
+                       smem_bwd, stream, jobs[0], jobs[1], jobs[2], jobs[3],
+                       T, H, K3pad, sync_p, ngrp, njobs, xpd);
+    // final dh0 per job (dstep(0) lives in ping-pong slot 0); each job
+    // writes its chunk's rows of the (B,H) dh0 output
+    for (int j = 0; j < njobs; ++j) {
+      const int dir = j & 1;
+      const int a = (j >> 1) * 32;
+      const int Bj = std::min(32, B - a);
+      hipLaunchKernelGGL(
+          nats_gru_step_bwd_gemm, dim3(ngrp), dim3(384), 0, stream,
+          dspj + (long)j * 2 * ds,
+          (const bf16_t*)(dir == 0 ? Ubwd0 : Ubwd1).data_ptr(),
+          ddj + (long)j * 32 * H,
+          (dir == 0 ? dh0_0 : dh0_1).data_ptr<float>() + (long)a * H, Bj,
+          H, K3pad);
+    }
     HIP_CHECK(hipGetLastError());
     return {dpre0, dh0_0, dpre1, dh0_1};
   }
 
+  TORCH_CHECK(B <= 32,
+              "bidir scan: B > 32 requires the persistent path "
+              "(gru_persistent_ok)");
   for (int t = T - 1; t >= 0; --t) {
     // parity: step t reads dstep[(t+1)%2], writes dstep[t%2]
     GruBwdArgs a0{
@@ -1311,6 +1372,21 @@ std::vector<torch::Tensor> gru_scan_bwd(torch::Tensor dh_out,
   return {dpre_all, dh_buf};
 }
 
+
+// Can the persistent bidirectional scans take this H (and hence batch
+// up to 64 via chunk jobs)? Mirrors the fwd/bwd gate conditions.
+bool gru_persistent_ok(long H) {
+  const long Hpad = (H + 31) / 32 * 32;
+  const long K3pad = (3 * H + 31) / 32 * 32;
+  const long ngrp = (H + JB - 1) / JB;
+  const size_t smem_fwd =
+      (size_t)3 * JB * Hpad * 2 + sizeof(float) * 3 * 32 * (JB + 1);
+  const size_t smem_bwd =
+      (size_t)JB * K3pad * 2 + sizeof(float) * 3 * 32 * (JB + 1);
+  return (2 * ngrp <= 192) && (smem_fwd <= 150 * 1024) &&
+         (smem_bwd <= 150 * 1024) &&
+         (getenv("NATS_NO_PERSISTENT") == nullptr);
+}
 
 // ---- grid-barrier micro-benchmark: isolates the per-step sync cost of
 // the persistent scans (timing evidence in profiles/README.md). ----
