@@ -82,6 +82,18 @@ def _batch_chunks(B):
             for b0 in range(0, B, MAX_KERNEL_BATCH)]
 
 
+_CHUNK_STREAMS = {}
+
+
+def _chunk_streams(device, n):
+    """Per-device side streams for concurrent batch-chunk chains."""
+    key = device.index or 0
+    pool = _CHUNK_STREAMS.setdefault(key, [])
+    while len(pool) < n:
+        pool.append(torch.cuda.Stream(device=device))
+    return pool[:n]
+
+
 def gru_scan_bidir(xg0, xc0, mask0, U0, Ux0, xg1, xc1, mask1, U1, Ux1):
     """Both encoder directions (independent scans) in one fused launch
     sequence on GPU; two eager scans on CPU. Returns (h_fwd, h_bwd).
@@ -124,12 +136,29 @@ def cond_gru_scan(y_gates, y_cand, mask, init_state, ctx, ctx_mask, pctx, P):
         if B <= MAX_KERNEL_BATCH:
             return cond_gru_scan_hip(y_gates, y_cand, mask, init_state, ctx,
                                      ctx_mask, pctx, P)
-        outs = [cond_gru_scan_hip(
-            y_gates[:, a:b], y_cand[:, a:b],
-            mask[:, a:b] if mask is not None else None,
-            init_state[a:b], ctx[:, a:b],
-            ctx_mask[:, a:b] if ctx_mask is not None else None,
-            pctx[:, a:b], P) for a, b in _batch_chunks(B)]
+        # batches beyond the 32-row MFMA tiling run as CONCURRENT chunk
+        # chains on separate HIP streams: the per-step decoder kernels
+        # occupy a fraction of the chip each, and the chains are
+        # independent (autograd replays each backward on its recorded
+        # stream, so the reverse chains overlap too)
+        chunks = _batch_chunks(B)
+        cur = torch.cuda.current_stream()
+        streams = _chunk_streams(y_gates.device, len(chunks))
+        outs = []
+        for (a, b), s in zip(chunks, streams):
+            s.wait_stream(cur)
+            with torch.cuda.stream(s):
+                outs.append(cond_gru_scan_hip(
+                    y_gates[:, a:b], y_cand[:, a:b],
+                    mask[:, a:b] if mask is not None else None,
+                    init_state[a:b], ctx[:, a:b],
+                    ctx_mask[:, a:b] if ctx_mask is not None else None,
+                    pctx[:, a:b], P))
+        for s in streams:
+            cur.wait_stream(s)
+        for o in outs:
+            for t in o:
+                t.record_stream(cur)
         return tuple(torch.cat([o[i] for o in outs],
                                dim=1 if i < 3 else 0) for i in range(5))
     return eager.cond_gru_scan(y_gates, y_cand, mask, init_state, ctx,
