@@ -123,7 +123,7 @@ def generate_file(model_path, dictionary, source_file, saveto, k=5,
         model = _make_model(model_path, options, devices[0])
         # batch several sentences' beams through each decode step (the
         # fused decoder kernels take up to 32 rows)
-        sent_batch = max(1, 32 // max(k, 1))
+        sent_batch = max(1, 64 // max(k, 1))
         from .batched import gen_sample_batched
         for base in range(0, n_samples, sent_batch):
             chunk = jobs[base:base + sent_batch]
@@ -157,7 +157,7 @@ def generate_file(model_path, dictionary, source_file, saveto, k=5,
                                   state_factor, dev, maxlen))
             p.start()
             procs.append(p)
-        sent_batch = max(1, 32 // max(k, 1))
+        sent_batch = max(1, 64 // max(k, 1))
         for base in range(0, n_samples, sent_batch):
             queue.put(jobs[base:base + sent_batch])
         for _ in range(n_process):

@@ -172,8 +172,30 @@ def cond_gru_step(h_prev, x_g, x_c, ctx, ctx_mask, pctx, acc_ctx, acc_alpha, P):
     """
     if _use_hip(h_prev, ctx):
         from .cond_gru import cond_gru_step_hip
-        return cond_gru_step_hip(h_prev, x_g, x_c, ctx, ctx_mask, pctx,
-                                 acc_ctx, acc_alpha, P)
+        B = h_prev.shape[0]
+        if B <= MAX_KERNEL_BATCH:
+            return cond_gru_step_hip(h_prev, x_g, x_c, ctx, ctx_mask, pctx,
+                                     acc_ctx, acc_alpha, P)
+        # beam batches beyond the 32-row tiling: concurrent chunk chains
+        # on side streams (doubles the serving micro-batch row budget)
+        chunks = _batch_chunks(B)
+        cur = torch.cuda.current_stream()
+        streams = _chunk_streams(h_prev.device, len(chunks))
+        outs = []
+        for (a, b), s in zip(chunks, streams):
+            s.wait_stream(cur)
+            with torch.cuda.stream(s):
+                outs.append(cond_gru_step_hip(
+                    h_prev[a:b], x_g[a:b], x_c[a:b], ctx[:, a:b],
+                    ctx_mask[:, a:b] if ctx_mask is not None else None,
+                    pctx[:, a:b], acc_ctx[a:b], acc_alpha[a:b], P))
+        for s in streams:
+            cur.wait_stream(s)
+        for o in outs:
+            for t in o:
+                t.record_stream(cur)
+        return tuple(torch.cat([o[i] for o in outs], dim=0)
+                     for i in range(5))
     return eager.cond_gru_step(h_prev, x_g, x_c, ctx, ctx_mask, pctx,
                                acc_ctx, acc_alpha, P)
 

@@ -2,7 +2,7 @@
 
 A thin production-serving layer over the batch decode engine
 (nats_amd.decode): requests are coalesced by a background worker into
-micro-batches of up to ``32 // beam`` sentences (the fused decoder
+micro-batches of up to ``64 // beam`` sentences (the fused decoder
 kernels take 32 beam rows per launch, ops/hip/cond_gru.hip) and decoded
 jointly with :func:`nats_amd.decode.batched.gen_sample_batched`, so
 concurrent requests share kernel launches instead of queueing whole
@@ -35,7 +35,7 @@ class SummarizerService:
 
     Thread-safe: ``summarize`` may be called from many request threads;
     a single worker thread owns the model/GPU and drains the queue in
-    batches of up to ``max_batch`` (default: 32 // k beam rows fit one
+    batches of up to ``max_batch`` (default: 64 // k beam rows fit the
     kernel launch), waiting at most ``max_wait_ms`` for co-batchable
     requests once one is pending.
     """
@@ -54,7 +54,7 @@ class SummarizerService:
         self.ctx_factor = float(ctx_factor)
         self.state_factor = float(state_factor)
         self.chr_level = bool(chr_level)
-        self.max_batch = int(max_batch or max(1, 32 // self.k))
+        self.max_batch = int(max_batch or max(1, 64 // self.k))
         self.max_wait_s = float(max_wait_ms) / 1000.0
         self.model_path = str(model_path)
 

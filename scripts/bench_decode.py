@@ -25,7 +25,7 @@ def run(model, xs, k, maxlen, use_graph, lam, batched=False):
     t0 = time.perf_counter()
     if batched:
         from nats_amd.decode.batched import gen_sample_batched
-        sb = max(1, 32 // k)
+        sb = max(1, 64 // k)
         for base in range(0, len(xs), sb):
             outs = gen_sample_batched(model, xs[base:base + sb], k=k,
                                       maxlen=maxlen, use_unk=True,
