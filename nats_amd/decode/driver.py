@@ -76,7 +76,13 @@ def seqs2words(caps, pos, word_idict):
         for w, p in zip(cc, pp):
             if w == 0:
                 break
-            ww.append(word_idict[w])
+            # ids in [len(dict), n_words) are valid model outputs when the
+            # corpus vocabulary is smaller than n_words (a briefly-trained
+            # model emits them); map to UNK like the trainer's sample
+            # printer rather than crashing (the reference would KeyError
+            # here, gen.py:93 — only because its corpora are larger than
+            # its n_words caps)
+            ww.append(word_idict.get(w, "UNK"))
             ww.append("[{0}]".format(p))
         capsw.append(" ".join(ww))
     return capsw
