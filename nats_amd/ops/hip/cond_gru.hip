@@ -211,23 +211,22 @@ __global__ void cond_attn_ctx_partial(
   const int chunk = (Ts + SCH - 1) / SCH;
   const int sbeg = blockIdx.z * chunk;
   const int send = min(Ts, sbeg + chunk);
-  // 8-deep s-unroll: each thread strides 80KB between s-reads, so the
-  // kernel is pure memory-level parallelism — 8 independent loads in
-  // flight per thread (the 4-deep version measured 13.9us = 4.6 TB/s)
-  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
   int s = sbeg;
-  for (; s + 8 <= send; s += 8) {
-#pragma unroll
-    for (int u = 0; u < 8; ++u)
-      acc[u] += (float)ctx_bf[((long)(s + u) * B + b) * C + c] *
-                alphas_t[(long)b * Ts + s + u];
+  for (; s + 4 <= send; s += 4) {
+    s0 += (float)ctx_bf[((long)(s + 0) * B + b) * C + c] *
+          alphas_t[(long)b * Ts + s + 0];
+    s1 += (float)ctx_bf[((long)(s + 1) * B + b) * C + c] *
+          alphas_t[(long)b * Ts + s + 1];
+    s2 += (float)ctx_bf[((long)(s + 2) * B + b) * C + c] *
+          alphas_t[(long)b * Ts + s + 2];
+    s3 += (float)ctx_bf[((long)(s + 3) * B + b) * C + c] *
+          alphas_t[(long)b * Ts + s + 3];
   }
   for (; s < send; ++s)
-    acc[0] += (float)ctx_bf[((long)s * B + b) * C + c] *
-              alphas_t[(long)b * Ts + s];
-  atomicAdd(&ctxpre_f32[(long)b * C + c],
-            ((acc[0] + acc[1]) + (acc[2] + acc[3])) +
-                ((acc[4] + acc[5]) + (acc[6] + acc[7])));
+    s0 += (float)ctx_bf[((long)s * B + b) * C + c] *
+          alphas_t[(long)b * Ts + s];
+  atomicAdd(&ctxpre_f32[(long)b * C + c], s0 + s1 + s2 + s3);
 }
 
 // ---------------- distraction gate + acc_ctx update ----------------
