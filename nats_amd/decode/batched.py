@@ -54,6 +54,12 @@ def gen_sample_batched(model, xs, k=1, maxlen=30, use_unk=False,
     # of S separate T-step launches.
     lens = [int(x.shape[0]) for x in xs]
     Ts = max(lens)
+    if use_graph and device.type == "cuda":
+        # bucket the padded source length so serving traffic with varied
+        # lengths reuses a handful of captured graphs instead of one per
+        # distinct Ts (padding is exact: ctx_mask zeroes the pad columns
+        # through attention and the encoder mask-blend)
+        Ts = (Ts + 63) // 64 * 64
     x_pad = torch.zeros(Ts, S, dtype=torch.int64, device=device)
     src_mask = torch.zeros(Ts, S, device=device)
     for i, x in enumerate(xs):

@@ -43,6 +43,11 @@ def get_stepper(model, ctx0, pctx0, k):
     key = (int(ctx0.shape[0]), int(k))
     st = per_model.get(key)
     if st is None:
+        plain_keys = [kk for kk in per_model
+                      if isinstance(kk, tuple) and kk[0] != "batched"
+                      and kk != "__ver__"]
+        if len(plain_keys) >= 8:
+            per_model.pop(plain_keys[0], None)
         st = GraphDecodeStepper(model, ctx0, pctx0, k)
         per_model[key] = st
     else:
@@ -61,6 +66,11 @@ def get_batched_stepper(model, ctx_pad, ctx_mask, pctx_pad, R):
     key = ("batched", int(ctx_pad.shape[0]), int(ctx_pad.shape[1]), int(R))
     st = per_model.get(key)
     if st is None:
+        # bound capture memory: keep the most recent shapes only
+        batched_keys = [k for k in per_model
+                        if isinstance(k, tuple) and k[0] == "batched"]
+        if len(batched_keys) >= 8:
+            per_model.pop(batched_keys[0], None)
         st = BatchedGraphStepper(model, ctx_pad, ctx_mask, pctx_pad, R)
         per_model[key] = st
     else:
