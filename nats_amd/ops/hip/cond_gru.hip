@@ -148,9 +148,15 @@ __global__ __launch_bounds__(256) void cond_attn_softmax(
   __shared__ float red[256 / NATS_WAVE];
   __shared__ float bcast;
 
+  // masked positions must not influence the max: they are zeroed AFTER
+  // exp, but a padded row's e shifting M changes rounding for the real
+  // rows — decode results would then depend on how far the source was
+  // padded (the graph-decode length bucketing pads to 64)
   float lmax = -INFINITY;
-  for (int s = threadIdx.x; s < Ts; s += blockDim.x)
+  for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
+    if (ctx_mask != nullptr && ctx_mask[(long)s * B + b] == 0.f) continue;
     lmax = fmaxf(lmax, e_buf[(long)s * B + b]);
+  }
 #pragma unroll
   for (int off = NATS_WAVE / 2; off > 0; off >>= 1)
     lmax = fmaxf(lmax, __shfl_down(lmax, off));
@@ -164,7 +170,9 @@ __global__ __launch_bounds__(256) void cond_attn_softmax(
     bcast = M;
   }
   __syncthreads();
-  const float M = bcast;
+  // all-masked rows cannot occur (every sequence has >= 1 real token),
+  // but guard the degenerate -inf max anyway
+  const float M = isfinite(bcast) ? bcast : 0.f;
 
   float lsum = 0.f;
   for (int s = threadIdx.x; s < Ts; s += blockDim.x) {
