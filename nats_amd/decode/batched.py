@@ -15,6 +15,13 @@ are the top-k candidate ids/costs and the selected attention rows for
 the alignment output (two small synchronisations per step — the naive
 numpy bookkeeping round-trip was ~25% of decode time in the kernel
 trace, profiles/decode_kernel_stats.csv).
+
+Numerics note: results are mathematically identical to per-sentence
+gen_sample, but not bitwise — the batched masked-mean init state and
+torch reductions over different padded lengths round differently at the
+last ulp. On a trained model beams are insensitive to this; a
+random-init model with near-tie hypotheses can flip a beam choice
+(observed only with inflated-readout test models).
 """
 
 import math
