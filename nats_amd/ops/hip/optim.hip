@@ -70,14 +70,42 @@ __global__ void fused_adadelta_kernel(
     const float g2 = *g2_in;
     if (g2 > clip_c * clip_c) scale = clip_c / sqrtf(g2);
   }
-  for (long i = threadIdx.x; i < n; i += BLOCK) {
+  // float4 accesses: the scalar version measured 436 us/step for the
+  // 27.5M-param model (~2 TB/s) — 8 scalar 4B streams per element left
+  // too few lines in flight; 16B vectors recover the roofline
+  const long n4 = n & ~(long)3;
+  for (long i = (long)threadIdx.x * 4; i < n4; i += (long)BLOCK * 4) {
+    float4 gv4 = *(float4*)(g + i);
+    float4 r24 = *(float4*)(rg2 + i);
+    float4 ru4 = *(float4*)(ru2 + i);
+    float4 pv4 = *(float4*)(p + i);
+    float* gv = &gv4.x;
+    float* r2 = &r24.x;
+    float* ru = &ru4.x;
+    float* pv = &pv4.x;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      const float gc = gv[k] * scale;
+      const float r2n = rho * r2[k] + (1.f - rho) * gc * gc;
+      const float ud = -sqrtf(ru[k] + eps) / sqrtf(r2n + eps) * gc;
+      ru[k] = rho * ru[k] + (1.f - rho) * ud * ud;
+      r2[k] = r2n;
+      pv[k] += ud;
+      gv[k] = gc;  // leave the clipped gradient visible (parity w/ ref)
+    }
+    *(float4*)(g + i) = gv4;
+    *(float4*)(rg2 + i) = r24;
+    *(float4*)(ru2 + i) = ru4;
+    *(float4*)(p + i) = pv4;
+  }
+  for (long i = n4 + threadIdx.x; i < n; i += BLOCK) {
     const float gv = g[i] * scale;
     const float r2 = rho * rg2[i] + (1.f - rho) * gv * gv;
     rg2[i] = r2;
     const float ud = -sqrtf(ru2[i] + eps) / sqrtf(r2 + eps) * gv;
     ru2[i] = rho * ru2[i] + (1.f - rho) * ud * ud;
     p[i] += ud;
-    g[i] = gv;  // leave the clipped gradient visible (parity with ref)
+    g[i] = gv;
   }
 }
 
