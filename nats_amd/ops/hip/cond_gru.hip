@@ -435,7 +435,8 @@ __global__ void cond_gru1_step_pointwise(
 // last step's scatter/reduce before this kernel runs — stream order) all
 // share one grid-stride index space.
 __global__ void cond_gru1_bwd_pointwise(
-    const float* __restrict__ dh_carry,   // [B][H]
+    const float* __restrict__ dh_carry,   // [B][H] (split-K half 0)
+    const float* __restrict__ dh_carry2,  // [B][H] half 1 or null
     const float* __restrict__ dh2_all_t,  // [B][H] or null
     const bf16_t* __restrict__ saved1_t,  // [B][4H]
     const float* __restrict__ h1_all_t,   // [B][H]
@@ -473,6 +474,7 @@ __global__ void cond_gru1_bwd_pointwise(
     const int b = idx / H;
     const int j = idx % H;
     float dh2 = dh_carry[idx];
+    if (dh_carry2 != nullptr) dh2 += dh_carry2[idx];
     if (dh2_all_t != nullptr) dh2 += dh2_all_t[idx];
     const float r2 = (float)saved1_t[(long)b * 4 * H + j];
     const float u2 = (float)saved1_t[(long)b * 4 * H + H + j];
@@ -604,6 +606,52 @@ __device__ __forceinline__ bf16x8 frag_a_f32pad(const float* A, int row0,
     v[i] = (bf16_t)((r < Rows && k < Kcols) ? A[(long)r * ld + k] : 0.f);
   }
   return v;
+}
+
+// dh_{t-1} carry GEMM with grid.y split-K and CONSUMER-side combine:
+// the unsplit kernel ran 63 blocks (a quarter of the chip) at 12.6 us;
+// round 1's atomic-combine split measured neutral (atomics + re-zero),
+// so here each half writes its own [B][H] partial — full coverage, no
+// zeroing — and the next step's GRU_1 pointwise sums the two.
+__global__ __launch_bounds__(384) void cond_dh_carry_gemm_split(
+    const bf16_t* __restrict__ dstep,  // [32][Kpad]
+    const bf16_t* __restrict__ Wt,     // [ngrp*16][Kpad]
+    const float* __restrict__ ddirect, // [B][H] (added by grid.y == 0)
+    float* __restrict__ out,           // [2][B][H] partials
+    int B, int H, int Kpad) {
+  __shared__ float part[3][32][JB + 1];
+  const int wg = blockIdx.x;
+  const int ks2 = blockIdx.y;        // K half
+  const int wave = threadIdx.x / NATS_WAVE;
+  const int m = wave / 3;
+  const int ks = wave % 3;
+  const int i0 = wg * JB;
+  const int khalf = ((Kpad / 2 + 31) / 32) * 32;
+  const int hbeg = min(Kpad, ks2 * khalf);
+  const int hend = min(Kpad, hbeg + khalf);
+  const int kchunk = (((hend - hbeg) / 3 + 31) / 32) * 32;
+  const int kbeg = hbeg + ks * kchunk;
+  const int kend = min(hend, kbeg + kchunk);
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  NATS_MFMA_KLOOP(acc, dstep, 16 * m, Kpad, Wt, i0, Kpad, kbeg, kend);
+  {
+    const int lane = threadIdx.x & (NATS_WAVE - 1);
+    const int col = lane & 15;
+    const int rbase = 16 * m + (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) part[ks][rbase + i][col] = acc[i];
+  }
+  __syncthreads();
+  float* o = out + (long)ks2 * B * H;
+  for (int idx = threadIdx.x; idx < B * JB; idx += blockDim.x) {
+    const int b = idx / JB;
+    const int cc = idx % JB;
+    const int i = i0 + cc;
+    if (i >= H) continue;
+    float v = part[0][b][cc] + part[1][b][cc] + part[2][b][cc];
+    if (ks2 == 0) v += ddirect[(long)b * H + i];
+    o[(long)b * H + i] = v;
+  }
 }
 
 __global__ __launch_bounds__(384) void cond_dh1_att_gemm(
@@ -1089,7 +1137,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
   auto gdUatt = torch::zeros({A}, optsF);
   auto gdcatt = torch::zeros({1}, optsF);
 
-  auto dh_carry = torch::zeros({B, H}, optsF);
+  auto dh_carry = torch::zeros({2, B, H}, optsF);
   auto dh1_buf = torch::zeros({B, H}, optsF);
   auto ddirect_h1 = torch::empty({B, H}, optsF);
   auto ddirect2 = torch::empty({B, H}, optsF);
@@ -1143,6 +1191,7 @@ std::vector<torch::Tensor> cond_gru_bwd(
     // b1 (fused): GRU_1 pointwise + dctx passthrough + dot_buf re-zero
     hipLaunchKernelGGL(cond_gru1_bwd_pointwise, dim3(pwHC), dim3(256), 0,
                        stream, dh_carry.data_ptr<float>(),
+                       dh_carry.data_ptr<float>() + (long)B * H,
                        dh2_c.data_ptr<float>() + (long)t * B * H,
                        (const bf16_t*)saved1.data_ptr() + (long)t * B * 4 * H,
                        h1_all.data_ptr<float>() + (long)t * B * H,
@@ -1228,15 +1277,18 @@ std::vector<torch::Tensor> cond_gru_bwd(
                        ddirect2.data_ptr<float>(),
                        (bf16_t*)dpre2_all.data_ptr() + (long)t * B * 4 * H, B,
                        H);
-    // b9: dh_{t-1} = [dpr1|dpu1|dpxl1] @ [U|Ux]^T + passthrough
-    hipLaunchKernelGGL(nats_gru_step_bwd_gemm, dim3(ngrpH), dim3(384), 0,
-                       stream, (const bf16_t*)dstep2.data_ptr(),
+    // b9: dh_{t-1} = [dpr1|dpu1|dpxl1] @ [U|Ux]^T + passthrough, split
+    // over K halves written as two partials (summed by the next step's
+    // GRU_1 pointwise — no atomics, no zeroing)
+    hipLaunchKernelGGL(cond_dh_carry_gemm_split, dim3(ngrpH, 2), dim3(384),
+                       0, stream, (const bf16_t*)dstep2.data_ptr(),
                        (const bf16_t*)U2cat.data_ptr(),
                        ddirect2.data_ptr<float>(), dh_carry.data_ptr<float>(),
                        B, H, K3Hpad);
   }
   HIP_CHECK(hipGetLastError());
+  auto dh_carry_sum = dh_carry[0] + dh_carry[1];
   return {dpre1_all, dpre2_all, dctxpre_all, gdUcon, dpstate_all,
-          dpctx_acc, gdDwei, gdUatt, gdcatt, dh_carry, daccC, daccA,
+          dpctx_acc, gdDwei, gdUatt, gdcatt, dh_carry_sum, daccC, daccA,
           gdWcon};
 }
