@@ -23,10 +23,19 @@
 // reductions chunk their serial axis, with fp32 partials summed by the
 // NEXT kernel in the chain rather than an extra pass (gru1/gru2 split-K
 // -> pointwise combine; pstate split-K -> escore/softmax inline sums;
-// escore/scatter A-chunks -> e_buf/daccA atomics). Scratch that is
+// escore/scatter A-chunks -> e_buf/daccA atomics; dh_carry split-K ->
+// next step's GRU_1 pointwise sums the halves). Scratch that is
 // atomically accumulated each step (e_buf, ctxpre_f32) is re-zeroed by
 // its consumer in the same pass, so the steady-state loop launches no
-// memsets. Measured ladder in profiles/README.md.
+// memsets.
+//
+// Round-2 fusions (each GPU-measured; negatives reverted in history):
+// the dctx passthrough + dot-buffer zeroing merged into the GRU_1
+// backward pointwise; the distraction-gate backward plus the
+// dU_con/dW_con column reductions live in the dctx GEMM's epilogue;
+// dh1 += dpstate @ W_att converts fp32 A-fragments in-register; the
+// e-score kernel stages the combined pstate + attention vectors in LDS.
+// Ladder and profiles in profiles/README.md.
 
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
