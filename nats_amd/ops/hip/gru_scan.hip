@@ -509,29 +509,7 @@ __device__ __forceinline__ bf16x8 frag_bt_lds_swz(const bf16_t* lds, int row,
   do {                                                                       \
     const int _ke = (KEND);                                                  \
     int _k = (KBEG);                                                         \
-    if (_k + 256 <= _ke) {                                                   \
-      /* 8-deep A-side pipeline: the A operand is the JUST-PUBLISHED      */ \
-      /* hidden state, L2-cold after the barrier acquire — eight          */ \
-      /* fragments in flight amortise the MALL round trip (4-deep left    */ \
-      /* ~2 RTTs serialized per step at Hpad=1024)                        */ \
-      bf16x8 _a[8];                                                          \
-      _Pragma("unroll")                                                      \
-      for (int _i = 0; _i < 8; ++_i)                                         \
-        _a[_i] = frag_a_rowmajor((APTR), (AROW), (ALD), _k + 32 * _i);       \
-      for (_k += 256; _k + 32 <= _ke; _k += 32) {                            \
-        bf16x8 _an = frag_a_rowmajor((APTR), (AROW), (ALD), _k);             \
-        bf16x8 _b = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 256);        \
-        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a[0], _b, ACC, 0, 0, 0);\
-        _Pragma("unroll")                                                    \
-        for (int _i = 0; _i < 7; ++_i) _a[_i] = _a[_i + 1];                  \
-        _a[7] = _an;                                                         \
-      }                                                                      \
-      _Pragma("unroll")                                                      \
-      for (int _i = 0; _i < 8; ++_i) {                                       \
-        bf16x8 _b = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 256 + 32 * _i);\
-        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a[_i], _b, ACC, 0, 0, 0);\
-      }                                                                      \
-    } else if (_k + 128 <= _ke) {                                            \
+    if (_k + 128 <= _ke) {                                                   \
       bf16x8 _a0 = frag_a_rowmajor((APTR), (AROW), (ALD), _k);               \
       bf16x8 _a1 = frag_a_rowmajor((APTR), (AROW), (ALD), _k + 32);          \
       bf16x8 _a2 = frag_a_rowmajor((APTR), (AROW), (ALD), _k + 64);          \
@@ -561,6 +539,11 @@ __device__ __forceinline__ bf16x8 frag_bt_lds_swz(const bf16_t* lds, int row,
     }                                                                        \
   } while (0)
 
+// One persistent-scan job = one (direction, batch-chunk): pointers are
+// pre-offset to the chunk's first row and strides skip the FULL batch
+// per timestep, so jobs of a B>32 call share the (T, Btot, ...) output
+// tensors without copies. Up to 4 jobs run concurrently in one launch
+// (2 directions x 2 chunks of the 32-row MFMA tiling).
 struct GruPersistFwd {
   const bf16_t* xg;    // -> [t][b][2H] rows of this chunk
   const bf16_t* xc;
