@@ -478,122 +478,6 @@ __device__ __forceinline__ bool nats_grid_barrier(unsigned* sync,
 
 #define NATS_SYNC_WORDS 27
 
-// ---- fence-free h-exchange (guide: "sc1 write-through payload; sc1
-// loads may replace the acquire only when the producer stored sc1").
-// The producing waves drain vmcnt before arriving, the payload is
-// published with 16-B sc1 (write-through) buffer stores, and consumers
-// read it with sc1 buffer loads — no buffer_wbl2 release and no
-// per-CU buffer_inv acquire in the per-step barrier. Measured acquire
-// alone costs ~1.3-1.7 us per CU per step (MI355X_MICROARCH.md).
-
-typedef int i32x4_t __attribute__((ext_vector_type(4)));
-
-template <typename R>
-__device__ __forceinline__ bf16x8 frag_a_sc1(R rsrc, int base_byte,
-                                             int row0, int ld, int k) {
-  const int lane = threadIdx.x & (NATS_WAVE - 1);
-  const int off =
-      base_byte + ((row0 + (lane & 15)) * ld + k + (lane >> 4) * 8) * 2;
-  i32x4_t v = __builtin_amdgcn_raw_buffer_load_b128(rsrc, off, 0, 16);
-  bf16x8 b;
-  __builtin_memcpy(&b, &v, 16);
-  return b;
-}
-
-#define NATS_MFMA_KLOOP_SC1(ACC, RSRC, ABASE, AROW, ALD, LDSB, BROW, BLD,    \
-                            KBEG, KEND)                                      \
-  do {                                                                       \
-    const int _ke = (KEND);                                                  \
-    int _k = (KBEG);                                                         \
-    if (_k + 128 <= _ke) {                                                   \
-      bf16x8 _a0 = frag_a_sc1((RSRC), (ABASE), (AROW), (ALD), _k);           \
-      bf16x8 _a1 = frag_a_sc1((RSRC), (ABASE), (AROW), (ALD), _k + 32);      \
-      bf16x8 _a2 = frag_a_sc1((RSRC), (ABASE), (AROW), (ALD), _k + 64);      \
-      bf16x8 _a3 = frag_a_sc1((RSRC), (ABASE), (AROW), (ALD), _k + 96);      \
-      for (_k += 128; _k + 32 <= _ke; _k += 32) {                            \
-        bf16x8 _an = frag_a_sc1((RSRC), (ABASE), (AROW), (ALD), _k);         \
-        bf16x8 _b = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 128);        \
-        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b, ACC, 0, 0, 0);\
-        _a0 = _a1; _a1 = _a2; _a2 = _a3; _a3 = _an;                          \
-      }                                                                      \
-      {                                                                      \
-        bf16x8 _b = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 128);        \
-        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b, ACC, 0, 0, 0);\
-        _b = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 96);                \
-        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a1, _b, ACC, 0, 0, 0);\
-        _b = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 64);                \
-        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a2, _b, ACC, 0, 0, 0);\
-        _b = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k - 32);                \
-        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a3, _b, ACC, 0, 0, 0);\
-      }                                                                      \
-    } else {                                                                 \
-      for (; _k < _ke; _k += 32) {                                           \
-        bf16x8 _a0 = frag_a_sc1((RSRC), (ABASE), (AROW), (ALD), _k);         \
-        bf16x8 _b0 = frag_bt_lds_swz((LDSB), (BROW), (BLD), _k);             \
-        ACC = __builtin_amdgcn_mfma_f32_16x16x32_bf16(_a0, _b0, ACC, 0, 0, 0);\
-      }                                                                      \
-    }                                                                        \
-  } while (0)
-
-// grid barrier WITHOUT the release/acquire cache fences: valid ONLY
-// when every cross-block payload is sc1-stored and sc1-loaded (above).
-// Keeps the arrive/top/generation structure, the vmcnt drain before
-// arrival (the recipe requires every storing wave drained) and the
-// bounded-spin give-up.
-__device__ __forceinline__ bool nats_grid_barrier_relaxed(
-    unsigned* sync, unsigned epoch, const NatsBarrierCtx& ctx) {
-  __shared__ unsigned ok_sh;
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // publishes drained
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    unsigned* arrive = sync + ctx.bucket;
-    unsigned* top = sync + 8;
-    unsigned* gen = sync + 9 + ctx.bucket;
-    unsigned* give_up = sync + 17;
-    unsigned ok = 1u, spins = 0u;
-    const unsigned prev = __hip_atomic_fetch_add(
-        arrive, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    if (prev + 1u == ctx.nper * epoch) {
-      __hip_atomic_fetch_add(top, 1u, __ATOMIC_RELAXED,
-                             __HIP_MEMORY_SCOPE_AGENT);
-      for (;;) {
-        if (__hip_atomic_load(top, __ATOMIC_RELAXED,
-                              __HIP_MEMORY_SCOPE_AGENT) >= ctx.ntop * epoch)
-          break;
-        if (__hip_atomic_load(give_up, __ATOMIC_RELAXED,
-                              __HIP_MEMORY_SCOPE_AGENT) != 0u ||
-            ++spins > 200000000u) {
-          __hip_atomic_store(give_up, 1u, __ATOMIC_RELAXED,
-                             __HIP_MEMORY_SCOPE_AGENT);
-          ok = 0u;
-          break;
-        }
-        __builtin_amdgcn_s_sleep(2);
-      }
-      __hip_atomic_store(gen, epoch, __ATOMIC_RELAXED,
-                         __HIP_MEMORY_SCOPE_AGENT);
-    } else {
-      for (;;) {
-        if (__hip_atomic_load(gen, __ATOMIC_RELAXED,
-                              __HIP_MEMORY_SCOPE_AGENT) >= epoch)
-          break;
-        if (__hip_atomic_load(give_up, __ATOMIC_RELAXED,
-                              __HIP_MEMORY_SCOPE_AGENT) != 0u ||
-            ++spins > 200000000u) {
-          __hip_atomic_store(give_up, 1u, __ATOMIC_RELAXED,
-                             __HIP_MEMORY_SCOPE_AGENT);
-          ok = 0u;
-          break;
-        }
-        __builtin_amdgcn_s_sleep(2);
-      }
-    }
-    ok_sh = ok;
-  }
-  __syncthreads();
-  return ok_sh != 0u;
-}
-
 // stage a [rows][Kpad] bf16 slice into LDS with the (row&15)<<4 byte-XOR
 // swizzle; read back with the same XOR (write+read swizzled together).
 __device__ __forceinline__ void stage_weights_lds(bf16_t* lds,
@@ -755,14 +639,6 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
   const int g = wave % 3;
   const int j0 = wg * JB;
   const long hb = (long)32 * Hpad;
-  // h is exchanged write-through: hnew staged in LDS, published as 16-B
-  // sc1 stores by wave 0, and read back as sc1 fragments — the per-step
-  // barrier then needs no cache fences (nats_grid_barrier_relaxed)
-  __shared__ __attribute__((aligned(16))) bf16_t hpub[32][JB];
-  for (int i = threadIdx.x; i < 32 * JB; i += blockDim.x)
-    ((bf16_t*)hpub)[i] = (bf16_t)0.f;
-  const auto hrsrc = __builtin_amdgcn_make_buffer_rsrc(
-      (void*)p.h_bf, /*stride*/ 0, (int)(2 * hb * 2), 0x00020000);
 
   // register-carry of this thread's own h columns (column-local: the same
   // thread wrote them last step) — removes an L2 round trip per step
@@ -826,6 +702,7 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
 
     const float* mask_t = p.mask ? p.mask + (long)t * p.smask : nullptr;
     float* h_out = p.h_all + (long)t * p.sh;
+    bf16_t* h_bf_out = p.h_bf + ((t + 1) % 2) * hb;
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
       if (!own[it]) continue;
@@ -846,31 +723,16 @@ __global__ __launch_bounds__(384) void nats_gru_persistent_fwd(
       }
       h_keep[it] = hnew;
       h_out[(long)b * H + j] = hnew;
-      hpub[b][c] = (bf16_t)hnew;
+      h_bf_out[(long)b * Hpad + j] = (bf16_t)hnew;
       pend_r[it] = r;
       pend_u[it] = u;
       pend_px[it] = px;
     }
     pend_saved = p.saved + (long)t * p.ssaved;
     have_pend = true;
-    // publish this step's h tile: 32 rows x 16 cols bf16 = 64 sc1 16-B
-    // chunks, exactly one wave (rows >= B / cols >= H stay zero from the
-    // one-time hpub init, keeping the MFMA padding exact)
-    __syncthreads();
-    if (wave == 0) {
-      const int lane = threadIdx.x;
-      const int r = lane >> 1;
-      const int half = lane & 1;
-      const int off = (int)(((t + 1) % 2) * hb * 2) +
-                      ((int)r * Hpad + j0 + half * 8) * 2;
-      i32x4_t w;
-      __builtin_memcpy(&w, &hpub[r][half * 8], 16);
-      __builtin_amdgcn_raw_buffer_store_b128(w, hrsrc, off, 0, 16);
-    }
     if (unsafe_nobarrier) {  // TIMING EXPERIMENTS ONLY (racy!)
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __syncthreads();
-    } else if (!nats_grid_barrier_relaxed(sync, (unsigned)(t + 1), bctx)) {
+    } else if (!nats_grid_barrier(sync, (unsigned)(t + 1), bctx)) {
       // poison output so a barrier give-up surfaces as NaN, never a hang
       if (threadIdx.x == 0) p.h_all[0] = __builtin_nanf("");
       return;
