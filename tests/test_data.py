@@ -125,3 +125,13 @@ def test_synthetic_batch_layout():
     assert x.shape == (11, 3) and y.shape == (5, 3)
     assert (x[:10] >= 2).all() and (x[10] == 0).all()
     assert xm.dtype == numpy.float32 and (xm == 1).all()
+
+
+def test_seqs2words_oov_ids_map_to_unk():
+    """Model-emitted ids beyond the dictionary (n_words > corpus vocab)
+    must render as UNK, not crash (found by the GPU e2e pipeline: a
+    42-word toy corpus trained at n_words=64 emits ids 42..63)."""
+    from nats_amd.decode.driver import seqs2words
+    word_idict = {0: "<eos>", 1: "UNK", 2: "alpha", 3: "beta"}
+    lines = seqs2words([[2, 45, 3, 0]], [[0, 1, 2, 3]], word_idict)
+    assert lines == ["alpha [0] UNK [1] beta [2]"]
