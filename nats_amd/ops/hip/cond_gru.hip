@@ -132,7 +132,7 @@ __global__ __launch_bounds__(256) void cond_attn_escore(
 }
 
 // ---------------- softmax finish + acc_alpha update (one WG per b) ----
-__global__ __launch_bounds__(256) void cond_attn_softmax(
+__global__ __launch_bounds__(1024) void cond_attn_softmax(
     float* __restrict__ accA,            // [B][Ts] in/out
     float* __restrict__ accA_used_t,     // [B][Ts] out (pre-update copy)
     const float* __restrict__ ctx_mask,  // [Ts][B] or null
@@ -1050,7 +1050,9 @@ std::vector<torch::Tensor> cond_gru_fwd(
                        accA.data_ptr<float>(), Dwei.data_ptr<float>(),
                        Uatt.data_ptr<float>(), catt.data_ptr<float>(),
                        e_buf.data_ptr<float>(), B, Ts, A);
-    hipLaunchKernelGGL(cond_attn_softmax, dim3(B), dim3(256), 0, stream,
+    // 1024 threads: the kernel runs only B blocks (latency-bound) —
+    // wider blocks quarter the serial passes over Ts
+    hipLaunchKernelGGL(cond_attn_softmax, dim3(B), dim3(1024), 0, stream,
                        accA.data_ptr<float>(),
                        accA_used.data_ptr<float>() + (long)t * B * Ts,
                        cmask_p, mt, e_buf.data_ptr<float>(),
