@@ -135,23 +135,3 @@ def test_seqs2words_oov_ids_map_to_unk():
     word_idict = {0: "<eos>", 1: "UNK", 2: "alpha", 3: "beta"}
     lines = seqs2words([[2, 45, 3, 0]], [[0, 1, 2, 3]], word_idict)
     assert lines == ["alpha [0] UNK [1] beta [2]"]
-
-
-def test_iterator_gzip_transparent(tmp_path):
-    """.gz corpora decompress transparently (data_iterator.py fopen)."""
-    import gzip
-    import pickle
-    src = tmp_path / "s.txt.gz"
-    tgt = tmp_path / "t.txt.gz"
-    with gzip.open(src, "wt") as f:
-        f.write("a b\nc a\n")
-    with gzip.open(tgt, "wt") as f:
-        f.write("b\na c\n")
-    dic = tmp_path / "d.pkl"
-    with open(dic, "wb") as f:
-        pickle.dump({"<eos>": 0, "UNK": 1, "a": 2, "b": 3, "c": 4}, f)
-    from nats_amd.data.iterator import TextIterator
-    it = TextIterator(str(src), str(tgt), str(dic), batch_size=2)
-    xs, ys = next(it)
-    assert xs == [[2, 3], [4, 2]]
-    assert ys == [[3], [2, 4]]
